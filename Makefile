@@ -1,0 +1,33 @@
+# Build the xps core extension in-tree (the .so travels to the GPU box).
+HIPCC ?= hipcc
+ARCH ?= gfx950
+PYEXT := $(shell python3-config --extension-suffix)
+PYINC := $(shell python3 -m pybind11 --includes)
+CXXFLAGS := -O3 -std=c++17 -fPIC -Wall -Wno-unused-function --offload-arch=$(ARCH) $(PYINC)
+LDFLAGS := -shared -fPIC
+
+SRCS := csrc/env.cc csrc/wire.cc csrc/tcp.cc csrc/van.cc csrc/postoffice.cc \
+        csrc/customer.cc csrc/resender.cc csrc/hip_util.cc csrc/hip_pool.cc \
+        csrc/gpu_plane.cc csrc/ps.cc csrc/pybind.cc
+HIPSRCS := $(wildcard csrc/*.hip)
+OBJS := $(SRCS:%.cc=build/%.o) $(HIPSRCS:%.hip=build/%.hip.o)
+
+TARGET := ps_lite_amd/_core$(PYEXT)
+
+all: $(TARGET)
+
+build/%.o: %.cc csrc/*.h
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+build/%.hip.o: %.hip csrc/*.h
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(TARGET): $(OBJS)
+	$(HIPCC) $(LDFLAGS) $(OBJS) -o $@
+
+clean:
+	rm -rf build $(TARGET)
+
+.PHONY: all clean

@@ -1,0 +1,57 @@
+// Customer: per-app request tracker + receiving thread.
+//
+// Reference parity: ps-lite include/ps/internal/customer.h,
+// src/customer.cc (tracker_ of (expected, received) pairs, recv thread
+// draining a queue into the app's handle).
+#pragma once
+
+#include <functional>
+#include <thread>
+#include <utility>
+#include <vector>
+
+#include "message.h"
+#include "queue.h"
+
+namespace xps {
+
+class Postoffice;
+
+class Customer {
+ public:
+  using RecvHandle = std::function<void(const Message&)>;
+
+  Customer(int app_id, int customer_id, RecvHandle handle, Postoffice* po);
+  ~Customer();
+  Customer(const Customer&) = delete;
+
+  int app_id() const { return app_id_; }
+  int customer_id() const { return customer_id_; }
+
+  // open a request slot destined for `recver` (node id or group mask);
+  // returns the timestamp used to track its responses.
+  int NewRequest(int recver);
+  // block until all expected responses for `ts` arrived.
+  void WaitRequest(int ts);
+  int NumResponse(int ts);
+  void AddResponse(int ts, int num = 1);
+  bool IsFinished(int ts);
+
+  // called by the Van for every data message addressed to this customer
+  void Accept(Message msg) { queue_.Push(std::move(msg)); }
+
+ private:
+  void Receiving();
+
+  int app_id_;
+  int customer_id_;
+  RecvHandle handle_;
+  Postoffice* po_;
+  ThreadsafeQueue<Message> queue_;
+  std::thread thread_;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::vector<std::pair<int, int>> tracker_;  // (expected, received) per ts
+};
+
+}  // namespace xps

@@ -1,0 +1,458 @@
+// KVWorker / KVServer: sharded key-value push/pull over the Van.
+//
+// Reference parity: ps-lite include/ps/kv_app.h (KVPairs, KVWorker
+// Push/Pull/ZPush/ZPull/Wait :66-318, DefaultSlicer :566, KVServer
+// Process/Response :346-564, KVServerDefaultHandle :431). Re-designed:
+// pull requests advertise the destination HBM-pool offset (meta.addr) so
+// the MI355X data plane can write responses in place over xGMI
+// (zero-copy pull, the rdma_transport.h:369-398 analog); responses
+// delivered in place carry kOptInPlace and skip the worker-side merge.
+#pragma once
+
+#include <algorithm>
+#include <functional>
+#include <unordered_map>
+#include <vector>
+
+#include "hip_pool.h"
+#include "simple_app.h"
+
+namespace xps {
+
+// meta.option bits
+static const int kOptInPlace = 1;  // pull response already written into dst buffer
+
+template <typename V>
+struct KVPairs {
+  SArray<Key> keys;
+  SArray<V> vals;
+  SArray<int> lens;
+};
+
+struct KVMeta {
+  int cmd = 0;
+  bool push = false;
+  bool pull = false;
+  int sender = kEmptyNodeID;
+  int timestamp = -1;
+  int customer_id = 0;
+  Key key = 0;          // fast-path single key
+  uint64_t addr = 0;    // pull: requester's destination pool offset
+  int64_t val_len = 0;
+  int option = 0;
+  int src_dev = kCPU;
+  int dst_dev = kCPU;
+};
+
+template <typename V>
+class KVWorker : public SimpleApp {
+ public:
+  using Callback = std::function<void()>;
+  using SlicedKVs = std::vector<std::pair<bool, KVPairs<V>>>;
+  using Slicer = std::function<void(const KVPairs<V>&, const std::vector<Range>&, SlicedKVs*)>;
+
+  explicit KVWorker(int app_id, int customer_id, int instance_idx = 0)
+      : SimpleApp(Postoffice::GetWorker(instance_idx), DeferCustomer{}) {
+    slicer_ = [this](const KVPairs<V>& s, const std::vector<Range>& r, SlicedKVs* out) {
+      DefaultSlicer(s, r, out);
+    };
+    obj_.reset(new Customer(app_id, customer_id,
+                            [this](const Message& m) { Process(m); }, po_));
+  }
+
+  int Push(const std::vector<Key>& keys, const std::vector<V>& vals,
+           const std::vector<int>& lens = {}, int cmd = 0, const Callback& cb = nullptr) {
+    return ZPush(SArray<Key>(keys), SArray<V>(vals), SArray<int>(lens), cmd, cb);
+  }
+
+  int Pull(const std::vector<Key>& keys, std::vector<V>* vals, std::vector<int>* lens = nullptr,
+           int cmd = 0, const Callback& cb = nullptr) {
+    SArray<Key> skeys(keys);
+    int ts = AddPullRequest_(skeys, vals, lens, cmd, cb);
+    return ts;
+  }
+
+  // zero-copy push: keys/vals/lens remain owned by the caller until done
+  int ZPush(const SArray<Key>& keys, const SArray<V>& vals, const SArray<int>& lens = {},
+            int cmd = 0, const Callback& cb = nullptr) {
+    int ts = obj_->NewRequest(kServerGroup);
+    AddCallback(ts, cb);
+    KVPairs<V> kvs;
+    kvs.keys = keys;
+    kvs.vals = vals;
+    kvs.lens = lens;
+    Send(ts, true, false, cmd, kvs);
+    return ts;
+  }
+
+  // zero-copy pull into a pre-allocated vals buffer (device or host)
+  int ZPull(const SArray<Key>& keys, SArray<V>* vals, SArray<int>* lens = nullptr, int cmd = 0,
+            const Callback& cb = nullptr) {
+    XPS_CHECK(vals && vals->size()) << "ZPull needs a pre-sized vals buffer";
+    int ts = obj_->NewRequest(kServerGroup);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      pull_dst_[ts] = {vals, lens};
+    }
+    AddCallback(ts, cb);
+    KVPairs<V> kvs;
+    kvs.keys = keys;
+    kvs.vals = *vals;  // carried for slicing geometry; not sent in requests
+    if (lens) kvs.lens = *lens;
+    Send(ts, false, true, cmd, kvs);
+    return ts;
+  }
+
+  void Wait(int timestamp) { obj_->WaitRequest(timestamp); }
+
+  void set_slicer(Slicer s) { slicer_ = std::move(s); }
+
+  // Default: binary-search keys into per-server ranges; vals/lens are
+  // zero-copy segments (ps-lite kv_app.h:566-621 behavior).
+  void DefaultSlicer(const KVPairs<V>& send, const std::vector<Range>& ranges, SlicedKVs* sliced) {
+    sliced->resize(ranges.size());
+    size_t n = send.keys.size();
+    std::vector<size_t> pos(ranges.size() + 1);
+    const Key* begin = send.keys.begin();
+    const Key* end = send.keys.end();
+    for (size_t i = 0; i < ranges.size(); ++i) {
+      pos[i] = std::lower_bound(begin, end, ranges[i].begin) - begin;
+    }
+    pos[ranges.size()] = n;
+    // per-key val offsets (elements)
+    std::vector<size_t> val_off(n + 1, 0);
+    if (!send.lens.empty()) {
+      XPS_CHECK_EQ(send.lens.size(), n);
+      for (size_t i = 0; i < n; ++i) val_off[i + 1] = val_off[i] + send.lens[i];
+    } else if (n > 0) {
+      size_t k = send.vals.size() / n;
+      for (size_t i = 0; i <= n; ++i) val_off[i] = i * k;
+    }
+    for (size_t i = 0; i < ranges.size(); ++i) {
+      size_t a = pos[i], b = pos[i + 1];
+      auto& out = (*sliced)[i];
+      out.first = b > a;
+      if (!out.first) continue;
+      out.second.keys = send.keys.Segment(a, b);
+      if (!send.vals.empty()) {
+        out.second.vals = send.vals.Segment(val_off[a], val_off[b]);
+      }
+      if (!send.lens.empty()) {
+        out.second.lens = send.lens.Segment(a, b);
+      }
+    }
+  }
+
+ private:
+  int AddPullRequest_(const SArray<Key>& keys, std::vector<V>* vals, std::vector<int>* lens,
+                      int cmd, const Callback& cb) {
+    // vector-based Pull: size the buffer by asking with lens unknown —
+    // allocate after responses arrive. We use a two-phase: buffer grows in merge.
+    int ts = obj_->NewRequest(kServerGroup);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      vec_pull_dst_[ts] = {vals, lens};
+    }
+    AddCallback(ts, cb);
+    KVPairs<V> kvs;
+    kvs.keys = keys;
+    Send(ts, false, true, cmd, kvs);
+    return ts;
+  }
+
+  void AddCallback(int ts, const Callback& cb) {
+    if (!cb) return;
+    std::lock_guard<std::mutex> lk(mu_);
+    callbacks_[ts] = cb;
+  }
+
+  void Send(int ts, bool push, bool pull, int cmd, KVPairs<V>& kvs) {
+    SlicedKVs sliced;
+    slicer_(kvs, po_->GetServerKeyRanges(), &sliced);
+    int skipped = 0;
+    for (auto& s : sliced)
+      if (!s.first) ++skipped;
+    if (skipped) obj_->AddResponse(ts, skipped);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      expected_[ts] = static_cast<int>(sliced.size()) - skipped;
+    }
+    if (static_cast<int>(sliced.size()) == skipped) {
+      // nothing to send; run callback now
+      RunCallback(ts);
+      return;
+    }
+    // byte offset of each slice's vals within the original vals buffer
+    // (for in-place pull destination advertisement)
+    size_t val_elem_off = 0;
+    for (size_t i = 0; i < sliced.size(); ++i) {
+      if (!sliced[i].first) continue;
+      auto& s = sliced[i].second;
+      Message msg;
+      msg.meta.app_id = obj_->app_id();
+      msg.meta.customer_id = obj_->customer_id();
+      msg.meta.request = true;
+      msg.meta.push = push;
+      msg.meta.pull = pull;
+      msg.meta.head = cmd;
+      msg.meta.timestamp = ts;
+      msg.meta.recver = ServerRankToID(static_cast<int>(i));
+      if (s.keys.size() == 1) msg.meta.key = s.keys[0];
+      msg.meta.src_dev = push ? s.vals.device() : kvs.vals.device();
+      msg.meta.dst_dev = msg.meta.src_dev;
+      msg.AddData(s.keys);
+      if (push) {
+        msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
+        msg.AddData(s.vals);
+        if (!s.lens.empty()) msg.AddData(s.lens);
+      } else {
+        // pull request: keys (+lens geometry) only; advertise destination
+        msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
+        if (!s.vals.empty() && s.vals.on_device()) {
+          uint64_t off = 0;
+          if (HbmPool::Get()->OffsetOf(s.vals.data(), &off)) {
+            msg.meta.addr = off;
+          }
+        }
+        if (!s.lens.empty()) msg.AddData(s.lens);
+      }
+      po_->van()->Send(msg);
+      (void)val_elem_off;
+    }
+  }
+
+  void Process(const Message& msg) {
+    if (msg.meta.request) return;  // workers only receive responses
+    int ts = msg.meta.timestamp;
+    bool last = false;
+    KVPairs<V> kvs;
+    if (msg.meta.pull) {
+      if (!msg.data.empty()) {
+        kvs.keys = SArray<Key>::View(msg.data[0]);
+        if (msg.data.size() > 1) kvs.vals = SArray<V>::View(msg.data[1]);
+        if (msg.data.size() > 2) kvs.lens = SArray<int>::View(msg.data[2]);
+      } else {
+        kvs.keys = SArray<Key>({msg.meta.key});
+      }
+    }
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      if (msg.meta.pull) {
+        auto& got = recv_kvs_[ts];
+        got.emplace_back(kvs, msg.meta.option);
+        last = static_cast<int>(got.size()) >= expected_[ts];
+      } else {
+        int n = ++push_acks_[ts];
+        last = n >= expected_[ts];
+      }
+    }
+    if (last) {
+      if (msg.meta.pull) MergePull(ts);
+      RunCallback(ts);
+    }
+  }
+
+  void MergePull(int ts) {
+    std::vector<std::pair<KVPairs<V>, int>> got;
+    SArray<V>* dst_vals = nullptr;
+    SArray<int>* dst_lens = nullptr;
+    std::vector<V>* vec_vals = nullptr;
+    std::vector<int>* vec_lens = nullptr;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      got.swap(recv_kvs_[ts]);
+      recv_kvs_.erase(ts);
+      auto it = pull_dst_.find(ts);
+      if (it != pull_dst_.end()) {
+        dst_vals = it->second.first;
+        dst_lens = it->second.second;
+        pull_dst_.erase(it);
+      }
+      auto it2 = vec_pull_dst_.find(ts);
+      if (it2 != vec_pull_dst_.end()) {
+        vec_vals = it2->second.first;
+        vec_lens = it2->second.second;
+        vec_pull_dst_.erase(it2);
+      }
+    }
+    // order slices by their first key
+    std::sort(got.begin(), got.end(), [](const auto& a, const auto& b) {
+      Key ka = a.first.keys.empty() ? 0 : a.first.keys[0];
+      Key kb = b.first.keys.empty() ? 0 : b.first.keys[0];
+      return ka < kb;
+    });
+    if (vec_vals) {
+      // vector Pull: concatenate
+      vec_vals->clear();
+      if (vec_lens) vec_lens->clear();
+      for (auto& g : got) {
+        vec_vals->insert(vec_vals->end(), g.first.vals.begin(), g.first.vals.end());
+        if (vec_lens) {
+          for (size_t i = 0; i < g.first.lens.size(); ++i) vec_lens->push_back(g.first.lens[i]);
+        }
+      }
+      return;
+    }
+    if (!dst_vals) return;
+    size_t off = 0;
+    for (auto& g : got) {
+      if (g.second & kOptInPlace) {
+        // data plane already wrote into the destination buffer
+        off += g.first.vals.size();
+        continue;
+      }
+      size_t n = g.first.vals.size();
+      if (n == 0) continue;
+      XPS_CHECK_LE(off + n, dst_vals->size()) << "pull response overflows dst";
+      if (dst_vals->on_device()) {
+        gpu::CopyHostToDevice(dst_vals->data() + off, g.first.vals.data(), n * sizeof(V),
+                              dst_vals->device());
+      } else {
+        memcpy(dst_vals->data() + off, g.first.vals.data(), n * sizeof(V));
+      }
+      off += n;
+    }
+    if (dst_lens && !got.empty()) {
+      size_t loff = 0;
+      for (auto& g : got) {
+        for (size_t i = 0; i < g.first.lens.size() && loff < dst_lens->size(); ++i) {
+          (*dst_lens)[loff++] = g.first.lens[i];
+        }
+      }
+    }
+  }
+
+  void RunCallback(int ts) {
+    Callback cb;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = callbacks_.find(ts);
+      if (it != callbacks_.end()) {
+        cb = it->second;
+        callbacks_.erase(it);
+      }
+      expected_.erase(ts);
+      push_acks_.erase(ts);
+    }
+    if (cb) cb();
+  }
+
+  Slicer slicer_;
+  std::mutex mu_;
+  std::unordered_map<int, Callback> callbacks_;
+  std::unordered_map<int, int> expected_;
+  std::unordered_map<int, int> push_acks_;
+  std::unordered_map<int, std::vector<std::pair<KVPairs<V>, int>>> recv_kvs_;
+  std::unordered_map<int, std::pair<SArray<V>*, SArray<int>*>> pull_dst_;
+  std::unordered_map<int, std::pair<std::vector<V>*, std::vector<int>*>> vec_pull_dst_;
+};
+
+template <typename V>
+class KVServer : public SimpleApp {
+ public:
+  using ReqHandle = std::function<void(const KVMeta&, const KVPairs<V>&, KVServer<V>*)>;
+
+  explicit KVServer(int app_id, int instance_idx = 0)
+      : SimpleApp(Postoffice::GetServer(instance_idx), DeferCustomer{}) {
+    obj_.reset(new Customer(app_id, app_id, [this](const Message& m) { Process(m); }, po_));
+  }
+
+  void set_request_handle(ReqHandle h) { request_handle2_ = std::move(h); }
+
+  void Response(const KVMeta& req, const KVPairs<V>& res = KVPairs<V>()) {
+    Message msg;
+    msg.meta.app_id = obj_->app_id();
+    msg.meta.customer_id = req.customer_id;
+    msg.meta.request = false;
+    msg.meta.push = req.push;
+    msg.meta.pull = req.pull;
+    msg.meta.head = req.cmd;
+    msg.meta.timestamp = req.timestamp;
+    msg.meta.recver = req.sender;
+    msg.meta.key = req.key;
+    msg.meta.addr = req.addr;         // in-place pull destination (pool offset)
+    msg.meta.option = req.option;
+    msg.meta.src_dev = res.vals.device();
+    msg.meta.dst_dev = req.dst_dev;
+    if (req.pull && !res.keys.empty()) {
+      msg.meta.val_len = static_cast<int64_t>(res.vals.nbytes());
+      msg.AddData(res.keys);
+      msg.AddData(res.vals);
+      if (!res.lens.empty()) msg.AddData(res.lens);
+    }
+    po_->van()->Send(msg);
+  }
+
+ private:
+  void Process(const Message& msg) {
+    if (msg.meta.simple_app) {
+      SimpleApp::Process(msg);
+      return;
+    }
+    KVMeta meta;
+    meta.cmd = msg.meta.head;
+    meta.push = msg.meta.push;
+    meta.pull = msg.meta.pull;
+    meta.sender = msg.meta.sender;
+    meta.timestamp = msg.meta.timestamp;
+    meta.customer_id = msg.meta.customer_id;
+    meta.key = msg.meta.key;
+    meta.addr = msg.meta.addr;
+    meta.val_len = msg.meta.val_len;
+    meta.option = msg.meta.option;
+    meta.src_dev = msg.meta.src_dev;
+    meta.dst_dev = msg.meta.dst_dev;
+    KVPairs<V> kvs;
+    if (!msg.data.empty()) {
+      kvs.keys = SArray<Key>::View(msg.data[0]);
+      if (msg.data.size() > 1 && msg.meta.push) kvs.vals = SArray<V>::View(msg.data[1]);
+      size_t lens_idx = msg.meta.push ? 2 : 1;
+      if (msg.data.size() > lens_idx) kvs.lens = SArray<int>::View(msg.data[lens_idx]);
+    } else if (msg.meta.key || msg.meta.val_len) {
+      kvs.keys = SArray<Key>({msg.meta.key});
+    }
+    XPS_CHECK(request_handle2_) << "KVServer has no request handle";
+    request_handle2_(meta, kvs, this);
+  }
+
+  ReqHandle request_handle2_;
+};
+
+// Default CPU handle: store[key] op= vals; pull echoes the store
+// (ps-lite kv_app.h:431-452 behavior).
+template <typename V>
+struct KVServerDefaultHandle {
+  void operator()(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
+    size_t n = kvs.keys.size();
+    KVPairs<V> res;
+    if (req.push) {
+      XPS_CHECK(!kvs.vals.on_device()) << "default handle is CPU-only";
+      size_t off = 0;
+      for (size_t i = 0; i < n; ++i) {
+        size_t len = kvs.lens.empty() ? kvs.vals.size() / n : kvs.lens[i];
+        auto& entry = store[kvs.keys[i]];
+        if (entry.size() < len) entry.resize(len, V(0));
+        for (size_t j = 0; j < len; ++j) entry[j] += kvs.vals[off + j];
+        off += len;
+      }
+    } else if (req.pull) {
+      res.keys = kvs.keys;
+      size_t total = 0;
+      for (size_t i = 0; i < n; ++i) total += store[kvs.keys[i]].size();
+      res.vals.Resize(total);
+      SArray<int> lens(n);
+      size_t off = 0;
+      for (size_t i = 0; i < n; ++i) {
+        auto& entry = store[kvs.keys[i]];
+        std::copy(entry.begin(), entry.end(), res.vals.data() + off);
+        off += entry.size();
+        lens[i] = static_cast<int>(entry.size());
+      }
+      res.lens = lens;
+    }
+    server->Response(req, res);
+  }
+  std::unordered_map<Key, std::vector<V>> store;
+};
+
+}  // namespace xps

@@ -1,0 +1,283 @@
+// Python bindings for the xps core (package ps_lite_amd).
+#include <pybind11/functional.h>
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "gpu_plane.h"
+#include "hip_pool.h"
+#include "hip_util.h"
+#include "kv_app.h"
+#include "ps.h"
+#include "simple_app.h"
+#include "van.h"
+
+namespace py = pybind11;
+using namespace xps;
+
+namespace {
+
+struct PoolBuffer {
+  SArray<char> arr;
+  uintptr_t ptr() const { return reinterpret_cast<uintptr_t>(arr.data()); }
+  size_t nbytes() const { return arr.size(); }
+  int device() const { return arr.device(); }
+  void copy_from(py::array src) {
+    py::buffer_info info = src.request();
+    size_t n = static_cast<size_t>(info.size) * info.itemsize;
+    XPS_CHECK_LE(n, arr.size());
+    gpu::CopyHostToDevice(arr.data(), info.ptr, n, arr.device());
+  }
+  py::array_t<float> to_numpy_f32() const {
+    size_t n = arr.size() / sizeof(float);
+    py::array_t<float> out(n);
+    gpu::CopyDeviceToHost(out.mutable_data(), arr.data(), n * sizeof(float));
+    return out;
+  }
+};
+
+class PyKVWorker {
+ public:
+  PyKVWorker(int app_id, int customer_id) : w_(app_id, customer_id) {}
+
+  int Push(py::array_t<uint64_t> keys, py::array_t<float> vals, py::array_t<int> lens) {
+    // copying variant (safe w.r.t. numpy lifetimes)
+    SArray<Key> k;
+    k.CopyFrom(keys.data(), keys.size());
+    SArray<float> v;
+    v.CopyFrom(vals.data(), vals.size());
+    SArray<int> l;
+    if (lens.size()) l.CopyFrom(lens.data(), lens.size());
+    py::gil_scoped_release rel;
+    return w_.ZPush(k, v, l);
+  }
+
+  py::array_t<float> PullBlocking(py::array_t<uint64_t> keys) {
+    std::vector<Key> k(keys.data(), keys.data() + keys.size());
+    std::vector<float> vals;
+    std::vector<int> lens;
+    {
+      py::gil_scoped_release rel;
+      int ts = w_.Pull(k, &vals, &lens);
+      w_.Wait(ts);
+    }
+    py::array_t<float> out(vals.size());
+    std::copy(vals.begin(), vals.end(), out.mutable_data());
+    return out;
+  }
+
+  int ZPushPtr(py::array_t<uint64_t> keys, uintptr_t vals_ptr, size_t vals_bytes, int device,
+               py::array_t<int> lens) {
+    SArray<Key> k;
+    k.CopyFrom(keys.data(), keys.size());
+    SArray<float> v(reinterpret_cast<float*>(vals_ptr), vals_bytes / sizeof(float), device);
+    SArray<int> l;
+    if (lens.size()) l.CopyFrom(lens.data(), lens.size());
+    py::gil_scoped_release rel;
+    return w_.ZPush(k, v, l);
+  }
+
+  int ZPullPtr(py::array_t<uint64_t> keys, uintptr_t vals_ptr, size_t vals_bytes, int device,
+               py::array_t<int> lens) {
+    SArray<Key> k;
+    k.CopyFrom(keys.data(), keys.size());
+    auto* v = new SArray<float>(reinterpret_cast<float*>(vals_ptr), vals_bytes / sizeof(float),
+                                device);
+    SArray<int>* l = nullptr;
+    if (lens.size()) {
+      l = new SArray<int>();
+      l->CopyFrom(lens.data(), lens.size());
+    }
+    py::gil_scoped_release rel;
+    // delete the temporaries when the pull completes
+    return w_.ZPull(k, v, l, 0, [v, l]() {
+      delete v;
+      delete l;
+    });
+  }
+
+  void Wait(int ts) {
+    py::gil_scoped_release rel;
+    w_.Wait(ts);
+  }
+
+ private:
+  KVWorker<float> w_;
+};
+
+class PyKVServer {
+ public:
+  explicit PyKVServer(int app_id) : s_(app_id) {}
+
+  void SetDefaultHandle() {
+    auto h = std::make_shared<KVServerDefaultHandle<float>>();
+    s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
+      (*h)(m, kvs, srv);
+    });
+  }
+
+  // fn(meta: dict, keys: ndarray[u64], vals: ndarray[f32]) -> ndarray[f32] | None
+  void SetPythonHandle(py::function fn) {
+    s_.set_request_handle([fn](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
+      KVPairs<float> res;
+      {
+        py::gil_scoped_acquire gil;
+        py::dict meta;
+        meta["cmd"] = m.cmd;
+        meta["push"] = m.push;
+        meta["pull"] = m.pull;
+        meta["sender"] = m.sender;
+        meta["timestamp"] = m.timestamp;
+        meta["key"] = m.key;
+        meta["val_len"] = m.val_len;
+        py::array_t<uint64_t> keys(kvs.keys.size());
+        std::copy(kvs.keys.begin(), kvs.keys.end(), keys.mutable_data());
+        py::array_t<float> vals(kvs.vals.size());
+        if (kvs.vals.size()) {
+          XPS_CHECK(!kvs.vals.on_device()) << "python handle got device vals";
+          std::copy(kvs.vals.begin(), kvs.vals.end(), vals.mutable_data());
+        }
+        py::object out = fn(meta, keys, vals);
+        if (m.pull) {
+          XPS_CHECK(!out.is_none()) << "pull handler must return vals";
+          auto arr = py::cast<py::array_t<float>>(out);
+          res.keys = kvs.keys;
+          res.vals.CopyFrom(arr.data(), arr.size());
+          size_t n = kvs.keys.size();
+          SArray<int> lens(n);
+          for (size_t i = 0; i < n; ++i) lens[i] = static_cast<int>(arr.size() / n);
+          res.lens = lens;
+        }
+      }
+      srv->Response(m, res);
+    });
+  }
+
+ private:
+  KVServer<float> s_;
+};
+
+class PySimpleApp {
+ public:
+  // server-side customers register under app_id (the Van routes requests
+  // to (app_id, app_id) on non-worker nodes — same convention as KVServer)
+  PySimpleApp(const std::string& role, int app_id, int customer_id)
+      : app_(app_id, role == "worker" ? customer_id : app_id, GetPO(role)) {}
+
+  void SetRequestHandle(py::function fn) {
+    app_.set_request_handle([fn](const SimpleData& d, SimpleApp* app) {
+      std::string reply;
+      {
+        py::gil_scoped_acquire gil;
+        py::object out = fn(d.head, py::bytes(d.body));
+        if (!out.is_none()) reply = py::cast<std::string>(out);
+      }
+      app->Response(d, reply);
+    });
+  }
+
+  void SetResponseHandle(py::function fn) {
+    app_.set_response_handle([fn](const SimpleData& d, SimpleApp*) {
+      py::gil_scoped_acquire gil;
+      fn(d.head, py::bytes(d.body));
+    });
+  }
+
+  int Request(int head, const std::string& body, int recver) {
+    py::gil_scoped_release rel;
+    return app_.Request(head, body, recver);
+  }
+
+  void Wait(int ts) {
+    py::gil_scoped_release rel;
+    app_.Wait(ts);
+  }
+
+ private:
+  SimpleApp app_;
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "xps: MI355X-native parameter-server core";
+
+  m.def("init_env", [](const std::map<std::string, std::string>& kv) {
+    Environment::Get()->Init(kv);
+  });
+  m.def("env", [](const std::string& k) { return Environment::Get()->GetStr(k, ""); });
+  m.def("start",
+        [](int customer_id, const std::string& role, int rank, bool barrier, int device) {
+          py::gil_scoped_release rel;
+          Start(customer_id, role, rank, barrier, device);
+        },
+        py::arg("customer_id") = 0, py::arg("role") = "worker", py::arg("rank") = -1,
+        py::arg("barrier") = true, py::arg("device") = -2);
+  m.def("finalize",
+        [](int customer_id, const std::string& role, bool barrier) {
+          py::gil_scoped_release rel;
+          Finalize(customer_id, role, barrier);
+        },
+        py::arg("customer_id") = 0, py::arg("role") = "worker", py::arg("barrier") = true);
+  m.def("clear_registry", []() { Postoffice::ClearRegistry(); });
+  m.def("barrier", [](const std::string& role, int group) {
+    py::gil_scoped_release rel;
+    GetPO(role)->Barrier(0, group);
+  });
+  m.def("num_workers", &NumWorkers);
+  m.def("num_servers", &NumServers);
+  m.def("my_rank", [](const std::string& role) { return GetPO(role)->my_rank(); });
+  m.def("node_id", [](const std::string& role) { return GetPO(role)->node_id(); });
+  m.def("send_bytes", [](const std::string& role) {
+    auto* van = GetPO(role)->van();
+    return van ? van->send_bytes_.load() : int64_t(0);
+  });
+  m.def("recv_bytes", [](const std::string& role) {
+    auto* van = GetPO(role)->van();
+    return van ? van->recv_bytes_.load() : int64_t(0);
+  });
+  m.def("gpu_count", []() { return gpu::DeviceCount(); });
+
+  m.attr("SCHEDULER_GROUP") = kScheduler;
+  m.attr("SERVER_GROUP") = kServerGroup;
+  m.attr("WORKER_GROUP") = kWorkerGroup;
+
+  m.def("pool_init", [](int device, size_t capacity) { HbmPool::Get()->Init(device, capacity); },
+        py::arg("device"), py::arg("capacity") = 0);
+  m.def("pool_in_use", []() { return HbmPool::Get()->bytes_in_use(); });
+  py::class_<PoolBuffer>(m, "PoolBuffer")
+      .def_property_readonly("ptr", &PoolBuffer::ptr)
+      .def_property_readonly("nbytes", &PoolBuffer::nbytes)
+      .def_property_readonly("device", &PoolBuffer::device)
+      .def("copy_from", &PoolBuffer::copy_from)
+      .def("to_numpy_f32", &PoolBuffer::to_numpy_f32);
+  m.def("pool_alloc", [](size_t nbytes) {
+    PoolBuffer b;
+    b.arr = HbmPool::Get()->AllocArray(nbytes);
+    return b;
+  });
+
+  py::class_<PyKVWorker>(m, "KVWorker")
+      .def(py::init<int, int>(), py::arg("app_id") = 0, py::arg("customer_id") = 0)
+      .def("push", &PyKVWorker::Push, py::arg("keys"), py::arg("vals"),
+           py::arg("lens") = py::array_t<int>())
+      .def("pull", &PyKVWorker::PullBlocking, py::arg("keys"))
+      .def("zpush_ptr", &PyKVWorker::ZPushPtr, py::arg("keys"), py::arg("vals_ptr"),
+           py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>())
+      .def("zpull_ptr", &PyKVWorker::ZPullPtr, py::arg("keys"), py::arg("vals_ptr"),
+           py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>())
+      .def("wait", &PyKVWorker::Wait);
+
+  py::class_<PyKVServer>(m, "KVServer")
+      .def(py::init<int>(), py::arg("app_id") = 0)
+      .def("set_default_handle", &PyKVServer::SetDefaultHandle)
+      .def("set_python_handle", &PyKVServer::SetPythonHandle);
+
+  py::class_<PySimpleApp>(m, "SimpleApp")
+      .def(py::init<const std::string&, int, int>(), py::arg("role"), py::arg("app_id") = 10,
+           py::arg("customer_id") = 0)
+      .def("set_request_handle", &PySimpleApp::SetRequestHandle)
+      .def("set_response_handle", &PySimpleApp::SetResponseHandle)
+      .def("request", &PySimpleApp::Request)
+      .def("wait", &PySimpleApp::Wait);
+}

@@ -1,0 +1,60 @@
+"""ps_lite_amd — MI355X-native parameter-server communication framework.
+
+A from-scratch rebuild of the capabilities of bytedance/ps-lite (byteps
+branch) designed for AMD Instinct MI355X: KVWorker/KVServer ZPush/ZPull
+over a worker/server/scheduler process model, with device payloads in a
+pre-registered HBM pool, a TCP side channel for control, and a same-host
+data plane on shm rings + hipIpc/xGMI (see docs/DESIGN.md).
+"""
+
+import os
+
+# the extension must be the in-tree build (fails loudly if missing)
+try:
+    from . import _core
+except ImportError as e:  # pragma: no cover
+    raise ImportError(
+        "ps_lite_amd._core extension not built. Run `make` at the repo root "
+        "(hipcc --offload-arch=gfx950)."
+    ) from e
+
+from ._core import (  # noqa: F401
+    KVWorker,
+    KVServer,
+    SimpleApp,
+    PoolBuffer,
+    SCHEDULER_GROUP,
+    SERVER_GROUP,
+    WORKER_GROUP,
+    barrier,
+    clear_registry,
+    finalize,
+    gpu_count,
+    init_env,
+    my_rank,
+    node_id,
+    num_servers,
+    num_workers,
+    pool_alloc,
+    pool_in_use,
+    pool_init,
+    recv_bytes,
+    send_bytes,
+    start,
+)
+
+__version__ = "0.1.0"
+
+
+def setup_env(num_workers, num_servers, root_uri="127.0.0.1", root_port=9100, **extra):
+    """Set the DMLC_* environment both for this process and for children."""
+    env = {
+        "DMLC_NUM_WORKER": str(num_workers),
+        "DMLC_NUM_SERVER": str(num_servers),
+        "DMLC_PS_ROOT_URI": root_uri,
+        "DMLC_PS_ROOT_PORT": str(root_port),
+    }
+    env.update({k: str(v) for k, v in extra.items()})
+    os.environ.update(env)
+    init_env(env)
+    return env

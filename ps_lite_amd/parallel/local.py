@@ -1,0 +1,106 @@
+"""Local multi-process cluster launcher.
+
+Reference parity: ps-lite tests/local.sh + tracker/dmlc_local.py — spawn
+1 scheduler + S servers + W workers as OS processes on this host, wired
+by DMLC_* env vars. Used by the CPU test-suite and bench.py.
+"""
+
+import multiprocessing as mp
+import os
+import random
+import traceback
+
+
+def _default_server_fn(ps, rank):
+    server = ps.KVServer(0)
+    server.set_default_handle()
+    return None, server  # keep the server alive until finalize
+
+
+def _child_main(role, rank, env, fn, args, q, device):
+    try:
+        os.environ.update(env)
+        import ps_lite_amd as ps
+
+        ps.init_env(env)
+        ps.start(role=role, rank=rank, device=device)
+        result = None
+        keepalive = None
+        if fn is not None:
+            result = fn(ps, rank, *args)
+            # convention: fn may return (payload, keepalive) where keepalive
+            # holds live objects (e.g. a KVServer) until the finalize barrier
+            if isinstance(result, tuple) and len(result) == 2:
+                result, keepalive = result
+        ps.finalize(role=role)
+        del keepalive
+        q.put((role, rank, "ok", result))
+    except Exception:
+        q.put((role, rank, "error", traceback.format_exc()))
+
+
+class LocalCluster:
+    """Spawn scheduler + servers + workers; collect worker results."""
+
+    def __init__(self, num_workers, num_servers, env_extra=None, root_port=None,
+                 joint=False, devices=None):
+        self.num_workers = num_workers
+        self.num_servers = num_servers
+        self.joint = joint
+        self.devices = devices or {}
+        port = root_port or random.randint(20000, 50000)
+        self.env = {
+            "DMLC_NUM_WORKER": str(num_workers),
+            "DMLC_NUM_SERVER": str(num_servers),
+            "DMLC_PS_ROOT_URI": "127.0.0.1",
+            "DMLC_PS_ROOT_PORT": str(port),
+        }
+        if env_extra:
+            self.env.update({k: str(v) for k, v in env_extra.items()})
+
+    def run(self, worker_fn, server_fn=None, worker_args=(), timeout=120):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = []
+
+        def spawn(role, rank, fn, args, device=-1):
+            p = ctx.Process(target=_child_main, args=(role, rank, self.env, fn, args, q, device))
+            p.daemon = True
+            p.start()
+            procs.append(p)
+
+        spawn("scheduler", -1, None, ())
+        if self.joint:
+            for r in range(self.num_workers):
+                spawn("joint", r, worker_fn, worker_args, self.devices.get(r, -1))
+        else:
+            sfn = server_fn or _default_server_fn
+            for r in range(self.num_servers):
+                spawn("server", r, sfn, (), self.devices.get(("server", r), -1))
+            for r in range(self.num_workers):
+                spawn("worker", r, worker_fn, worker_args, self.devices.get(("worker", r), -1))
+
+        expected = len(procs)
+        results = []
+        for _ in range(expected):
+            role, rank, status, payload = q.get(timeout=timeout)
+            if status == "error":
+                for p in procs:
+                    p.terminate()
+                raise RuntimeError(f"{role}:{rank} failed:\n{payload}")
+            results.append((role, rank, payload))
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        out = {}
+        for role, rank, payload in results:
+            if role in ("worker", "joint"):
+                out[rank] = payload
+        return out
+
+
+def launch_local(num_workers, num_servers, worker_fn, server_fn=None, env_extra=None,
+                 joint=False, timeout=120, worker_args=()):
+    c = LocalCluster(num_workers, num_servers, env_extra=env_extra, joint=joint)
+    return c.run(worker_fn, server_fn=server_fn, timeout=timeout, worker_args=worker_args)

@@ -1,0 +1,162 @@
+"""KV push/pull correctness over the localhost TCP van (CPU).
+
+Restores the upstream-style unit coverage the byteps branch dropped
+(test_connection / test_kv_app / test_kv_app_multi_workers).
+"""
+import threading
+
+import numpy as np
+import pytest
+
+import ps_lite_amd as ps
+from ps_lite_amd.parallel import launch_local
+
+_PORT = [21000]
+
+
+def _inproc_cluster(num_workers=1, num_servers=1):
+    _PORT[0] += 7
+    ps.setup_env(num_workers, num_servers, root_port=_PORT[0])
+    ths = [
+        threading.Thread(target=ps.start, kwargs=dict(role=r, device=-1))
+        for r in ("scheduler", "server", "worker")
+    ]
+    [t.start() for t in ths]
+    [t.join() for t in ths]
+
+
+def _inproc_teardown():
+    ths = [
+        threading.Thread(target=ps.finalize, kwargs=dict(role=r))
+        for r in ("scheduler", "server", "worker")
+    ]
+    [t.start() for t in ths]
+    [t.join() for t in ths]
+    ps.clear_registry()
+
+
+def test_push_pull_default_handle():
+    _inproc_cluster()
+    try:
+        server = ps.KVServer(0)
+        server.set_default_handle()
+        worker = ps.KVWorker(0, 0)
+        keys = np.array([1, 5, 9], dtype=np.uint64)
+        vals = np.arange(9, dtype=np.float32)
+        lens = np.array([3, 3, 3], dtype=np.int32)
+        n_push = 4
+        tss = [worker.push(keys, vals, lens) for _ in range(n_push)]
+        for ts in tss:
+            worker.wait(ts)
+        out = worker.pull(keys)
+        assert np.allclose(out, n_push * vals)
+    finally:
+        _inproc_teardown()
+
+
+def test_python_handle_assign():
+    _inproc_cluster()
+    try:
+        store = {}
+
+        def handle(meta, keys, vals):
+            if meta["push"]:
+                off = 0
+                k = len(vals) // len(keys)
+                for key in keys:
+                    store[int(key)] = vals[off:off + k].copy()
+                    off += k
+                return None
+            return np.concatenate([store[int(key)] for key in keys])
+
+        server = ps.KVServer(0)
+        server.set_python_handle(handle)
+        worker = ps.KVWorker(0, 0)
+        keys = np.array([2, 4], dtype=np.uint64)
+        vals = np.array([1.0, 2.0, 3.0, 4.0], dtype=np.float32)
+        ts = worker.push(keys, vals, np.array([2, 2], dtype=np.int32))
+        worker.wait(ts)
+        out = worker.pull(keys)
+        assert np.allclose(out, vals)
+    finally:
+        _inproc_teardown()
+
+
+def test_zpush_zpull_host_buffers():
+    _inproc_cluster()
+    try:
+        server = ps.KVServer(0)
+        server.set_default_handle()
+        worker = ps.KVWorker(0, 0)
+        keys = np.array([7], dtype=np.uint64)
+        vals = np.full(1024, 2.5, dtype=np.float32)
+        ts = worker.zpush_ptr(keys, vals.ctypes.data, vals.nbytes, -1,
+                              np.array([1024], dtype=np.int32))
+        worker.wait(ts)
+        dst = np.zeros(1024, dtype=np.float32)
+        ts = worker.zpull_ptr(keys, dst.ctypes.data, dst.nbytes, -1,
+                              np.array([1024], dtype=np.int32))
+        worker.wait(ts)
+        assert np.allclose(dst, vals)
+    finally:
+        _inproc_teardown()
+
+
+# ---------------- multi-process tests (the reference's local.sh pattern) ---
+
+
+def _worker_push_pull(ps_mod, rank):
+    worker = ps_mod.KVWorker(0, 0)
+    num_servers = ps_mod.num_servers()
+    # keys spanning every server's range: server i owns [i*2^64/S, ...)
+    step = (1 << 64) // num_servers
+    keys = np.array([i * step + 17 for i in range(num_servers)], dtype=np.uint64)
+    vals = (np.arange(8 * num_servers) + rank).astype(np.float32)
+    lens = np.full(num_servers, 8, dtype=np.int32)
+    ts = worker.push(keys, vals, lens)
+    worker.wait(ts)
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    out = worker.pull(keys)
+    return out.tolist()
+
+
+def test_multiprocess_2workers_2servers():
+    results = launch_local(2, 2, _worker_push_pull, timeout=180)
+    assert set(results.keys()) == {0, 1}
+    # sum over both workers: vals_r = arange(16) + r => total = 2*arange(16)+1
+    expect = 2 * np.arange(16, dtype=np.float32) + 1
+    for rank, out in results.items():
+        assert np.allclose(np.array(out), expect), (rank, out)
+
+
+def _worker_single(ps_mod, rank):
+    worker = ps_mod.KVWorker(0, 0)
+    keys = np.array([3], dtype=np.uint64)
+    vals = np.ones(64, dtype=np.float32) * (rank + 1)
+    ts = worker.push(keys, vals, np.array([64], dtype=np.int32))
+    worker.wait(ts)
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    return worker.pull(keys).tolist()
+
+
+def _joint_fn(ps_mod, rank):
+    server = ps_mod.KVServer(0)
+    server.set_default_handle()
+    # one worker instance per joint process: a worker-group barrier
+    # synchronizes every process after server creation
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    out = _worker_single(ps_mod, rank)
+    return out, server
+
+
+def test_multiprocess_joint_mode():
+    results = launch_local(2, 2, _joint_fn, joint=True, timeout=180)
+    expect = np.full(64, 3.0)  # ranks 1+2
+    for rank, out in results.items():
+        assert np.allclose(np.array(out), expect)
+
+
+def test_resend_with_drop():
+    env = {"PS_RESEND": "1", "PS_RESEND_TIMEOUT": "200", "PS_DROP_MSG": "10"}
+    results = launch_local(1, 1, _worker_single, env_extra=env, timeout=180)
+    assert np.allclose(np.array(results[0]), np.ones(64))
