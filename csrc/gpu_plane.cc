@@ -1,38 +1,409 @@
-// Placeholder plane: fills the Node's pool/device fields so ADD_NODE
-// carries the hipIpc handle; the fast send path lands in the next phase
-// (shm rings + xGMI copies). TCP staging keeps device payloads correct
-// meanwhile.
 #include "gpu_plane.h"
+
+#include <hip/hip_runtime.h>
+#include <unistd.h>
 
 #include <cstring>
 
 #include "hip_pool.h"
 #include "postoffice.h"
+#include "wire.h"
 
 namespace xps {
 
-class StubPlane : public DataPlane {
- public:
-  StubPlane(Postoffice* po, int device) : po_(po), device_(device) {}
-  bool CanSend(const Message&, const Node&) override { return false; }
-  int64_t Send(Message&, const Node&) override { return -1; }
-  void FillSelf(Node* self) override {
-    self->dev_id = device_;
+#define XPS_HIP_CHECK(cmd)                                                            \
+  do {                                                                                \
+    hipError_t e_ = (cmd);                                                            \
+    XPS_CHECK(e_ == hipSuccess) << "HIP error: " << hipGetErrorString(e_) << " in " #cmd; \
+  } while (0)
+
+namespace {
+// process-wide dedup of hipIpcOpenMemHandle (two vans of a joint process
+// and multiple planes share peer pool mappings; never closed — pools are
+// process-lifetime)
+std::mutex g_map_mu;
+struct HandleKey {
+  char h[kIpcHandleBytes];
+  bool operator==(const HandleKey& o) const { return memcmp(h, o.h, kIpcHandleBytes) == 0; }
+};
+struct HandleHash {
+  size_t operator()(const HandleKey& k) const {
+    size_t v = 1469598103934665603ull;
+    for (char c : k.h) {
+      v ^= static_cast<unsigned char>(c);
+      v *= 1099511628211ull;
+    }
+    return v;
+  }
+};
+std::unordered_map<HandleKey, void*, HandleHash> g_mapped;
+}  // namespace
+
+GpuPlane::GpuPlane(Postoffice* po, int device) : po_(po), device_(device) {
+  my_host_hash_ = HostHash();
+}
+
+GpuPlane::~GpuPlane() { Stop(); }
+
+void GpuPlane::FillSelf(Node* self) {
+  self->dev_id = device_;
+  auto* pool = HbmPool::Get();
+  if (pool->initialized()) {
+    self->pool_capacity = pool->capacity();
+    memcpy(self->pool_handle, pool->ipc_handle(), kIpcHandleBytes);
+  }
+  if (!started_) {
+    XPS_CHECK(in_ring_.Create(self->shm_uid)) << "cannot create shm ring";
+    started_ = true;
+    poll_thread_ = std::thread([this] { RingPollLoop(); });
+    comp_thread_ = std::thread([this] { CompletionLoop(); });
+  }
+}
+
+void GpuPlane::Stop() {
+  if (stop_.exchange(true)) return;
+  if (poll_thread_.joinable()) poll_thread_.join();
+  if (comp_thread_.joinable()) comp_thread_.join();
+  // drain pending sends synchronously so responses are not lost on stop
+  {
+    std::lock_guard<std::mutex> lk(pend_mu_);
+    for (auto& kv : pending_) {
+      for (auto& p : kv.second) {
+        hipEventSynchronize(p.ev);
+        Peer* peer = GetPeer(p.peer_id);
+        if (peer && EnsureRing(peer)) {
+          peer->ring.Push(p.payload.data(), static_cast<uint32_t>(p.payload.size()));
+        }
+        hipEventDestroy(p.ev);
+      }
+    }
+    pending_.clear();
+  }
+  in_ring_.CloseAndUnlink();
+  std::lock_guard<std::mutex> lk(ev_mu_);
+  for (auto ev : event_pool_) hipEventDestroy(ev);
+  event_pool_.clear();
+}
+
+void GpuPlane::OnPeer(const Node& peer) {
+  if (peer.host_hash != my_host_hash_) return;
+  std::lock_guard<std::mutex> lk(peers_mu_);
+  auto& p = peers_[peer.id];
+  if (!p) p.reset(new Peer());
+  p->node = peer;
+}
+
+GpuPlane::Peer* GpuPlane::GetPeer(int id) {
+  std::lock_guard<std::mutex> lk(peers_mu_);
+  auto& p = peers_[id];
+  if (!p) {
+    p.reset(new Peer());
+    p->node = po_->van()->GetNode(id);
+  }
+  return p.get();
+}
+
+bool GpuPlane::EnsureRing(Peer* p) {
+  std::lock_guard<std::mutex> lk(p->mu);
+  if (p->ring.ok()) return true;
+  if (p->ring_tried) return false;
+  p->ring_tried = true;
+  if (p->node.shm_uid == 0) return false;
+  return p->ring.Open(p->node.shm_uid);
+}
+
+void* GpuPlane::PeerPoolBase(Peer* p) {
+  {
+    std::lock_guard<std::mutex> lk(p->mu);
+    if (p->pool_tried) return p->pool_base;
+    p->pool_tried = true;
+  }
+  void* base = nullptr;
+  if (p->node.pool_capacity) {
     auto* pool = HbmPool::Get();
-    if (pool->initialized()) {
-      self->pool_capacity = pool->capacity();
-      memcpy(self->pool_handle, pool->ipc_handle(), kIpcHandleBytes);
+    if (pool->initialized() &&
+        memcmp(p->node.pool_handle, pool->ipc_handle(), kIpcHandleBytes) == 0) {
+      base = pool->base();  // same process (joint): use the local mapping
+    } else {
+      HandleKey key;
+      memcpy(key.h, p->node.pool_handle, kIpcHandleBytes);
+      std::lock_guard<std::mutex> lk(g_map_mu);
+      auto it = g_mapped.find(key);
+      if (it != g_mapped.end()) {
+        base = it->second;
+      } else {
+        XPS_HIP_CHECK(hipSetDevice(device_));
+        hipIpcMemHandle_t h;
+        memcpy(&h, p->node.pool_handle, sizeof(h));
+        hipError_t e = hipIpcOpenMemHandle(&base, h, hipIpcMemLazyEnablePeerAccess);
+        if (e != hipSuccess) {
+          XPS_LOG(Warning) << "hipIpcOpenMemHandle(peer " << p->node.id
+                           << ") failed: " << hipGetErrorString(e);
+          base = nullptr;
+        }
+        g_mapped[key] = base;
+      }
     }
   }
+  std::lock_guard<std::mutex> lk(p->mu);
+  p->pool_base = base;
+  return base;
+}
 
- private:
-  Postoffice* po_;
-  int device_;
-};
+hipStream_t GpuPlane::StreamForPeer(int node_id) {
+  Peer* p = GetPeer(node_id);
+  std::lock_guard<std::mutex> lk(p->mu);
+  if (!p->stream) {
+    XPS_HIP_CHECK(hipSetDevice(device_));
+    XPS_HIP_CHECK(hipStreamCreateWithFlags(&p->stream, hipStreamNonBlocking));
+  }
+  return p->stream;
+}
+
+hipEvent_t GpuPlane::GetEvent() {
+  {
+    std::lock_guard<std::mutex> lk(ev_mu_);
+    if (!event_pool_.empty()) {
+      hipEvent_t ev = event_pool_.back();
+      event_pool_.pop_back();
+      return ev;
+    }
+  }
+  hipEvent_t ev;
+  XPS_HIP_CHECK(hipSetDevice(device_));
+  XPS_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  return ev;
+}
+
+void GpuPlane::PutEvent(hipEvent_t ev) {
+  std::lock_guard<std::mutex> lk(ev_mu_);
+  event_pool_.push_back(ev);
+}
+
+bool GpuPlane::CanSend(const Message& msg, const Node& peer) {
+  if (!started_ || stop_.load()) return false;
+  if (peer.host_hash != my_host_hash_ || peer.shm_uid == 0) return false;
+  size_t est = 160 + msg.meta.body.size();
+  for (size_t i = 0; i < msg.data.size(); ++i) {
+    const auto& d = msg.data[i];
+    if (d.on_device()) {
+      uint64_t off;
+      if (!HbmPool::Get()->OffsetOf(d.data(), &off)) return false;  // not our pool -> TCP
+      est += 24;
+      // device vals in a pull response need an in-place destination
+      if (!msg.meta.request && msg.meta.pull && i == 1) {
+        if (!(msg.meta.option & kOptPullAddr) || peer.pool_capacity == 0) return false;
+      }
+    } else {
+      est += 24 + d.size();
+    }
+  }
+  return est <= ShmRing::MaxPayload();
+}
+
+bool GpuPlane::Serialize(const Message& msg, const std::vector<char>& by_ref, std::string* out) {
+  std::string meta;
+  PackMeta(msg.meta, &meta);
+  out->clear();
+  out->reserve(meta.size() + 64);
+  ByteWriter w(out);
+  w.U64(meta.size());
+  w.Raw(meta.data(), meta.size());
+  w.U8(static_cast<uint8_t>(msg.data.size()));
+  for (size_t i = 0; i < msg.data.size(); ++i) {
+    const auto& d = msg.data[i];
+    if (by_ref[i]) {
+      uint64_t off = 0;
+      XPS_CHECK(HbmPool::Get()->OffsetOf(d.data(), &off));
+      w.U8(1);
+      w.U64(off);
+      w.U64(d.size());
+    } else {
+      w.U8(0);
+      w.U64(d.size());
+      w.Raw(d.data(), d.size());
+    }
+  }
+  return out->size() <= ShmRing::MaxPayload();
+}
+
+int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
+  Peer* p = GetPeer(peer_node.id);
+  {
+    std::lock_guard<std::mutex> lk(p->mu);
+    if (p->node.shm_uid == 0) p->node = peer_node;
+  }
+  bool response = !msg.meta.request;
+  if (!EnsureRing(p)) {
+    // TCP fallback for a response must not outrun handler kernels still
+    // running on this peer's stream (the worker may reuse buffers on ack)
+    if (response) hipStreamSynchronize(StreamForPeer(peer_node.id));
+    return -1;
+  }
+
+  // ---- pull response with device vals: one-sided xGMI write ----------
+  if (response && msg.meta.pull && msg.data.size() > 1 && msg.data[1].on_device() &&
+      (msg.meta.option & kOptPullAddr)) {
+    void* base = PeerPoolBase(p);
+    if (!base) return -1;
+    SArray<char> vals = msg.data[1];
+    XPS_CHECK_LE(msg.meta.addr + vals.size(), p->node.pool_capacity)
+        << "in-place pull write out of peer pool bounds";
+    char* dst = static_cast<char*>(base) + msg.meta.addr;
+    hipStream_t stream = StreamForPeer(peer_node.id);
+    XPS_HIP_CHECK(hipSetDevice(device_));
+    XPS_HIP_CHECK(hipMemcpyAsync(dst, vals.data(), vals.size(), hipMemcpyDefault, stream));
+    Message meta_msg;
+    meta_msg.meta = msg.meta;
+    meta_msg.meta.option |= kOptInPlace;
+    meta_msg.meta.val_len = static_cast<int64_t>(vals.size());
+    meta_msg.meta.data_type.clear();
+    // keep host-side keys/lens blobs for the merge bookkeeping
+    for (size_t i = 0; i < msg.data.size(); ++i) {
+      if (i == 1 || msg.data[i].on_device()) continue;
+      meta_msg.data.push_back(msg.data[i]);
+      meta_msg.meta.data_type.push_back(msg.meta.data_type[i]);
+    }
+    std::vector<char> by_ref(meta_msg.data.size(), 0);
+    std::string payload;
+    if (!Serialize(meta_msg, by_ref, &payload)) {
+      meta_msg.data.clear();
+      meta_msg.meta.data_type.clear();
+      XPS_CHECK(Serialize(meta_msg, {}, &payload));
+    }
+    int64_t bytes = static_cast<int64_t>(vals.size() + payload.size());
+    msg.data.clear();  // keepalive: vals (and the store view) live in meta_msg? no —
+    // the original msg owns vals; keep it alive until the event fires
+    Message keepalive;
+    keepalive.data.push_back(vals);
+    DeferSendInternal(p, peer_node.id, std::move(keepalive), std::move(payload), bytes);
+    return bytes;
+  }
+
+  // ---- general path: inline blobs + by-ref device blobs --------------
+  std::vector<char> by_ref(msg.data.size(), 0);
+  int64_t ref_bytes = 0;
+  for (size_t i = 0; i < msg.data.size(); ++i) {
+    if (msg.data[i].on_device()) {
+      by_ref[i] = 1;
+      ref_bytes += msg.data[i].size();
+      if (i == 1) msg.meta.option |= kOptValsByRef;
+    }
+  }
+  std::string payload;
+  if (!Serialize(msg, by_ref, &payload)) {
+    if (response) hipStreamSynchronize(StreamForPeer(peer_node.id));
+    return -1;
+  }
+  int64_t bytes = static_cast<int64_t>(payload.size()) + ref_bytes;
+  if (response) {
+    // order behind any handler kernels on this peer's stream
+    Message keepalive = msg;
+    DeferSendInternal(p, peer_node.id, std::move(keepalive), std::move(payload), bytes);
+  } else {
+    if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
+  }
+  return bytes;
+}
+
+void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::string payload,
+                                 int64_t bytes) {
+  hipStream_t stream = StreamForPeer(peer_id);
+  hipEvent_t ev = GetEvent();
+  XPS_HIP_CHECK(hipSetDevice(device_));
+  XPS_HIP_CHECK(hipEventRecord(ev, stream));
+  std::lock_guard<std::mutex> lk(pend_mu_);
+  pending_[peer_id].push_back(Pending{ev, peer_id, std::move(payload), std::move(keepalive), bytes});
+  pending_count_.fetch_add(1);
+}
+
+void GpuPlane::CompletionLoop() {
+  XPS_HIP_CHECK(hipSetDevice(device_));
+  int idle = 0;
+  while (!stop_.load()) {
+    bool did = false;
+    if (pending_count_.load() > 0) {
+      std::lock_guard<std::mutex> lk(pend_mu_);
+      for (auto& kv : pending_) {
+        auto& dq = kv.second;
+        while (!dq.empty()) {
+          Pending& front = dq.front();
+          hipError_t e = hipEventQuery(front.ev);
+          if (e == hipErrorNotReady) break;
+          XPS_CHECK(e == hipSuccess) << "hipEventQuery: " << hipGetErrorString(e);
+          Peer* peer = GetPeer(front.peer_id);
+          if (peer && EnsureRing(peer)) {
+            peer->ring.Push(front.payload.data(), static_cast<uint32_t>(front.payload.size()));
+          }
+          PutEvent(front.ev);
+          dq.pop_front();
+          pending_count_.fetch_sub(1);
+          did = true;
+        }
+      }
+    }
+    if (did) {
+      idle = 0;
+    } else if (++idle > 2000) {
+      usleep(20);
+    }
+  }
+}
+
+void GpuPlane::RingPollLoop() {
+  XPS_HIP_CHECK(hipSetDevice(device_));
+  std::vector<char> buf(ShmRing::MaxPayload());
+  int idle = 0;
+  while (!stop_.load()) {
+    uint32_t n = in_ring_.Pop(buf.data());
+    if (n == 0) {
+      if (++idle > 3000) usleep(20);
+      continue;
+    }
+    idle = 0;
+    ByteReader r(buf.data(), n);
+    uint64_t meta_len = r.U64();
+    Message msg;
+    {
+      std::vector<char> meta(meta_len);
+      r.Raw(meta.data(), meta_len);
+      UnpackMeta(meta.data(), meta_len, &msg.meta);
+    }
+    int nblobs = r.U8();
+    int64_t ref_bytes = 0;
+    bool ok = true;
+    for (int i = 0; i < nblobs; ++i) {
+      int kind = r.U8();
+      if (kind == 0) {
+        uint64_t len = r.U64();
+        SArray<char> d(len);
+        r.Raw(d.data(), len);
+        msg.data.push_back(d);
+      } else {
+        uint64_t off = r.U64();
+        uint64_t len = r.U64();
+        Peer* sender = GetPeer(msg.meta.sender);
+        void* base = sender ? PeerPoolBase(sender) : nullptr;
+        if (!base) {
+          XPS_LOG(Warning) << "dropping by-ref blob: sender pool not mapped (from "
+                           << msg.meta.sender << ")";
+          ok = false;
+          break;
+        }
+        msg.data.push_back(
+            SArray<char>(static_cast<char*>(base) + off, len, device_));
+        ref_bytes += len;
+      }
+    }
+    if (!ok) continue;
+    po_->van()->recv_bytes_ += n + ref_bytes;
+    po_->van()->Deliver(std::move(msg));
+  }
+}
 
 std::shared_ptr<DataPlane> CreateGpuPlane(Postoffice* po, int device) {
   if (device < 0) return nullptr;
-  return std::make_shared<StubPlane>(po, device);
+  return std::make_shared<GpuPlane>(po, device);
 }
 
 }  // namespace xps

@@ -5,16 +5,98 @@
 //
 // Reference parity: replaces ps-lite's RDMA data path — rendezvous +
 // one-sided RDMA_WRITE (src/rdma_van.h, rdma_transport.h) and the POSIX
-// shm IPCTransport (rdma_transport.h:591-617) — per SURVEY.md §5.8.
+// shm IPCTransport (rdma_transport.h:591-617) — per SURVEY.md §5.8:
+//  * per-(key,peer) buffer rendezvous -> a single pool hipIpc handle
+//    exchanged once at ADD_NODE (steady state ships {offset, len});
+//  * unsignaled RDMA_WRITE + write-with-imm -> async copy/kernel on the
+//    peer's stream + ring metadata sent after the completion event;
+//  * PollCQ busy-spin thread -> the event-poller (completion) thread;
+//  * IPCTransport shm segments -> the worker pool itself (zero copy).
 #pragma once
 
-#include <memory>
+#include <hip/hip_runtime.h>
 
+#include <deque>
+#include <memory>
+#include <thread>
+#include <unordered_map>
+
+#include "shm_ring.h"
 #include "van.h"
 
 namespace xps {
 
 class Postoffice;
+
+class GpuPlane : public DataPlane {
+ public:
+  GpuPlane(Postoffice* po, int device);
+  ~GpuPlane() override;
+
+  void FillSelf(Node* self) override;
+  bool CanSend(const Message& msg, const Node& peer) override;
+  int64_t Send(Message& msg, const Node& peer) override;
+  void OnPeer(const Node& peer) override;
+  void Stop() override;
+
+  // ---- used by the GPU server handlers -------------------------------
+  // the per-peer HIP stream every op destined for / received from `peer`
+  // runs on (gives the per-(key,peer) ordering guarantee)
+  hipStream_t StreamForPeer(int node_id);
+  int device() const { return device_; }
+
+ private:
+  struct Peer {
+    Node node;
+    ShmRing ring;  // producer handle on the peer's inbound ring
+    bool ring_tried = false;
+    void* pool_base = nullptr;  // peer pool mapped into our address space
+    bool pool_tried = false;
+    hipStream_t stream = nullptr;
+    std::mutex mu;
+  };
+
+  struct Pending {
+    hipEvent_t ev;
+    int peer_id;
+    std::string payload;
+    Message keepalive;  // holds pool temporaries until the event fires
+    int64_t bytes;
+  };
+
+  Peer* GetPeer(int id);
+  bool EnsureRing(Peer* p);
+  void* PeerPoolBase(Peer* p);
+  // serialize msg (meta + blobs, by-ref where flagged) into `out`;
+  // by_ref[i] true => blob i encoded as pool offset
+  bool Serialize(const Message& msg, const std::vector<char>& by_ref, std::string* out);
+  void RingPollLoop();
+  void CompletionLoop();
+  void DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::string payload,
+                         int64_t bytes);
+  hipEvent_t GetEvent();
+  void PutEvent(hipEvent_t ev);
+
+  Postoffice* po_;
+  int device_;
+  uint64_t my_host_hash_;
+  ShmRing in_ring_;
+  bool started_ = false;
+
+  std::mutex peers_mu_;
+  std::unordered_map<int, std::unique_ptr<Peer>> peers_;
+
+  std::thread poll_thread_;
+  std::thread comp_thread_;
+  std::atomic<bool> stop_{false};
+
+  std::mutex pend_mu_;
+  std::unordered_map<int, std::deque<Pending>> pending_;  // per peer, FIFO
+  std::atomic<int> pending_count_{0};
+
+  std::mutex ev_mu_;
+  std::vector<hipEvent_t> event_pool_;
+};
 
 // Returns the plane for this van, or nullptr when no GPU is attached.
 std::shared_ptr<DataPlane> CreateGpuPlane(Postoffice* po, int device);

@@ -1,0 +1,36 @@
+// Host-side launchers for the CDNA4 server kernels (kernels.hip).
+//
+// These implement the compute the ps-lite server handler API implies
+// (dense accumulate/assign replacing KVServerDefaultHandle
+// kv_app.h:441-448, sparse gather/scatter over a key-indexed table) —
+// all-new HIP code, the reference has no kernels.
+#pragma once
+
+#include <cstddef>
+#include <cstdint>
+
+#include <hip/hip_runtime.h>
+
+namespace xps {
+namespace kern {
+
+// dst[i] = src[i] (bytes; 16B-vectorized grid-stride copy kernel)
+void DenseAssign(void* dst, const void* src, size_t nbytes, hipStream_t s);
+// dst[i] += src[i] (fp32)
+void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s);
+// dst[i] += sum_j srcs[j][i], up to 8 sources in one pass (one read of
+// dst, one write — HBM-optimal multi-worker reduction)
+void DenseSumMultiF32(float* dst, const float* const* srcs_host, int nsrc, size_t n,
+                      hipStream_t s);
+// out[r][:] = table[rows[r]][:] for r in [0, nrows)
+void SparseGatherF32(const float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
+                     float* out, hipStream_t s);
+// table[rows[r]][:] += src[r][:]; atomic=true tolerates duplicate rows
+void SparseScatterAddF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
+                         const float* src, bool atomic, hipStream_t s);
+// table[rows[r]][:] = src[r][:]
+void SparseScatterAssignF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
+                            const float* src, hipStream_t s);
+
+}  // namespace kern
+}  // namespace xps

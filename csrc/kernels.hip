@@ -1,0 +1,227 @@
+// CDNA4 (gfx950) kernels for the KVServer request handlers.
+//
+// All kernels are HBM-bandwidth-bound streaming ops, so the design rules
+// are (cdna_hip_programming.md §2): 16 B per lane per instruction
+// (uint4/float4), 64-wide wavefronts, grid-stride loops sized >> 256
+// workgroups to fill 8 XCDs, and a single pass over dst for the
+// multi-source reduction. Inputs may be hipIpc-mapped peer-GPU memory —
+// the loads then ride xGMI.
+#include <hip/hip_runtime.h>
+
+#include "kernels.h"
+
+namespace xps {
+namespace kern {
+
+namespace {
+
+constexpr int kBlock = 256;
+
+inline int GridFor(size_t work_items, int cap = 8192) {
+  size_t g = (work_items + kBlock - 1) / kBlock;
+  if (g > static_cast<size_t>(cap)) g = cap;
+  if (g == 0) g = 1;
+  return static_cast<int>(g);
+}
+
+__global__ void assign_kernel(uint4* __restrict__ dst, const uint4* __restrict__ src, size_t n4) {
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (; i < n4; i += stride) dst[i] = src[i];
+}
+
+__global__ void assign_tail_kernel(char* __restrict__ dst, const char* __restrict__ src,
+                                   size_t begin, size_t end) {
+  size_t i = begin + blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  if (i < end) dst[i] = src[i];
+}
+
+__global__ void sum_kernel_f32(float4* __restrict__ dst, const float4* __restrict__ src,
+                               size_t n4) {
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 d = dst[i];
+    float4 s = src[i];
+    d.x += s.x;
+    d.y += s.y;
+    d.z += s.z;
+    d.w += s.w;
+    dst[i] = d;
+  }
+}
+
+__global__ void sum_tail_f32(float* __restrict__ dst, const float* __restrict__ src, size_t begin,
+                             size_t end) {
+  size_t i = begin + blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  if (i < end) dst[i] += src[i];
+}
+
+struct SrcList {
+  const float4* p[8];
+};
+
+__global__ void sum_multi_f32(float4* __restrict__ dst, SrcList srcs, int nsrc, size_t n4) {
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 d = dst[i];
+    for (int j = 0; j < nsrc; ++j) {
+      float4 s = srcs.p[j][i];
+      d.x += s.x;
+      d.y += s.y;
+      d.z += s.z;
+      d.w += s.w;
+    }
+    dst[i] = d;
+  }
+}
+
+// one wave per row-chunk: rows are contiguous float runs; row_len4 is the
+// row length in float4. Each block strides over rows; lanes stride the row.
+__global__ void gather_rows_f32(const float4* __restrict__ table, const uint64_t* __restrict__ rows,
+                                size_t nrows, size_t row_len4, float4* __restrict__ out) {
+  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
+    const float4* src = table + rows[r] * row_len4;
+    float4* dst = out + r * row_len4;
+    for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) dst[c] = src[c];
+  }
+}
+
+__global__ void scatter_add_rows_f32(float4* __restrict__ table, const uint64_t* __restrict__ rows,
+                                     size_t nrows, size_t row_len4,
+                                     const float4* __restrict__ src) {
+  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
+    float4* dst = table + rows[r] * row_len4;
+    const float4* s = src + r * row_len4;
+    for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) {
+      float4 d = dst[c];
+      float4 v = s[c];
+      d.x += v.x;
+      d.y += v.y;
+      d.z += v.z;
+      d.w += v.w;
+      dst[c] = d;
+    }
+  }
+}
+
+__global__ void gather_rows_scalar_f32(const float* __restrict__ table,
+                                       const uint64_t* __restrict__ rows, size_t nrows,
+                                       size_t row_len, float* __restrict__ out) {
+  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
+    const float* src = table + rows[r] * row_len;
+    float* dst = out + r * row_len;
+    for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) dst[c] = src[c];
+  }
+}
+
+__global__ void scatter_assign_rows_f32(float4* __restrict__ table,
+                                        const uint64_t* __restrict__ rows, size_t nrows,
+                                        size_t row_len4, const float4* __restrict__ src) {
+  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
+    float4* dst = table + rows[r] * row_len4;
+    const float4* s = src + r * row_len4;
+    for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) dst[c] = s[c];
+  }
+}
+
+__global__ void scatter_assign_rows_scalar_f32(float* __restrict__ table,
+                                               const uint64_t* __restrict__ rows, size_t nrows,
+                                               size_t row_len, const float* __restrict__ src) {
+  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
+    float* dst = table + rows[r] * row_len;
+    const float* s = src + r * row_len;
+    for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) dst[c] = s[c];
+  }
+}
+
+__global__ void scatter_add_rows_atomic_f32(float* __restrict__ table,
+                                            const uint64_t* __restrict__ rows, size_t nrows,
+                                            size_t row_len, const float* __restrict__ src) {
+  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
+    float* dst = table + rows[r] * row_len;
+    const float* s = src + r * row_len;
+    for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) {
+      atomicAdd(&dst[c], s[c]);
+    }
+  }
+}
+
+}  // namespace
+
+void DenseAssign(void* dst, const void* src, size_t nbytes, hipStream_t s) {
+  size_t n4 = nbytes / 16;
+  if (n4) {
+    hipLaunchKernelGGL(assign_kernel, dim3(GridFor(n4)), dim3(kBlock), 0, s,
+                       static_cast<uint4*>(dst), static_cast<const uint4*>(src), n4);
+  }
+  if (nbytes % 16) {
+    hipLaunchKernelGGL(assign_tail_kernel, dim3(1), dim3(kBlock), 0, s, static_cast<char*>(dst),
+                       static_cast<const char*>(src), n4 * 16, nbytes);
+  }
+}
+
+void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s) {
+  size_t n4 = n / 4;
+  if (n4) {
+    hipLaunchKernelGGL(sum_kernel_f32, dim3(GridFor(n4)), dim3(kBlock), 0, s,
+                       reinterpret_cast<float4*>(dst), reinterpret_cast<const float4*>(src), n4);
+  }
+  if (n % 4) {
+    hipLaunchKernelGGL(sum_tail_f32, dim3(1), dim3(kBlock), 0, s, dst, src, n4 * 4, n);
+  }
+}
+
+void DenseSumMultiF32(float* dst, const float* const* srcs_host, int nsrc, size_t n,
+                      hipStream_t s) {
+  SrcList list{};
+  for (int i = 0; i < nsrc && i < 8; ++i) list.p[i] = reinterpret_cast<const float4*>(srcs_host[i]);
+  size_t n4 = n / 4;
+  hipLaunchKernelGGL(sum_multi_f32, dim3(GridFor(n4)), dim3(kBlock), 0, s,
+                     reinterpret_cast<float4*>(dst), list, nsrc, n4);
+  for (int j = 0; j < nsrc; ++j) {
+    if (n % 4) {
+      hipLaunchKernelGGL(sum_tail_f32, dim3(1), dim3(kBlock), 0, s, dst, srcs_host[j], n4 * 4, n);
+    }
+  }
+}
+
+void SparseGatherF32(const float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
+                     float* out, hipStream_t s) {
+  if (row_len % 4 == 0) {
+    hipLaunchKernelGGL(gather_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)), dim3(kBlock),
+                       0, s, reinterpret_cast<const float4*>(table), rows_dev, nrows, row_len / 4,
+                       reinterpret_cast<float4*>(out));
+  } else {
+    hipLaunchKernelGGL(gather_rows_scalar_f32, dim3(GridFor(nrows * row_len, 16384)), dim3(kBlock),
+                       0, s, table, rows_dev, nrows, row_len, out);
+  }
+}
+
+void SparseScatterAssignF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
+                            const float* src, hipStream_t s) {
+  if (row_len % 4 == 0) {
+    hipLaunchKernelGGL(scatter_assign_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)),
+                       dim3(kBlock), 0, s, reinterpret_cast<float4*>(table), rows_dev, nrows,
+                       row_len / 4, reinterpret_cast<const float4*>(src));
+  } else {
+    hipLaunchKernelGGL(scatter_assign_rows_scalar_f32, dim3(GridFor(nrows * row_len, 16384)),
+                       dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src);
+  }
+}
+
+void SparseScatterAddF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
+                         const float* src, bool atomic, hipStream_t s) {
+  if (atomic || row_len % 4 != 0) {
+    hipLaunchKernelGGL(scatter_add_rows_atomic_f32, dim3(GridFor(nrows * row_len, 16384)),
+                       dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src);
+  } else {
+    hipLaunchKernelGGL(scatter_add_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)),
+                       dim3(kBlock), 0, s, reinterpret_cast<float4*>(table), rows_dev, nrows,
+                       row_len / 4, reinterpret_cast<const float4*>(src));
+  }
+}
+
+}  // namespace kern
+}  // namespace xps

@@ -15,18 +15,20 @@
 #include <vector>
 
 #include "hip_pool.h"
+#include "hip_util.h"
 #include "simple_app.h"
 
 namespace xps {
 
-// meta.option bits
-static const int kOptInPlace = 1;  // pull response already written into dst buffer
-
 template <typename V>
 struct KVPairs {
-  SArray<Key> keys;
+  SArray<Key> keys;      // host keys (slicing always runs host-side)
   SArray<V> vals;
   SArray<int> lens;
+  // optional device mirror of `keys` (sparse path): when set, the wire
+  // message carries this blob by pool reference so server kernels can
+  // index the table without a host->device staging copy
+  SArray<Key> keys_dev;
 };
 
 struct KVMeta {
@@ -74,20 +76,21 @@ class KVWorker : public SimpleApp {
 
   // zero-copy push: keys/vals/lens remain owned by the caller until done
   int ZPush(const SArray<Key>& keys, const SArray<V>& vals, const SArray<int>& lens = {},
-            int cmd = 0, const Callback& cb = nullptr) {
+            int cmd = 0, const Callback& cb = nullptr, const SArray<Key>& keys_dev = {}) {
     int ts = obj_->NewRequest(kServerGroup);
     AddCallback(ts, cb);
     KVPairs<V> kvs;
     kvs.keys = keys;
     kvs.vals = vals;
     kvs.lens = lens;
+    kvs.keys_dev = keys_dev;
     Send(ts, true, false, cmd, kvs);
     return ts;
   }
 
   // zero-copy pull into a pre-allocated vals buffer (device or host)
   int ZPull(const SArray<Key>& keys, SArray<V>* vals, SArray<int>* lens = nullptr, int cmd = 0,
-            const Callback& cb = nullptr) {
+            const Callback& cb = nullptr, const SArray<Key>& keys_dev = {}) {
     XPS_CHECK(vals && vals->size()) << "ZPull needs a pre-sized vals buffer";
     int ts = obj_->NewRequest(kServerGroup);
     {
@@ -99,6 +102,7 @@ class KVWorker : public SimpleApp {
     kvs.keys = keys;
     kvs.vals = *vals;  // carried for slicing geometry; not sent in requests
     if (lens) kvs.lens = *lens;
+    kvs.keys_dev = keys_dev;
     Send(ts, false, true, cmd, kvs);
     return ts;
   }
@@ -134,6 +138,9 @@ class KVWorker : public SimpleApp {
       out.first = b > a;
       if (!out.first) continue;
       out.second.keys = send.keys.Segment(a, b);
+      if (!send.keys_dev.empty()) {
+        out.second.keys_dev = send.keys_dev.Segment(a, b);
+      }
       if (!send.vals.empty()) {
         out.second.vals = send.vals.Segment(val_off[a], val_off[b]);
       }
@@ -197,10 +204,14 @@ class KVWorker : public SimpleApp {
       msg.meta.head = cmd;
       msg.meta.timestamp = ts;
       msg.meta.recver = ServerRankToID(static_cast<int>(i));
-      if (s.keys.size() == 1) msg.meta.key = s.keys[0];
+      if (s.keys.size() >= 1) msg.meta.key = s.keys[0];
       msg.meta.src_dev = push ? s.vals.device() : kvs.vals.device();
       msg.meta.dst_dev = msg.meta.src_dev;
-      msg.AddData(s.keys);
+      if (!s.keys_dev.empty()) {
+        msg.AddData(s.keys_dev);
+      } else {
+        msg.AddData(s.keys);
+      }
       if (push) {
         msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
         msg.AddData(s.vals);
@@ -212,6 +223,7 @@ class KVWorker : public SimpleApp {
           uint64_t off = 0;
           if (HbmPool::Get()->OffsetOf(s.vals.data(), &off)) {
             msg.meta.addr = off;
+            msg.meta.option |= kOptPullAddr;  // server may write the response in place
           }
         }
         if (!s.lens.empty()) msg.AddData(s.lens);
@@ -231,7 +243,9 @@ class KVWorker : public SimpleApp {
         kvs.keys = SArray<Key>::View(msg.data[0]);
         if (msg.data.size() > 1) kvs.vals = SArray<V>::View(msg.data[1]);
         if (msg.data.size() > 2) kvs.lens = SArray<int>::View(msg.data[2]);
-      } else {
+      }
+      if (kvs.keys.empty() || kvs.keys.on_device()) {
+        // device keys (sparse path) are never dereferenced host-side
         kvs.keys = SArray<Key>({msg.meta.key});
       }
     }
@@ -239,7 +253,7 @@ class KVWorker : public SimpleApp {
       std::lock_guard<std::mutex> lk(mu_);
       if (msg.meta.pull) {
         auto& got = recv_kvs_[ts];
-        got.emplace_back(kvs, msg.meta.option);
+        got.push_back(RecvSlice{kvs, msg.meta.option, msg.meta.val_len});
         last = static_cast<int>(got.size()) >= expected_[ts];
       } else {
         int n = ++push_acks_[ts];
@@ -253,7 +267,7 @@ class KVWorker : public SimpleApp {
   }
 
   void MergePull(int ts) {
-    std::vector<std::pair<KVPairs<V>, int>> got;
+    std::vector<RecvSlice> got;
     SArray<V>* dst_vals = nullptr;
     SArray<int>* dst_lens = nullptr;
     std::vector<V>* vec_vals = nullptr;
@@ -276,9 +290,9 @@ class KVWorker : public SimpleApp {
       }
     }
     // order slices by their first key
-    std::sort(got.begin(), got.end(), [](const auto& a, const auto& b) {
-      Key ka = a.first.keys.empty() ? 0 : a.first.keys[0];
-      Key kb = b.first.keys.empty() ? 0 : b.first.keys[0];
+    std::sort(got.begin(), got.end(), [](const RecvSlice& a, const RecvSlice& b) {
+      Key ka = a.kvs.keys.empty() ? 0 : a.kvs.keys[0];
+      Key kb = b.kvs.keys.empty() ? 0 : b.kvs.keys[0];
       return ka < kb;
     });
     if (vec_vals) {
@@ -286,9 +300,9 @@ class KVWorker : public SimpleApp {
       vec_vals->clear();
       if (vec_lens) vec_lens->clear();
       for (auto& g : got) {
-        vec_vals->insert(vec_vals->end(), g.first.vals.begin(), g.first.vals.end());
+        vec_vals->insert(vec_vals->end(), g.kvs.vals.begin(), g.kvs.vals.end());
         if (vec_lens) {
-          for (size_t i = 0; i < g.first.lens.size(); ++i) vec_lens->push_back(g.first.lens[i]);
+          for (size_t i = 0; i < g.kvs.lens.size(); ++i) vec_lens->push_back(g.kvs.lens[i]);
         }
       }
       return;
@@ -296,27 +310,27 @@ class KVWorker : public SimpleApp {
     if (!dst_vals) return;
     size_t off = 0;
     for (auto& g : got) {
-      if (g.second & kOptInPlace) {
-        // data plane already wrote into the destination buffer
-        off += g.first.vals.size();
+      if (g.option & kOptInPlace) {
+        // the data plane already wrote this slice into the destination
+        off += static_cast<size_t>(g.val_len) / sizeof(V);
         continue;
       }
-      size_t n = g.first.vals.size();
+      size_t n = g.kvs.vals.size();
       if (n == 0) continue;
       XPS_CHECK_LE(off + n, dst_vals->size()) << "pull response overflows dst";
       if (dst_vals->on_device()) {
-        gpu::CopyHostToDevice(dst_vals->data() + off, g.first.vals.data(), n * sizeof(V),
+        gpu::CopyHostToDevice(dst_vals->data() + off, g.kvs.vals.data(), n * sizeof(V),
                               dst_vals->device());
       } else {
-        memcpy(dst_vals->data() + off, g.first.vals.data(), n * sizeof(V));
+        memcpy(dst_vals->data() + off, g.kvs.vals.data(), n * sizeof(V));
       }
       off += n;
     }
     if (dst_lens && !got.empty()) {
       size_t loff = 0;
       for (auto& g : got) {
-        for (size_t i = 0; i < g.first.lens.size() && loff < dst_lens->size(); ++i) {
-          (*dst_lens)[loff++] = g.first.lens[i];
+        for (size_t i = 0; i < g.kvs.lens.size() && loff < dst_lens->size(); ++i) {
+          (*dst_lens)[loff++] = g.kvs.lens[i];
         }
       }
     }
@@ -337,12 +351,18 @@ class KVWorker : public SimpleApp {
     if (cb) cb();
   }
 
+  struct RecvSlice {
+    KVPairs<V> kvs;
+    int option = 0;
+    int64_t val_len = 0;
+  };
+
   Slicer slicer_;
   std::mutex mu_;
   std::unordered_map<int, Callback> callbacks_;
   std::unordered_map<int, int> expected_;
   std::unordered_map<int, int> push_acks_;
-  std::unordered_map<int, std::vector<std::pair<KVPairs<V>, int>>> recv_kvs_;
+  std::unordered_map<int, std::vector<RecvSlice>> recv_kvs_;
   std::unordered_map<int, std::pair<SArray<V>*, SArray<int>*>> pull_dst_;
   std::unordered_map<int, std::pair<std::vector<V>*, std::vector<int>*>> vec_pull_dst_;
 };

@@ -65,6 +65,11 @@ struct Control {
   }
 };
 
+// meta.option bits (shared by kv_app and the data plane)
+static const int kOptInPlace = 1;   // pull response already written into dst buffer
+static const int kOptValsByRef = 2; // vals travel as {pool offset, len}, not bytes
+static const int kOptPullAddr = 4;  // meta.addr holds a valid pull-destination offset
+
 enum DataType : int { kChar = 0, kInt32, kInt64, kUint64, kFloat32, kFloat64, kUint8 };
 inline size_t DataTypeSize(int t) {
   switch (t) {

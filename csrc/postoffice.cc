@@ -96,8 +96,10 @@ Customer* Postoffice::GetCustomer(int app_id, int customer_id, int timeout_sec) 
 
 void Postoffice::Barrier(int customer_id, int group) {
   XPS_CHECK(van_->IsReady());
-  std::unique_lock<std::mutex> lk(barrier_mu_);
-  barrier_done_ = false;
+  {
+    std::lock_guard<std::mutex> lk(barrier_mu_);
+    barrier_done_ = false;
+  }
   Message req;
   req.meta.control.cmd = Control::BARRIER;
   req.meta.control.barrier_group = group;
@@ -105,7 +107,11 @@ void Postoffice::Barrier(int customer_id, int group) {
   req.meta.recver = kScheduler;
   req.meta.app_id = 0;
   req.meta.customer_id = customer_id;
+  // NOTE: the lock must NOT be held across Send — on the scheduler the
+  // loopback delivery can complete the barrier inline on this thread and
+  // re-enter Manage (which takes barrier_mu_)
   van_->Send(req);
+  std::unique_lock<std::mutex> lk(barrier_mu_);
   barrier_cv_.wait(lk, [this] { return barrier_done_; });
 }
 

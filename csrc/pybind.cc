@@ -7,8 +7,10 @@
 #include "gpu_plane.h"
 #include "hip_pool.h"
 #include "hip_util.h"
+#include "kernels.h"
 #include "kv_app.h"
 #include "ps.h"
+#include "server_handlers.h"
 #include "simple_app.h"
 #include "van.h"
 
@@ -67,18 +69,22 @@ class PyKVWorker {
   }
 
   int ZPushPtr(py::array_t<uint64_t> keys, uintptr_t vals_ptr, size_t vals_bytes, int device,
-               py::array_t<int> lens) {
+               py::array_t<int> lens, int cmd, uintptr_t keys_dev_ptr) {
     SArray<Key> k;
     k.CopyFrom(keys.data(), keys.size());
     SArray<float> v(reinterpret_cast<float*>(vals_ptr), vals_bytes / sizeof(float), device);
     SArray<int> l;
     if (lens.size()) l.CopyFrom(lens.data(), lens.size());
+    SArray<Key> kd;
+    if (keys_dev_ptr) {
+      kd = SArray<Key>(reinterpret_cast<Key*>(keys_dev_ptr), keys.size(), device);
+    }
     py::gil_scoped_release rel;
-    return w_.ZPush(k, v, l);
+    return w_.ZPush(k, v, l, cmd, nullptr, kd);
   }
 
   int ZPullPtr(py::array_t<uint64_t> keys, uintptr_t vals_ptr, size_t vals_bytes, int device,
-               py::array_t<int> lens) {
+               py::array_t<int> lens, int cmd, uintptr_t keys_dev_ptr) {
     SArray<Key> k;
     k.CopyFrom(keys.data(), keys.size());
     auto* v = new SArray<float>(reinterpret_cast<float*>(vals_ptr), vals_bytes / sizeof(float),
@@ -88,12 +94,16 @@ class PyKVWorker {
       l = new SArray<int>();
       l->CopyFrom(lens.data(), lens.size());
     }
+    SArray<Key> kd;
+    if (keys_dev_ptr) {
+      kd = SArray<Key>(reinterpret_cast<Key*>(keys_dev_ptr), keys.size(), device);
+    }
     py::gil_scoped_release rel;
     // delete the temporaries when the pull completes
-    return w_.ZPull(k, v, l, 0, [v, l]() {
+    return w_.ZPull(k, v, l, cmd, [v, l]() {
       delete v;
       delete l;
-    });
+    }, kd);
   }
 
   void Wait(int ts) {
@@ -115,6 +125,23 @@ class PyKVServer {
       (*h)(m, kvs, srv);
     });
   }
+
+  void SetGpuDenseHandle(bool accumulate) {
+    auto h = std::make_shared<GpuDenseHandler>(s_.postoffice(), accumulate);
+    s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
+      (*h)(m, kvs, srv);
+    });
+  }
+
+  void SetGpuSparseHandle(size_t rows, size_t row_len, bool accumulate) {
+    auto h = std::make_shared<GpuSparseHandler>(s_.postoffice(), rows, row_len, accumulate);
+    sparse_ = h;
+    s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
+      (*h)(m, kvs, srv);
+    });
+  }
+
+  uintptr_t SparseTablePtr() const { return sparse_ ? sparse_->table_ptr() : 0; }
 
   // fn(meta: dict, keys: ndarray[u64], vals: ndarray[f32]) -> ndarray[f32] | None
   void SetPythonHandle(py::function fn) {
@@ -155,6 +182,7 @@ class PyKVServer {
 
  private:
   KVServer<float> s_;
+  std::shared_ptr<GpuSparseHandler> sparse_;
 };
 
 class PySimpleApp {
@@ -263,15 +291,46 @@ PYBIND11_MODULE(_core, m) {
            py::arg("lens") = py::array_t<int>())
       .def("pull", &PyKVWorker::PullBlocking, py::arg("keys"))
       .def("zpush_ptr", &PyKVWorker::ZPushPtr, py::arg("keys"), py::arg("vals_ptr"),
-           py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>())
+           py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>(),
+           py::arg("cmd") = 0, py::arg("keys_dev_ptr") = 0)
       .def("zpull_ptr", &PyKVWorker::ZPullPtr, py::arg("keys"), py::arg("vals_ptr"),
-           py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>())
+           py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>(),
+           py::arg("cmd") = 0, py::arg("keys_dev_ptr") = 0)
       .def("wait", &PyKVWorker::Wait);
 
   py::class_<PyKVServer>(m, "KVServer")
       .def(py::init<int>(), py::arg("app_id") = 0)
       .def("set_default_handle", &PyKVServer::SetDefaultHandle)
+      .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("accumulate") = false)
+      .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),
+           py::arg("row_len"), py::arg("accumulate") = true)
+      .def("sparse_table_ptr", &PyKVServer::SparseTablePtr)
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
+
+  // raw kernel entry points (synchronous; for numerics tests)
+  m.def("k_dense_assign", [](uintptr_t dst, uintptr_t src, size_t nbytes) {
+    kern::DenseAssign(reinterpret_cast<void*>(dst), reinterpret_cast<void*>(src), nbytes, nullptr);
+    gpu::DeviceSync(-1);
+  });
+  m.def("k_dense_sum_f32", [](uintptr_t dst, uintptr_t src, size_t n) {
+    kern::DenseSumF32(reinterpret_cast<float*>(dst), reinterpret_cast<float*>(src), n, nullptr);
+    gpu::DeviceSync(-1);
+  });
+  m.def("k_sparse_gather_f32",
+        [](uintptr_t table, uintptr_t rows, size_t nrows, size_t row_len, uintptr_t out) {
+          kern::SparseGatherF32(reinterpret_cast<float*>(table),
+                                reinterpret_cast<uint64_t*>(rows), nrows, row_len,
+                                reinterpret_cast<float*>(out), nullptr);
+          gpu::DeviceSync(-1);
+        });
+  m.def("k_sparse_scatter_add_f32",
+        [](uintptr_t table, uintptr_t rows, size_t nrows, size_t row_len, uintptr_t src,
+           bool atomic) {
+          kern::SparseScatterAddF32(reinterpret_cast<float*>(table),
+                                    reinterpret_cast<uint64_t*>(rows), nrows, row_len,
+                                    reinterpret_cast<float*>(src), atomic, nullptr);
+          gpu::DeviceSync(-1);
+        });
 
   py::class_<PySimpleApp>(m, "SimpleApp")
       .def(py::init<const std::string&, int, int>(), py::arg("role"), py::arg("app_id") = 10,

@@ -1,0 +1,67 @@
+// Built-in GPU request handlers for KVServer<float>.
+//
+// Reference parity: ps-lite ships only CPU handlers — the tests'
+// EmptyHandler (tests/test_benchmark.cc:132-203, stores pushed buffers,
+// echoes on pull) and KVServerDefaultHandle (kv_app.h:431-452,
+// store[key] += vals). Here both behaviors are CDNA4 kernels on the
+// requester's per-peer HIP stream: push = DenseAssign / DenseSumF32
+// reading the worker's HBM directly (hipIpc-mapped, reads ride xGMI);
+// pull = zero-copy store view handed to the plane's in-place xGMI write.
+// The sparse handler owns a rows x row_len fp32 embedding table and uses
+// the gather/scatter kernels (BASELINE config #5).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <unordered_map>
+
+#include "kv_app.h"
+
+namespace xps {
+
+// worker-side cmd values (meta.head)
+static const int kCmdDefault = 0;  // handler's configured default op
+static const int kCmdAssign = 1;
+static const int kCmdSum = 2;
+
+class GpuDenseHandler {
+ public:
+  GpuDenseHandler(Postoffice* po, bool default_sum);
+  void operator()(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
+
+ private:
+  void HandlePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
+  void HandlePull(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
+  hipStream_t Stream(int sender);
+
+  Postoffice* po_;
+  bool default_sum_;
+  std::mutex mu_;
+  std::unordered_map<Key, SArray<char>> store_;
+  hipStream_t fallback_stream_ = nullptr;
+};
+
+class GpuSparseHandler {
+ public:
+  // allocates (and zeroes) a rows x row_len fp32 table in the pool
+  GpuSparseHandler(Postoffice* po, size_t rows, size_t row_len, bool accumulate);
+  void operator()(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
+  uintptr_t table_ptr() const { return reinterpret_cast<uintptr_t>(table_.data()); }
+
+ private:
+  hipStream_t Stream(int sender);
+  // keys usable by a kernel: device keys pass through; host keys are
+  // staged synchronously into a per-peer scratch row-id buffer
+  const uint64_t* DeviceKeys(const SArray<Key>& keys, int sender, hipStream_t s);
+
+  Postoffice* po_;
+  size_t rows_;
+  size_t row_len_;
+  bool accumulate_;
+  SArray<char> table_;
+  std::mutex mu_;
+  std::unordered_map<int, SArray<char>> key_scratch_;  // per sender
+  hipStream_t fallback_stream_ = nullptr;
+};
+
+}  // namespace xps

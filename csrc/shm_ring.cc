@@ -1,0 +1,139 @@
+#include "shm_ring.h"
+
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#include <cstdio>
+#include <cstring>
+#include <thread>
+
+#include "base.h"
+
+namespace xps {
+
+namespace {
+
+struct RingHeader {
+  alignas(64) std::atomic<uint64_t> head;  // producer ticket
+  alignas(64) std::atomic<uint64_t> tail;  // consumer position
+  alignas(64) uint64_t magic;
+};
+static const uint64_t kRingMagic = 0x587052696e673166ull;  // "XpRing1f"
+
+}  // namespace
+
+struct RingSlot {
+  std::atomic<uint64_t> seq;
+  uint32_t len;
+  uint32_t pad;
+  char payload[ShmRing::kSlotBytes - 16];
+};
+
+static RingHeader* Hdr(void* mem) { return static_cast<RingHeader*>(mem); }
+static RingSlot* Slot(void* mem, uint64_t i) {
+  return reinterpret_cast<RingSlot*>(static_cast<char*>(mem) + 4096 +
+                                     (i % ShmRing::kSlots) * sizeof(RingSlot));
+}
+
+static std::string RingName(uint64_t uid) {
+  char buf[64];
+  snprintf(buf, sizeof(buf), "/xps_ring_%016llx", static_cast<unsigned long long>(uid));
+  return buf;
+}
+
+ShmRing::~ShmRing() {
+  if (mem_) munmap(mem_, bytes_);
+  mem_ = nullptr;
+}
+
+bool ShmRing::Create(uint64_t uid) {
+  name_ = RingName(uid);
+  bytes_ = 4096 + static_cast<size_t>(kSlots) * sizeof(RingSlot);
+  shm_unlink(name_.c_str());  // stale segment from a crashed run
+  int fd = shm_open(name_.c_str(), O_CREAT | O_EXCL | O_RDWR, 0600);
+  if (fd < 0) {
+    XPS_LOG(Warning) << "shm_open create failed for " << name_;
+    return false;
+  }
+  if (ftruncate(fd, bytes_) != 0) {
+    close(fd);
+    return false;
+  }
+  mem_ = mmap(nullptr, bytes_, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
+  close(fd);
+  if (mem_ == MAP_FAILED) {
+    mem_ = nullptr;
+    return false;
+  }
+  memset(mem_, 0, bytes_);
+  for (uint64_t i = 0; i < kSlots; ++i) Slot(mem_, i)->seq.store(i, std::memory_order_relaxed);
+  Hdr(mem_)->head.store(0);
+  Hdr(mem_)->tail.store(0);
+  __atomic_store_n(&Hdr(mem_)->magic, kRingMagic, __ATOMIC_RELEASE);
+  owner_ = true;
+  return true;
+}
+
+bool ShmRing::Open(uint64_t uid) {
+  name_ = RingName(uid);
+  bytes_ = 4096 + static_cast<size_t>(kSlots) * sizeof(RingSlot);
+  int fd = -1;
+  for (int i = 0; i < 200; ++i) {  // the owner may still be creating it
+    fd = shm_open(name_.c_str(), O_RDWR, 0600);
+    if (fd >= 0) break;
+    usleep(10 * 1000);
+  }
+  if (fd < 0) return false;
+  mem_ = mmap(nullptr, bytes_, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
+  close(fd);
+  if (mem_ == MAP_FAILED) {
+    mem_ = nullptr;
+    return false;
+  }
+  for (int i = 0; i < 200; ++i) {
+    if (__atomic_load_n(&Hdr(mem_)->magic, __ATOMIC_ACQUIRE) == kRingMagic) return true;
+    usleep(10 * 1000);
+  }
+  munmap(mem_, bytes_);
+  mem_ = nullptr;
+  return false;
+}
+
+void ShmRing::CloseAndUnlink() {
+  if (mem_) {
+    munmap(mem_, bytes_);
+    mem_ = nullptr;
+  }
+  if (owner_ && !name_.empty()) shm_unlink(name_.c_str());
+}
+
+bool ShmRing::Push(const void* payload, uint32_t len) {
+  if (len > MaxPayload() || !mem_) return false;
+  uint64_t pos = Hdr(mem_)->head.fetch_add(1, std::memory_order_relaxed);
+  RingSlot* s = Slot(mem_, pos);
+  // wait for the consumer to free this slot (seq == pos)
+  int spins = 0;
+  while (s->seq.load(std::memory_order_acquire) != pos) {
+    if (++spins > 1000) std::this_thread::yield();
+  }
+  memcpy(s->payload, payload, len);
+  s->len = len;
+  s->seq.store(pos + 1, std::memory_order_release);
+  return true;
+}
+
+uint32_t ShmRing::Pop(void* buf) {
+  if (!mem_) return 0;
+  uint64_t pos = Hdr(mem_)->tail.load(std::memory_order_relaxed);
+  RingSlot* s = Slot(mem_, pos);
+  if (s->seq.load(std::memory_order_acquire) != pos + 1) return 0;  // empty
+  uint32_t len = s->len;
+  memcpy(buf, s->payload, len);
+  s->seq.store(pos + kSlots, std::memory_order_release);  // free for lap head+kSlots
+  Hdr(mem_)->tail.store(pos + 1, std::memory_order_relaxed);
+  return len;
+}
+
+}  // namespace xps

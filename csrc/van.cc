@@ -445,6 +445,10 @@ void Van::ProcessBarrierAtScheduler(Message& msg) {
     if (waiters.size() < expected) return;
     release.swap(waiters);
   }
+  // release self LAST: waking our own Finalize first would let Van::Stop
+  // close the connections before the other waiters get their responses
+  std::stable_partition(release.begin(), release.end(),
+                        [this](int id) { return id != my_node_.id; });
   Message res;
   res.meta.control.cmd = Control::BARRIER;
   res.meta.request = false;
