@@ -265,6 +265,10 @@ PYBIND11_MODULE(_core, m) {
     return van ? van->recv_bytes_.load() : int64_t(0);
   });
   m.def("gpu_count", []() { return gpu::DeviceCount(); });
+  m.def("device_sync", [](int dev) {
+    py::gil_scoped_release rel;
+    gpu::DeviceSync(dev);
+  }, py::arg("dev") = -1);
 
   m.attr("SCHEDULER_GROUP") = kScheduler;
   m.attr("SERVER_GROUP") = kServerGroup;

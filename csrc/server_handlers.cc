@@ -57,7 +57,12 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
     {
       std::lock_guard<std::mutex> lk(mu_);
       auto& e = store_[kvs.keys[i]];
-      if (e.size() < len) e = HbmPool::Get()->AllocArray(len);
+      if (e.size() < len) {
+        e = HbmPool::Get()->AllocArray(len);
+        // zero before publishing: another peer's stream may accumulate
+        // into this entry concurrently with our first push
+        XPS_HIP_CHECK(hipMemset(e.data(), 0, len));
+      }
       entry = e;
     }
     const char* src = reinterpret_cast<const char*>(kvs.vals.data()) + off;

@@ -1,0 +1,179 @@
+"""MI355X tests: HBM pool, CDNA4 kernel numerics (vs torch fp32), and the
+hipIpc/shm data plane end to end. All marked gpu."""
+import threading
+
+import numpy as np
+import pytest
+
+import ps_lite_amd as ps
+from ps_lite_amd.parallel import launch_local
+
+pytestmark = pytest.mark.gpu
+
+_PORT = [27100]
+
+
+def _boot_joint_inproc():
+    _PORT[0] += 7
+    ps.setup_env(1, 1, root_port=_PORT[0], XPS_DEV_ID=0, XPS_POOL_GB=4)
+    ths = [threading.Thread(target=ps.start, kwargs=dict(role="scheduler", device=-1)),
+           threading.Thread(target=ps.start, kwargs=dict(role="joint", device=0))]
+    [t.start() for t in ths]
+    [t.join() for t in ths]
+
+
+def _down_joint():
+    ths = [threading.Thread(target=ps.finalize, kwargs=dict(role="scheduler")),
+           threading.Thread(target=ps.finalize, kwargs=dict(role="joint"))]
+    [t.start() for t in ths]
+    [t.join() for t in ths]
+    ps.clear_registry()
+
+
+def test_pool_roundtrip():
+    ps.pool_init(0)
+    buf = ps.pool_alloc(1 << 20)
+    src = np.random.default_rng(0).standard_normal((1 << 20) // 4).astype(np.float32)
+    buf.copy_from(src)
+    out = buf.to_numpy_f32()
+    assert np.allclose(out, src)
+
+
+def test_kernel_dense_sum_vs_torch():
+    import torch
+
+    torch.manual_seed(0)
+    n = 1 << 20
+    a = torch.randn(n, device="cuda:0")
+    b = torch.randn(n, device="cuda:0")
+    ref = (a + b).cpu()
+    ps._core.k_dense_sum_f32(a.data_ptr(), b.data_ptr(), n)
+    assert torch.allclose(a.cpu(), ref)
+
+
+def test_kernel_dense_assign_vs_torch():
+    import torch
+
+    n = (1 << 20) + 3  # exercise the tail path
+    a = torch.zeros(n, device="cuda:0")
+    b = torch.randn(n, device="cuda:0")
+    ps._core.k_dense_assign(a.data_ptr(), b.data_ptr(), n * 4)
+    assert torch.equal(a.cpu(), b.cpu())
+
+
+def test_kernel_sparse_gather_scatter_vs_torch():
+    import torch
+
+    torch.manual_seed(1)
+    rows, width, nsel = 4096, 64, 512
+    table = torch.randn(rows, width, device="cuda:0")
+    # int64 row ids: same bit pattern as uint64 for values < 2^63
+    idx = torch.randperm(rows, device="cuda:0")[:nsel]
+    out = torch.empty(nsel, width, device="cuda:0")
+    ps._core.k_sparse_gather_f32(table.data_ptr(), idx.data_ptr(), nsel, width, out.data_ptr())
+    ref = table[idx]
+    assert torch.allclose(out.cpu(), ref.cpu())
+
+    grad = torch.randn(nsel, width, device="cuda:0")
+    ref2 = table.clone()
+    ref2[idx] += grad
+    ps._core.k_sparse_scatter_add_f32(table.data_ptr(), idx.data_ptr(), nsel, width,
+                                      grad.data_ptr(), False)
+    assert torch.allclose(table.cpu(), ref2.cpu(), atol=1e-6)
+
+
+def test_joint_push_pull_inproc_gpu():
+    _boot_joint_inproc()
+    try:
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(accumulate=False)
+        worker = ps.KVWorker(0, 0)
+        n = 1 << 18  # 1 MiB of floats
+        src = ps.pool_alloc(n * 4)
+        dst = ps.pool_alloc(n * 4)
+        vals = np.random.default_rng(2).standard_normal(n).astype(np.float32)
+        src.copy_from(vals)
+        keys = np.array([42], dtype=np.uint64)
+        lens = np.array([n], dtype=np.int32)
+        ts = worker.zpush_ptr(keys, src.ptr, n * 4, 0, lens, cmd=1)
+        worker.wait(ts)
+        ts = worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens, cmd=1)
+        worker.wait(ts)
+        out = dst.to_numpy_f32()
+        assert np.allclose(out, vals), "in-place zero-copy pull returned wrong data"
+    finally:
+        _down_joint()
+
+
+def test_joint_accumulate_gpu():
+    _boot_joint_inproc()
+    try:
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(accumulate=True)
+        worker = ps.KVWorker(0, 0)
+        n = 4096
+        src = ps.pool_alloc(n * 4)
+        dst = ps.pool_alloc(n * 4)
+        vals = np.arange(n, dtype=np.float32)
+        src.copy_from(vals)
+        keys = np.array([7], dtype=np.uint64)
+        lens = np.array([n], dtype=np.int32)
+        for _ in range(3):
+            worker.wait(worker.zpush_ptr(keys, src.ptr, n * 4, 0, lens, cmd=2))
+        worker.wait(worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens))
+        assert np.allclose(dst.to_numpy_f32(), 3 * vals)
+    finally:
+        _down_joint()
+
+
+def test_sparse_handler_gpu():
+    _boot_joint_inproc()
+    try:
+        rows, width, nsel = 1 << 14, 64, 256
+        server = ps.KVServer(0)
+        server.set_gpu_sparse_handle(rows, width, accumulate=True)
+        worker = ps.KVWorker(0, 0)
+        rng = np.random.default_rng(3)
+        idx = np.sort(rng.choice(rows, size=nsel, replace=False)).astype(np.uint64)
+        grads = rng.standard_normal((nsel, width)).astype(np.float32)
+        vbuf = ps.pool_alloc(grads.nbytes)
+        vbuf.copy_from(grads.reshape(-1))
+        dst = ps.pool_alloc(grads.nbytes)
+        lens = np.full(nsel, width, dtype=np.int32)
+        worker.wait(worker.zpush_ptr(idx, vbuf.ptr, grads.nbytes, 0, lens, cmd=2))
+        worker.wait(worker.zpull_ptr(idx, dst.ptr, grads.nbytes, 0, lens))
+        out = dst.to_numpy_f32().reshape(nsel, width)
+        assert np.allclose(out, grads, atol=1e-6)
+    finally:
+        _down_joint()
+
+
+# ------- cross-process hipIpc on a single GPU (config #2 layout x2) -------
+
+
+def _gpu_worker_fn(ps_mod, rank):
+    server = ps_mod.KVServer(0)
+    server.set_gpu_dense_handle(accumulate=True)
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker = ps_mod.KVWorker(0, 0)
+    n = 1 << 16
+    src = ps_mod.pool_alloc(n * 4)
+    dst = ps_mod.pool_alloc(n * 4)
+    vals = np.full(n, float(rank + 1), dtype=np.float32)
+    src.copy_from(vals)
+    keys = np.array([5], dtype=np.uint64)
+    lens = np.array([n], dtype=np.int32)
+    worker.wait(worker.zpush_ptr(keys, src.ptr, n * 4, 0, lens, cmd=2))
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker.wait(worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens))
+    out = dst.to_numpy_f32()
+    return (float(out[0]), float(out[-1])), server
+
+
+def test_multiprocess_hipipc_two_joint_on_one_gpu():
+    devices = {0: 0, 1: 0}
+    results = launch_local(2, 2, _gpu_worker_fn, joint=True, devices=devices,
+                           env_extra={"XPS_POOL_GB": 4}, timeout=300)
+    # both workers pushed (1+2) with accumulate -> 3 everywhere
+    for rank, (first, last) in results.items():
+        assert first == 3.0 and last == 3.0, results
