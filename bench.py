@@ -76,7 +76,7 @@ def main():
             ps.start(role="scheduler", device=-1)
             ps.finalize(role="scheduler")  # blocks in the final barrier
 
-        sched = threading.Thread(target=sched_main)
+        sched = threading.Thread(target=sched_main, daemon=True)
         sched.start()
     ps.start(role="joint", rank=rank, device=local)
 

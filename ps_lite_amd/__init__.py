@@ -9,6 +9,14 @@ data plane on shm rings + hipIpc/xGMI (see docs/DESIGN.md).
 
 import os
 
+# Load torch's bundled HIP runtime FIRST when torch is present: _core
+# links libamdhip64.so.7 and must bind to the same runtime instance as
+# torch, else whichever loads second fails to initialize the GPU.
+try:  # pragma: no cover
+    import torch  # noqa: F401
+except ImportError:
+    pass
+
 # the extension must be the in-tree build (fails loudly if missing)
 try:
     from . import _core
@@ -28,6 +36,7 @@ from ._core import (  # noqa: F401
     WORKER_GROUP,
     barrier,
     clear_registry,
+    device_sync,
     finalize,
     gpu_count,
     init_env,
