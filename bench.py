@@ -106,17 +106,20 @@ def main():
         push_bufs.append(b)
         pull_bufs.append(ps.pool_alloc(size))
 
+    keys_np = np.array(keys, dtype=np.uint64)
+    push_ptrs = None
+    pull_ptrs = None
+
     def one_round(bufs, pull):
-        tss = []
-        for i in range(len(keys)):
-            if pull:
-                tss.append(worker.zpull_ptr(key_arrs[i], bufs[i].ptr, size, local, lens,
-                                            cmd=cmd))
-            else:
-                tss.append(worker.zpush_ptr(key_arrs[i], bufs[i].ptr, size, local, lens,
-                                            cmd=cmd))
-        for ts in tss:
-            worker.wait(ts)
+        nonlocal push_ptrs, pull_ptrs
+        if pull:
+            if pull_ptrs is None:
+                pull_ptrs = [b.ptr for b in bufs]
+            worker.round(keys_np, pull_ptrs, size, local, cmd, True)
+        else:
+            if push_ptrs is None:
+                push_ptrs = [b.ptr for b in bufs]
+            worker.round(keys_np, push_ptrs, size, local, cmd, False)
 
     # warmup
     for _ in range(args.warmup):

@@ -394,9 +394,9 @@ class KVServer : public SimpleApp {
     msg.meta.option = req.option;
     msg.meta.src_dev = res.vals.device();
     msg.meta.dst_dev = req.dst_dev;
-    if (req.pull && !res.keys.empty()) {
+    if (req.pull && (!res.vals.empty() || !res.keys.empty())) {
       msg.meta.val_len = static_cast<int64_t>(res.vals.nbytes());
-      msg.AddData(res.keys);
+      msg.AddData(res.keys);  // may be empty (sparse: meta.key suffices)
       msg.AddData(res.vals);
       if (!res.lens.empty()) msg.AddData(res.lens);
     }

@@ -111,6 +111,33 @@ class PyKVWorker {
     w_.Wait(ts);
   }
 
+  // benchmark fast path: issue one single-key message per key (push or
+  // pull) and wait for all — the whole round runs in C++ (the reference
+  // benchmark is C++; this keeps the comparison honest at small sizes)
+  void Round(py::array_t<uint64_t> keys, const std::vector<uintptr_t>& ptrs, size_t nbytes,
+             int device, int cmd, bool pull) {
+    size_t nk = static_cast<size_t>(keys.size());
+    XPS_CHECK_EQ(ptrs.size(), nk);
+    size_t n = nbytes / sizeof(float);
+    std::vector<SArray<Key>> karrs(nk);
+    for (size_t i = 0; i < nk; ++i) karrs[i] = SArray<Key>({keys.data()[i]});
+    SArray<int> lens(1);
+    lens[0] = static_cast<int>(n);
+    py::gil_scoped_release rel;
+    std::vector<int> tss(nk);
+    std::vector<std::unique_ptr<SArray<float>>> dsts;
+    for (size_t i = 0; i < nk; ++i) {
+      if (pull) {
+        dsts.emplace_back(new SArray<float>(reinterpret_cast<float*>(ptrs[i]), n, device));
+        tss[i] = w_.ZPull(karrs[i], dsts.back().get(), nullptr, cmd);
+      } else {
+        SArray<float> v(reinterpret_cast<float*>(ptrs[i]), n, device);
+        tss[i] = w_.ZPush(karrs[i], v, lens, cmd);
+      }
+    }
+    for (int ts : tss) w_.Wait(ts);
+  }
+
  private:
   KVWorker<float> w_;
 };
@@ -300,7 +327,9 @@ PYBIND11_MODULE(_core, m) {
       .def("zpull_ptr", &PyKVWorker::ZPullPtr, py::arg("keys"), py::arg("vals_ptr"),
            py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>(),
            py::arg("cmd") = 0, py::arg("keys_dev_ptr") = 0)
-      .def("wait", &PyKVWorker::Wait);
+      .def("wait", &PyKVWorker::Wait)
+      .def("round", &PyKVWorker::Round, py::arg("keys"), py::arg("ptrs"), py::arg("nbytes"),
+           py::arg("device"), py::arg("cmd") = 0, py::arg("pull") = false);
 
   py::class_<PyKVServer>(m, "KVServer")
       .def(py::init<int>(), py::arg("app_id") = 0)

@@ -101,6 +101,7 @@ class LocalCluster:
 
 
 def launch_local(num_workers, num_servers, worker_fn, server_fn=None, env_extra=None,
-                 joint=False, timeout=120, worker_args=()):
-    c = LocalCluster(num_workers, num_servers, env_extra=env_extra, joint=joint)
+                 joint=False, timeout=120, worker_args=(), devices=None):
+    c = LocalCluster(num_workers, num_servers, env_extra=env_extra, joint=joint,
+                     devices=devices)
     return c.run(worker_fn, server_fn=server_fn, timeout=timeout, worker_args=worker_args)
