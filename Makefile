@@ -3,8 +3,18 @@ HIPCC ?= hipcc
 ARCH ?= gfx950
 PYEXT := $(shell python3-config --extension-suffix)
 PYINC := $(shell python3 -m pybind11 --includes)
+# Link against torch's bundled HIP runtime when torch is installed: torch
+# ships its own libamdhip64.so (soname without .so.7). If we linked the
+# system .so.7, a process importing both would run TWO HIP runtimes and
+# pointers could not cross (memory faults). With -L<torch/lib> first,
+# -lamdhip64 resolves to torch's copy and the dynamic loader shares one
+# runtime; on a torch-less box the rpath falls back to /opt/rocm/lib.
+TORCHLIB := $(shell python3 -c "import torch, os; print(os.path.join(os.path.dirname(torch.__file__), 'lib'))" 2>/dev/null)
 CXXFLAGS := -O3 -std=c++17 -fPIC -Wall -Wno-unused-function --offload-arch=$(ARCH) $(PYINC)
 LDFLAGS := -shared -fPIC
+ifneq ($(TORCHLIB),)
+LDFLAGS += -L$(TORCHLIB) -Wl,-rpath,$(TORCHLIB)
+endif
 
 SRCS := csrc/env.cc csrc/wire.cc csrc/tcp.cc csrc/van.cc csrc/postoffice.cc \
         csrc/customer.cc csrc/resender.cc csrc/hip_util.cc csrc/hip_pool.cc \
