@@ -1,14 +1,22 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: dense push+pull goodput + round-trip latency on
-the MI355X parameter server (BASELINE.json metric).
+"""Flagship benchmark: push+pull goodput + round-trip latency on the
+MI355X parameter server (BASELINE.json metric).
 
 Layout (BytePS): N ranks = N GPUs, each a JOINT process (one worker +
 one co-located server sharing the GPU); rank 0 additionally hosts the
 scheduler thread. Metadata rides the shm rings; payloads move GPU<->GPU
 over hipIpc/xGMI (cross-process even on one GPU).
 
-One step = ZPush of every key (wait all) + ZPull of every key (wait
-all), mirroring ps-lite tests/test_benchmark.cc PUSH_PULL mode.
+Modes:
+  dense (default, config #2/#3): one step = ZPush of every key (wait
+      all) + ZPull of every key (wait all) — ps-lite
+      tests/test_benchmark.cc PUSH_PULL semantics.
+  rn50 (config #4): ResNet-50 gradient buckets, each key's pull issued
+      right behind its push (overlapped).
+  sparse (config #5): 1M-row fp32 embedding, 8K hot keys/step,
+      server-side HIP scatter/gather.
+  --cpu (config #1): localhost TCP van, host buffers, CPU default
+      handler — the plumbing baseline; runs without a GPU.
 
 Run standalone (1 GPU) or under `python -m torch.distributed.run
 --nnodes=1 --nproc-per-node N bench.py --gpus N ...`.
@@ -32,15 +40,55 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--mode", choices=["dense", "rn50", "sparse"], default="dense")
+    p.add_argument("--cpu", action="store_true", help="config #1: TCP van, host buffers")
     p.add_argument("--size-mb", type=float, default=64.0,
-                   help="message size per key (headline: 64 MiB)")
+                   help="dense message size per key (headline: 64 MiB)")
     p.add_argument("--keys-per-server", type=int, default=8)
     p.add_argument("--op", choices=["assign", "sum"], default="assign")
+    p.add_argument("--hot-keys", type=int, default=8192, help="sparse keys per step")
+    p.add_argument("--emb-rows", type=int, default=1 << 20)
+    p.add_argument("--emb-width", type=int, default=64)
     p.add_argument("--rtt-iters", type=int, default=100)
     p.add_argument("--no-rtt", action="store_true")
     p.add_argument("--pool-gb", type=int, default=0, help="0 = auto-size")
     p.add_argument("--smoke", action="store_true", help="tiny correctness run")
     return p.parse_args()
+
+
+class Cluster:
+    def __init__(self, ps, n, rank, local, device, root_uri, root_port, pool_gb):
+        self.ps = ps
+        ps.setup_env(n, n, root_uri=root_uri, root_port=root_port,
+                     XPS_DEV_ID=device, XPS_POOL_GB=pool_gb)
+        self.sched = None
+        if rank == 0:
+            def sched_main():
+                ps.start(role="scheduler", device=-1)
+                ps.finalize(role="scheduler")  # blocks in the final barrier
+
+            self.sched = threading.Thread(target=sched_main, daemon=True)
+            self.sched.start()
+        ps.start(role="joint", rank=rank, device=device)
+
+    def finish(self):
+        self.ps.finalize(role="joint")
+        if self.sched is not None:
+            self.sched.join(timeout=60)
+
+
+def reduce_max(values, rank, world):
+    """MAX over ranks via gloo (driver launches us under torchrun)."""
+    if world <= 1:
+        return values
+    import torch
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+    t = torch.tensor(values, dtype=torch.float64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return t.tolist()
 
 
 def main():
@@ -49,146 +97,197 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", 1))
     local = int(os.environ.get("LOCAL_RANK", rank))
     n = max(args.gpus, world)
-    size = int(args.size_mb * (1 << 20))
-    keys_per_server = args.keys_per_server
     if args.smoke:
-        args.steps, args.warmup, keys_per_server = 2, 1, 2
-        size = min(size, 1 << 20)
-
-    total_keys = keys_per_server * n
-    # pool: push bufs + pull bufs (worker) + store + temps (server), per GPU
-    need_gb = max(2, int(4 * total_keys * size / (1 << 30)) + 2)
-    pool_gb = args.pool_gb or need_gb
+        args.steps, args.warmup = 2, 1
+        args.size_mb = min(args.size_mb, 1.0)
+        args.keys_per_server = 2
+        args.hot_keys = 256
 
     master = os.environ.get("MASTER_ADDR", "127.0.0.1")
     master_port = int(os.environ.get("MASTER_PORT", "29400"))
 
     import ps_lite_amd as ps
+    from ps_lite_amd.models import EmbeddingSpec, resnet50_grad_buckets
 
-    assert ps.gpu_count() > 0, "bench.py needs an MI355X (no GPU visible)"
-    ps.setup_env(n, n, root_uri=master if world > 1 else "127.0.0.1",
-                 root_port=master_port + 137,
-                 XPS_DEV_ID=local, XPS_POOL_GB=pool_gb)
+    if args.cpu:
+        assert args.mode == "dense", "--cpu supports the dense mode (config #1)"
+        device = -1
+    else:
+        count = ps.gpu_count()
+        assert count > 0, "bench.py needs an MI355X (no GPU visible); use --cpu for config #1"
+        device = local % count  # fallback lets N ranks share fewer GPUs in dev runs
+        if device != local and rank == 0:
+            print(f"# note: {n} ranks on {count} GPU(s); sharing devices", file=sys.stderr)
 
-    sched = None
-    if rank == 0:
-        def sched_main():
-            ps.start(role="scheduler", device=-1)
-            ps.finalize(role="scheduler")  # blocks in the final barrier
+    # ---- workload geometry ------------------------------------------------
+    step_range = (1 << 64) // n
+    cmd = 2 if args.op == "sum" else 1
+    if args.mode == "dense":
+        size = int(args.size_mb * (1 << 20))
+        msg_sizes = [size] * (args.keys_per_server * n)
+        keys = [s * step_range + i for s in range(n) for i in range(args.keys_per_server)]
+        model_name = "dense push+pull (ps-lite test_benchmark PUSH_PULL)"
+        overlap_pull = False
+    elif args.mode == "rn50":
+        buckets = resnet50_grad_buckets()
+        if args.smoke:
+            buckets = buckets[:8]
+        msg_sizes = [(b + 255) & ~255 for b in buckets]
+        keys = [(i % n) * step_range + (i // n) + 1 for i in range(len(buckets))]
+        cmd = 2 if args.op == "sum" else 1
+        model_name = f"resnet50 grad buckets x{len(buckets)} (BytePS layout)"
+        overlap_pull = True
+    else:  # sparse
+        spec = EmbeddingSpec(rows=args.emb_rows, width=args.emb_width)
+        model_name = f"sparse embedding {spec.rows}x{spec.width}, {args.hot_keys} hot keys/step"
+        msg_sizes = []
+        overlap_pull = False
 
-        sched = threading.Thread(target=sched_main, daemon=True)
-        sched.start()
-    ps.start(role="joint", rank=rank, device=local)
+    total_msg_bytes = sum(msg_sizes)
+    need_gb = max(2, int(4 * total_msg_bytes / (1 << 30)) + 2)
+    pool_gb = args.pool_gb or need_gb
+
+    cluster = Cluster(ps, n, rank, local, device,
+                      master if world > 1 else "127.0.0.1", master_port + 137, pool_gb)
 
     server = ps.KVServer(0)
-    server.set_gpu_dense_handle(accumulate=(args.op == "sum"))
+    if args.cpu:
+        server.set_default_handle()
+    elif args.mode == "sparse":
+        spec = EmbeddingSpec(rows=args.emb_rows, width=args.emb_width)
+        server.set_gpu_sparse_handle(spec.rows_local(n), spec.width, accumulate=True,
+                                     key_shift=spec.key_shift)
+    else:
+        server.set_gpu_dense_handle(accumulate=(args.op == "sum"))
     worker = ps.KVWorker(0, 0)
     ps.barrier("worker", ps.WORKER_GROUP)
 
-    cmd = 2 if args.op == "sum" else 1
-
-    # keys: server s owns [s * 2^64/n, ...); key i of server s = base + i
-    step = (1 << 64) // n
-    keys = []
-    for s in range(n):
-        for i in range(keys_per_server):
-            keys.append(s * step + i)
-    key_arrs = [np.array([k], dtype=np.uint64) for k in keys]
-    nfloats = size // 4
-    lens = np.array([nfloats], dtype=np.int32)
-
-    push_bufs, pull_bufs = [], []
     rng = np.random.default_rng(1234 + rank)
-    fill = rng.standard_normal(nfloats, dtype=np.float32)
-    for _ in keys:
-        b = ps.pool_alloc(size)
-        b.copy_from(fill)
-        push_bufs.append(b)
-        pull_bufs.append(ps.pool_alloc(size))
 
-    keys_np = np.array(keys, dtype=np.uint64)
-    push_ptrs = None
-    pull_ptrs = None
+    # ---- buffers + the per-step closure ----------------------------------
+    if args.mode == "sparse":
+        spec = EmbeddingSpec(rows=args.emb_rows, width=args.emb_width)
+        width = spec.width
+        hot = args.hot_keys
+        nsets = 4
+        key_sets, key_dev_bufs = [], []
+        for s in range(nsets):
+            k = spec.hot_batch(hot, seed=100 * rank + s)
+            key_sets.append(k)
+            kb = ps.pool_alloc(k.nbytes) if device >= 0 else None
+            if kb is not None:
+                kb.copy_from(k)
+            key_dev_bufs.append(kb)
+        vbytes = hot * width * 4
+        push_buf = ps.pool_alloc(vbytes)
+        pull_buf = ps.pool_alloc(vbytes)
+        push_buf.copy_from(rng.standard_normal(hot * width).astype(np.float32))
+        lens = np.full(hot, width, dtype=np.int32)
+        state = {"i": 0}
 
-    def one_round(bufs, pull):
-        nonlocal push_ptrs, pull_ptrs
-        if pull:
-            if pull_ptrs is None:
-                pull_ptrs = [b.ptr for b in bufs]
-            worker.round(keys_np, pull_ptrs, size, local, cmd, True)
+        def one_step():
+            i = state["i"] % nsets
+            state["i"] += 1
+            k, kb = key_sets[i], key_dev_bufs[i]
+            ts = worker.zpush_ptr(k, push_buf.ptr, vbytes, device, lens, cmd=2,
+                                  keys_dev_ptr=kb.ptr if kb else 0)
+            worker.wait(ts)
+            ts = worker.zpull_ptr(k, pull_buf.ptr, vbytes, device, lens, cmd=0,
+                                  keys_dev_ptr=kb.ptr if kb else 0)
+            worker.wait(ts)
+
+        bytes_per_worker_step = 2.0 * vbytes
+    else:
+        keys_np = np.array(keys, dtype=np.uint64)
+        push_bufs, pull_bufs = [], []
+        host_push, host_pull = [], []
+        for sz in msg_sizes:
+            if device >= 0:
+                b = ps.pool_alloc(sz)
+                b.copy_from(rng.standard_normal(sz // 4).astype(np.float32))
+                push_bufs.append(b)
+                pull_bufs.append(ps.pool_alloc(sz))
+            else:
+                a = rng.standard_normal(sz // 4).astype(np.float32)
+                host_push.append(a)
+                host_pull.append(np.zeros(sz // 4, dtype=np.float32))
+        if device >= 0:
+            push_ptrs = [b.ptr for b in push_bufs]
+            pull_ptrs = [b.ptr for b in pull_bufs]
         else:
-            if push_ptrs is None:
-                push_ptrs = [b.ptr for b in bufs]
-            worker.round(keys_np, push_ptrs, size, local, cmd, False)
+            push_ptrs = [a.ctypes.data for a in host_push]
+            pull_ptrs = [a.ctypes.data for a in host_pull]
+        uniform = len(set(msg_sizes)) == 1
 
-    # warmup
+        def one_step():
+            if uniform and not overlap_pull:
+                worker.round(keys_np, push_ptrs, msg_sizes[0], device, cmd, False)
+                worker.round(keys_np, pull_ptrs, msg_sizes[0], device, cmd, True)
+            else:
+                tss = []
+                for i, sz in enumerate(msg_sizes):
+                    l = np.array([sz // 4], dtype=np.int32)
+                    ka = keys_np[i:i + 1]
+                    tss.append(worker.zpush_ptr(ka, push_ptrs[i], sz, device, l, cmd=cmd))
+                    if overlap_pull:
+                        tss.append(worker.zpull_ptr(ka, pull_ptrs[i], sz, device, l, cmd=cmd))
+                if not overlap_pull:
+                    for ts in tss:
+                        worker.wait(ts)
+                    tss = [worker.zpull_ptr(keys_np[i:i + 1], pull_ptrs[i], sz, device,
+                                            np.array([sz // 4], dtype=np.int32), cmd=cmd)
+                           for i, sz in enumerate(msg_sizes)]
+                for ts in tss:
+                    worker.wait(ts)
+
+        bytes_per_worker_step = 2.0 * total_msg_bytes
+
     for _ in range(args.warmup):
-        one_round(push_bufs, pull=False)
-        one_round(pull_bufs, pull=True)
+        one_step()
 
-    if args.smoke:
+    if args.smoke and args.mode == "dense" and device >= 0:
         got = pull_bufs[0].to_numpy_f32()
-        assert np.allclose(got, fill, atol=1e-6), "smoke: pulled values mismatch"
+        want = push_bufs[0].to_numpy_f32()
+        assert np.allclose(got, want, atol=1e-6), "smoke: pulled values mismatch"
         print("SMOKE_OK: pulled values match pushed values")
 
-    try:
-        import torch
-        has_torch = torch.cuda.is_available()
-    except Exception:
-        torch, has_torch = None, False
-
     def sync():
-        if has_torch:
-            torch.cuda.synchronize(local)
-        else:
-            ps.device_sync(local)
+        if device >= 0:
+            ps.device_sync(device)
 
     ps.barrier("worker", ps.WORKER_GROUP)
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        one_round(push_bufs, pull=False)
-        one_round(pull_bufs, pull=True)
+        one_step()
     sync()
     t1 = time.perf_counter()
     ps.barrier("worker", ps.WORKER_GROUP)
     elapsed = t1 - t0
 
-    # p50 round-trip: one blocking push+pull of a single key
+    # p50 round trip: one blocking push+pull of a single key
     rtts = {}
-    if not args.no_rtt and not args.smoke:
+    if not args.no_rtt and not args.smoke and args.mode == "dense":
         for j, (sz_name, sz) in enumerate((("1mb", 1 << 20), ("64mb", 64 << 20))):
-            if sz > size:
+            if sz > max(msg_sizes):
                 continue
-            # dedicated key so the store entry matches this message size
             rtt_key = np.array([10_000_000 + 100 * rank + j], dtype=np.uint64)
             samples = []
             l = np.array([sz // 4], dtype=np.int32)
             for _ in range(args.rtt_iters):
                 a = time.perf_counter()
-                worker.wait(worker.zpush_ptr(rtt_key, push_bufs[0].ptr, sz, local, l,
-                                             cmd=cmd))
-                worker.wait(worker.zpull_ptr(rtt_key, pull_bufs[0].ptr, sz, local, l,
-                                             cmd=cmd))
+                worker.wait(worker.zpush_ptr(rtt_key, push_ptrs[0], sz, device, l, cmd=cmd))
+                worker.wait(worker.zpull_ptr(rtt_key, pull_ptrs[0], sz, device, l, cmd=cmd))
                 samples.append((time.perf_counter() - a) * 1e6)
             rtts[sz_name] = statistics.median(samples)
 
-    # MAX elapsed over ranks (gloo reduction when multi-rank)
-    if world > 1 and has_torch:
-        import torch.distributed as dist
-        dist.init_process_group("gloo", rank=rank, world_size=world)
-        t = torch.tensor([elapsed], dtype=torch.float64)
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.item())
-        for k in ("1mb", "64mb"):
-            if k in rtts:
-                t = torch.tensor([rtts[k]], dtype=torch.float64)
-                dist.all_reduce(t, op=dist.ReduceOp.MAX)
-                rtts[k] = float(t.item())
-        dist.destroy_process_group()
+    vals = reduce_max([elapsed] + [rtts.get("1mb", 0.0), rtts.get("64mb", 0.0)], rank, world)
+    elapsed = vals[0]
+    if "1mb" in rtts:
+        rtts["1mb"] = vals[1]
+    if "64mb" in rtts:
+        rtts["64mb"] = vals[2]
 
-    # bytes per worker per step: push size*keys + pull size*keys
-    bytes_per_worker_step = 2.0 * total_keys * size
     gbs_per_worker = bytes_per_worker_step * args.steps / elapsed / 1e9
     total_gbs = gbs_per_worker * n
     ms_per_step = elapsed / args.steps * 1e3
@@ -208,22 +307,21 @@ def main():
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
-                "model": "dense push+pull (ps-lite test_benchmark PUSH_PULL)",
-                "global_batch": total_keys,
-                "seq_len": size,
-                "parallelism": f"byteps-joint x{n} (worker+server per GPU)",
-                "msg_bytes": size,
-                "keys_per_server": keys_per_server,
+                "model": model_name,
+                "global_batch": len(msg_sizes) or args.hot_keys,
+                "seq_len": max(msg_sizes) if msg_sizes else args.emb_width,
+                "parallelism": ("cpu-tcp" if args.cpu else f"byteps-joint x{n} (worker+server per GPU)"),
+                "mode": args.mode + ("-cpu" if args.cpu else ""),
+                "msg_bytes": max(msg_sizes) if msg_sizes else args.hot_keys * args.emb_width * 4,
+                "keys_per_server": args.keys_per_server,
                 "op": args.op,
                 "gbs_per_worker": round(gbs_per_worker, 3),
                 "p50_rtt_us": {k: round(v, 1) for k, v in rtts.items()},
             },
         }
-        print(json.dumps(out))
+        print(json.dumps(out, ensure_ascii=False))
 
-    ps.finalize(role="joint")
-    if sched is not None:
-        sched.join()
+    cluster.finish()
 
 
 if __name__ == "__main__":

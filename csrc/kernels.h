@@ -22,15 +22,20 @@ void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s);
 // dst, one write — HBM-optimal multi-worker reduction)
 void DenseSumMultiF32(float* dst, const float* const* srcs_host, int nsrc, size_t n,
                       hipStream_t s);
-// out[r][:] = table[rows[r]][:] for r in [0, nrows)
+// Sparse ops: local row index = (rows[r] >> key_shift) - row_base, so a
+// server can index its local table shard from globally-sharded keys
+// (key = global_row << key_shift spreads rows over the PS key space).
+// out[r][:] = table[row(r)][:] for r in [0, nrows)
 void SparseGatherF32(const float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                     float* out, hipStream_t s);
-// table[rows[r]][:] += src[r][:]; atomic=true tolerates duplicate rows
+                     float* out, hipStream_t s, int key_shift = 0, uint64_t row_base = 0);
+// table[row(r)][:] += src[r][:]; atomic=true tolerates duplicate rows
 void SparseScatterAddF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                         const float* src, bool atomic, hipStream_t s);
-// table[rows[r]][:] = src[r][:]
+                         const float* src, bool atomic, hipStream_t s, int key_shift = 0,
+                         uint64_t row_base = 0);
+// table[row(r)][:] = src[r][:]
 void SparseScatterAssignF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                            const float* src, hipStream_t s);
+                            const float* src, hipStream_t s, int key_shift = 0,
+                            uint64_t row_base = 0);
 
 }  // namespace kern
 }  // namespace xps

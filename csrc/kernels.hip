@@ -80,9 +80,10 @@ __global__ void sum_multi_f32(float4* __restrict__ dst, SrcList srcs, int nsrc, 
 // one wave per row-chunk: rows are contiguous float runs; row_len4 is the
 // row length in float4. Each block strides over rows; lanes stride the row.
 __global__ void gather_rows_f32(const float4* __restrict__ table, const uint64_t* __restrict__ rows,
-                                size_t nrows, size_t row_len4, float4* __restrict__ out) {
+                                size_t nrows, size_t row_len4, float4* __restrict__ out,
+                                int shift, uint64_t base) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    const float4* src = table + rows[r] * row_len4;
+    const float4* src = table + ((rows[r] >> shift) - base) * row_len4;
     float4* dst = out + r * row_len4;
     for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) dst[c] = src[c];
   }
@@ -90,9 +91,9 @@ __global__ void gather_rows_f32(const float4* __restrict__ table, const uint64_t
 
 __global__ void scatter_add_rows_f32(float4* __restrict__ table, const uint64_t* __restrict__ rows,
                                      size_t nrows, size_t row_len4,
-                                     const float4* __restrict__ src) {
+                                     const float4* __restrict__ src, int shift, uint64_t base) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float4* dst = table + rows[r] * row_len4;
+    float4* dst = table + ((rows[r] >> shift) - base) * row_len4;
     const float4* s = src + r * row_len4;
     for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) {
       float4 d = dst[c];
@@ -108,9 +109,10 @@ __global__ void scatter_add_rows_f32(float4* __restrict__ table, const uint64_t*
 
 __global__ void gather_rows_scalar_f32(const float* __restrict__ table,
                                        const uint64_t* __restrict__ rows, size_t nrows,
-                                       size_t row_len, float* __restrict__ out) {
+                                       size_t row_len, float* __restrict__ out, int shift,
+                                       uint64_t base) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    const float* src = table + rows[r] * row_len;
+    const float* src = table + ((rows[r] >> shift) - base) * row_len;
     float* dst = out + r * row_len;
     for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) dst[c] = src[c];
   }
@@ -118,9 +120,10 @@ __global__ void gather_rows_scalar_f32(const float* __restrict__ table,
 
 __global__ void scatter_assign_rows_f32(float4* __restrict__ table,
                                         const uint64_t* __restrict__ rows, size_t nrows,
-                                        size_t row_len4, const float4* __restrict__ src) {
+                                        size_t row_len4, const float4* __restrict__ src,
+                                        int shift, uint64_t base) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float4* dst = table + rows[r] * row_len4;
+    float4* dst = table + ((rows[r] >> shift) - base) * row_len4;
     const float4* s = src + r * row_len4;
     for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) dst[c] = s[c];
   }
@@ -128,9 +131,10 @@ __global__ void scatter_assign_rows_f32(float4* __restrict__ table,
 
 __global__ void scatter_assign_rows_scalar_f32(float* __restrict__ table,
                                                const uint64_t* __restrict__ rows, size_t nrows,
-                                               size_t row_len, const float* __restrict__ src) {
+                                               size_t row_len, const float* __restrict__ src,
+                                               int shift, uint64_t base) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float* dst = table + rows[r] * row_len;
+    float* dst = table + ((rows[r] >> shift) - base) * row_len;
     const float* s = src + r * row_len;
     for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) dst[c] = s[c];
   }
@@ -138,9 +142,10 @@ __global__ void scatter_assign_rows_scalar_f32(float* __restrict__ table,
 
 __global__ void scatter_add_rows_atomic_f32(float* __restrict__ table,
                                             const uint64_t* __restrict__ rows, size_t nrows,
-                                            size_t row_len, const float* __restrict__ src) {
+                                            size_t row_len, const float* __restrict__ src,
+                                            int shift, uint64_t base) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float* dst = table + rows[r] * row_len;
+    float* dst = table + ((rows[r] >> shift) - base) * row_len;
     const float* s = src + r * row_len;
     for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) {
       atomicAdd(&dst[c], s[c]);
@@ -188,38 +193,41 @@ void DenseSumMultiF32(float* dst, const float* const* srcs_host, int nsrc, size_
 }
 
 void SparseGatherF32(const float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                     float* out, hipStream_t s) {
+                     float* out, hipStream_t s, int key_shift, uint64_t row_base) {
   if (row_len % 4 == 0) {
     hipLaunchKernelGGL(gather_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)), dim3(kBlock),
                        0, s, reinterpret_cast<const float4*>(table), rows_dev, nrows, row_len / 4,
-                       reinterpret_cast<float4*>(out));
+                       reinterpret_cast<float4*>(out), key_shift, row_base);
   } else {
     hipLaunchKernelGGL(gather_rows_scalar_f32, dim3(GridFor(nrows * row_len, 16384)), dim3(kBlock),
-                       0, s, table, rows_dev, nrows, row_len, out);
+                       0, s, table, rows_dev, nrows, row_len, out, key_shift, row_base);
   }
 }
 
 void SparseScatterAssignF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                            const float* src, hipStream_t s) {
+                            const float* src, hipStream_t s, int key_shift, uint64_t row_base) {
   if (row_len % 4 == 0) {
     hipLaunchKernelGGL(scatter_assign_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)),
                        dim3(kBlock), 0, s, reinterpret_cast<float4*>(table), rows_dev, nrows,
-                       row_len / 4, reinterpret_cast<const float4*>(src));
+                       row_len / 4, reinterpret_cast<const float4*>(src), key_shift, row_base);
   } else {
     hipLaunchKernelGGL(scatter_assign_rows_scalar_f32, dim3(GridFor(nrows * row_len, 16384)),
-                       dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src);
+                       dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src, key_shift,
+                       row_base);
   }
 }
 
 void SparseScatterAddF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                         const float* src, bool atomic, hipStream_t s) {
+                         const float* src, bool atomic, hipStream_t s, int key_shift,
+                         uint64_t row_base) {
   if (atomic || row_len % 4 != 0) {
     hipLaunchKernelGGL(scatter_add_rows_atomic_f32, dim3(GridFor(nrows * row_len, 16384)),
-                       dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src);
+                       dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src, key_shift,
+                       row_base);
   } else {
     hipLaunchKernelGGL(scatter_add_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)),
                        dim3(kBlock), 0, s, reinterpret_cast<float4*>(table), rows_dev, nrows,
-                       row_len / 4, reinterpret_cast<const float4*>(src));
+                       row_len / 4, reinterpret_cast<const float4*>(src), key_shift, row_base);
   }
 }
 

@@ -160,8 +160,9 @@ class PyKVServer {
     });
   }
 
-  void SetGpuSparseHandle(size_t rows, size_t row_len, bool accumulate) {
-    auto h = std::make_shared<GpuSparseHandler>(s_.postoffice(), rows, row_len, accumulate);
+  void SetGpuSparseHandle(size_t rows, size_t row_len, bool accumulate, int key_shift) {
+    auto h = std::make_shared<GpuSparseHandler>(s_.postoffice(), rows, row_len, accumulate,
+                                                key_shift);
     sparse_ = h;
     s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
       (*h)(m, kvs, srv);
@@ -336,7 +337,7 @@ PYBIND11_MODULE(_core, m) {
       .def("set_default_handle", &PyKVServer::SetDefaultHandle)
       .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("accumulate") = false)
       .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),
-           py::arg("row_len"), py::arg("accumulate") = true)
+           py::arg("row_len"), py::arg("accumulate") = true, py::arg("key_shift") = 0)
       .def("sparse_table_ptr", &PyKVServer::SparseTablePtr)
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
 

@@ -24,6 +24,12 @@ class ThreadsafeQueue {
   }
 
   void WaitAndPop(T* out) {
+    // brief spin before sleeping: saves a ~5-10 µs cv wakeup per hop on
+    // the hot path (customer recv thread), matching the data plane's
+    // spin-then-sleep pollers
+    for (int i = 0; i < 2000; ++i) {
+      if (TryPop(out)) return;
+    }
     std::unique_lock<std::mutex> lk(mu_);
     cv_.wait(lk, [this] { return !q_.empty(); });
     *out = std::move(q_.front());

@@ -145,10 +145,15 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
 
 // ----------------------------------------------------------------- sparse
 
-GpuSparseHandler::GpuSparseHandler(Postoffice* po, size_t rows, size_t row_len, bool accumulate)
-    : po_(po), rows_(rows), row_len_(row_len), accumulate_(accumulate) {
+GpuSparseHandler::GpuSparseHandler(Postoffice* po, size_t rows, size_t row_len, bool accumulate,
+                                   int key_shift)
+    : po_(po), rows_(rows), row_len_(row_len), accumulate_(accumulate), key_shift_(key_shift) {
   auto* pool = HbmPool::Get();
   XPS_CHECK(pool->initialized()) << "GpuSparseHandler needs the HBM pool";
+  if (key_shift_ > 0) {
+    int rank = po_->my_rank();
+    row_base_ = po_->GetServerKeyRanges()[rank].begin >> key_shift_;
+  }
   XPS_HIP_CHECK(hipSetDevice(pool->device()));
   table_ = pool->AllocArray(rows * row_len * sizeof(float));
   XPS_HIP_CHECK(hipMemset(table_.data(), 0, table_.size()));
@@ -190,9 +195,10 @@ void GpuSparseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
     const uint64_t* rows = DeviceKeys(kvs.keys, req.sender, stream);
     if (accumulate_ || req.cmd == kCmdSum) {
       kern::SparseScatterAddF32(table, rows, n, row_len_, kvs.vals.data(), /*atomic=*/false,
-                                stream);
+                                stream, key_shift_, row_base_);
     } else {
-      kern::SparseScatterAssignF32(table, rows, n, row_len_, kvs.vals.data(), stream);
+      kern::SparseScatterAssignF32(table, rows, n, row_len_, kvs.vals.data(), stream, key_shift_,
+                                   row_base_);
     }
     auto* plane = po_->van() ? po_->van()->plane() : nullptr;
     if (!plane) XPS_HIP_CHECK(hipStreamSynchronize(stream));
@@ -200,7 +206,8 @@ void GpuSparseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
   } else if (req.pull) {
     const uint64_t* rows = DeviceKeys(kvs.keys, req.sender, stream);
     SArray<char> out = pool->AllocArray(n * row_len_ * sizeof(float));
-    kern::SparseGatherF32(table, rows, n, row_len_, reinterpret_cast<float*>(out.data()), stream);
+    kern::SparseGatherF32(table, rows, n, row_len_, reinterpret_cast<float*>(out.data()), stream,
+                          key_shift_, row_base_);
     KVPairs<float> res;
     // keys stay meta-only: device keys must not be dereferenced host-side
     res.vals = SArray<float>::View(out);

@@ -43,8 +43,12 @@ class GpuDenseHandler {
 
 class GpuSparseHandler {
  public:
-  // allocates (and zeroes) a rows x row_len fp32 table in the pool
-  GpuSparseHandler(Postoffice* po, size_t rows, size_t row_len, bool accumulate);
+  // Allocates (and zeroes) a rows x row_len fp32 table shard in the pool.
+  // key_shift: workers send key = global_row << key_shift so the key-range
+  // slicer shards rows evenly; this server's local row =
+  // (key >> key_shift) - row_base, row_base derived from its key range.
+  GpuSparseHandler(Postoffice* po, size_t rows, size_t row_len, bool accumulate,
+                   int key_shift = 0);
   void operator()(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   uintptr_t table_ptr() const { return reinterpret_cast<uintptr_t>(table_.data()); }
 
@@ -58,6 +62,8 @@ class GpuSparseHandler {
   size_t rows_;
   size_t row_len_;
   bool accumulate_;
+  int key_shift_ = 0;
+  uint64_t row_base_ = 0;
   SArray<char> table_;
   std::mutex mu_;
   std::unordered_map<int, SArray<char>> key_scratch_;  // per sender
