@@ -182,7 +182,9 @@ def main():
         push_buf = ps.pool_alloc(vbytes)
         pull_buf = ps.pool_alloc(vbytes)
         push_buf.copy_from(rng.standard_normal(hot * width).astype(np.float32))
-        lens = np.full(hot, width, dtype=np.int32)
+        # uniform row width: lens stays empty (8K explicit lens would blow
+        # the ring-slot inline budget and force the TCP fallback)
+        lens = np.array([], dtype=np.int32)
         state = {"i": 0}
 
         def one_step():
@@ -224,21 +226,8 @@ def main():
                 worker.round(keys_np, push_ptrs, msg_sizes[0], device, cmd, False)
                 worker.round(keys_np, pull_ptrs, msg_sizes[0], device, cmd, True)
             else:
-                tss = []
-                for i, sz in enumerate(msg_sizes):
-                    l = np.array([sz // 4], dtype=np.int32)
-                    ka = keys_np[i:i + 1]
-                    tss.append(worker.zpush_ptr(ka, push_ptrs[i], sz, device, l, cmd=cmd))
-                    if overlap_pull:
-                        tss.append(worker.zpull_ptr(ka, pull_ptrs[i], sz, device, l, cmd=cmd))
-                if not overlap_pull:
-                    for ts in tss:
-                        worker.wait(ts)
-                    tss = [worker.zpull_ptr(keys_np[i:i + 1], pull_ptrs[i], sz, device,
-                                            np.array([sz // 4], dtype=np.int32), cmd=cmd)
-                           for i, sz in enumerate(msg_sizes)]
-                for ts in tss:
-                    worker.wait(ts)
+                worker.round_mixed(keys_np, push_ptrs, pull_ptrs, msg_sizes, device, cmd,
+                                   overlap_pull)
 
         bytes_per_worker_step = 2.0 * total_msg_bytes
 

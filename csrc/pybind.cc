@@ -111,6 +111,41 @@ class PyKVWorker {
     w_.Wait(ts);
   }
 
+  // rn50/mixed-size fast path: per key, push (and optionally the pull
+  // right behind it — per-peer ordering makes that safe), then wait all
+  void RoundMixed(py::array_t<uint64_t> keys, const std::vector<uintptr_t>& push_ptrs,
+                  const std::vector<uintptr_t>& pull_ptrs, const std::vector<size_t>& sizes,
+                  int device, int cmd, bool overlap) {
+    size_t nk = static_cast<size_t>(keys.size());
+    XPS_CHECK_EQ(push_ptrs.size(), nk);
+    std::vector<SArray<Key>> karrs(nk);
+    for (size_t i = 0; i < nk; ++i) karrs[i] = SArray<Key>({keys.data()[i]});
+    py::gil_scoped_release rel;
+    std::vector<int> tss;
+    tss.reserve(2 * nk);
+    std::vector<std::unique_ptr<SArray<float>>> dsts;
+    std::vector<int> pull_tss;
+    auto issue_pull = [&](size_t i) {
+      size_t n = sizes[i] / sizeof(float);
+      dsts.emplace_back(new SArray<float>(reinterpret_cast<float*>(pull_ptrs[i]), n, device));
+      return w_.ZPull(karrs[i], dsts.back().get(), nullptr, cmd);
+    };
+    for (size_t i = 0; i < nk; ++i) {
+      size_t n = sizes[i] / sizeof(float);
+      SArray<int> lens(1);
+      lens[0] = static_cast<int>(n);
+      SArray<float> v(reinterpret_cast<float*>(push_ptrs[i]), n, device);
+      tss.push_back(w_.ZPush(karrs[i], v, lens, cmd));
+      if (overlap && !pull_ptrs.empty()) tss.push_back(issue_pull(i));
+    }
+    if (!overlap && !pull_ptrs.empty()) {
+      for (int ts : tss) w_.Wait(ts);
+      tss.clear();
+      for (size_t i = 0; i < nk; ++i) tss.push_back(issue_pull(i));
+    }
+    for (int ts : tss) w_.Wait(ts);
+  }
+
   // benchmark fast path: issue one single-key message per key (push or
   // pull) and wait for all — the whole round runs in C++ (the reference
   // benchmark is C++; this keeps the comparison honest at small sizes)
@@ -330,7 +365,10 @@ PYBIND11_MODULE(_core, m) {
            py::arg("cmd") = 0, py::arg("keys_dev_ptr") = 0)
       .def("wait", &PyKVWorker::Wait)
       .def("round", &PyKVWorker::Round, py::arg("keys"), py::arg("ptrs"), py::arg("nbytes"),
-           py::arg("device"), py::arg("cmd") = 0, py::arg("pull") = false);
+           py::arg("device"), py::arg("cmd") = 0, py::arg("pull") = false)
+      .def("round_mixed", &PyKVWorker::RoundMixed, py::arg("keys"), py::arg("push_ptrs"),
+           py::arg("pull_ptrs"), py::arg("sizes"), py::arg("device"), py::arg("cmd") = 0,
+           py::arg("overlap") = true);
 
   py::class_<PyKVServer>(m, "KVServer")
       .def(py::init<int>(), py::arg("app_id") = 0)
