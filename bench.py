@@ -147,8 +147,13 @@ def main():
     need_gb = max(2, int(4 * total_msg_bytes / (1 << 30)) + 2)
     pool_gb = args.pool_gb or need_gb
 
+    def trace(msg):
+        if os.environ.get("XPS_BENCH_TRACE"):
+            print(f"# [rank {rank}] {msg}", file=sys.stderr, flush=True)
+
     cluster = Cluster(ps, n, rank, local, device,
                       master if world > 1 else "127.0.0.1", master_port + 137, pool_gb)
+    trace("cluster up")
 
     server = ps.KVServer(0)
     if args.cpu:
@@ -159,8 +164,10 @@ def main():
                                      key_shift=spec.key_shift)
     else:
         server.set_gpu_dense_handle(accumulate=(args.op == "sum"))
+    trace("server handler installed")
     worker = ps.KVWorker(0, 0)
     ps.barrier("worker", ps.WORKER_GROUP)
+    trace("post-handler barrier done")
 
     rng = np.random.default_rng(1234 + rank)
 
@@ -231,8 +238,10 @@ def main():
 
         bytes_per_worker_step = 2.0 * total_msg_bytes
 
-    for _ in range(args.warmup):
+    trace("buffers ready")
+    for w in range(args.warmup):
         one_step()
+        trace(f"warmup step {w} done")
 
     if args.smoke and args.mode == "dense" and device >= 0:
         got = pull_bufs[0].to_numpy_f32()
