@@ -46,8 +46,10 @@ class LogMessage {
  public:
   LogMessage(const char* file, int line, LogLevel lvl) : lvl_(lvl) {
     const char* base = strrchr(file, '/');
-    ss_ << "[xps " << LevelStr(lvl) << " " << (base ? base + 1 : file) << ":" << line << "] ";
+    ss_ << "[xps " << LevelStr(lvl) << " p" << Pid() << " " << (base ? base + 1 : file) << ":"
+        << line << "] ";
   }
+  static long Pid();
   ~LogMessage() noexcept(false) {
     ss_ << "\n";
     std::cerr << ss_.str() << std::flush;

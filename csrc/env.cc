@@ -1,6 +1,13 @@
+#include <unistd.h>
+
 #include "base.h"
 
 namespace xps {
+
+long LogMessage::Pid() {
+  static long pid = getpid();
+  return pid;
+}
 
 Environment* Environment::Get() {
   static Environment inst;

@@ -1,7 +1,10 @@
 #include "gpu_plane.h"
 
+#include <fcntl.h>
 #include <hip/hip_runtime.h>
+#include <semaphore.h>
 #include <sys/prctl.h>
+#include <time.h>
 #include <unistd.h>
 
 #include <cstring>
@@ -38,6 +41,37 @@ struct HandleHash {
   }
 };
 std::unordered_map<HandleKey, void*, HandleHash> g_mapped;
+
+// Host-wide lock around hipIpcOpenMemHandle. Two processes importing
+// each other's pools CONCURRENTLY deadlock inside the HIP runtime
+// (measured on ROCm 7.x: the import handshake needs the exporter's
+// runtime lock, held by its own in-flight import — cross-process AB-BA).
+// Serializing imports host-wide breaks the cycle.
+class IpcOpenLock {
+ public:
+  IpcOpenLock() {
+    sem_ = sem_open("/xps_ipc_open_lock", O_CREAT, 0600, 1);
+  }
+  void Lock() {
+    if (sem_ == SEM_FAILED) return;
+    struct timespec ts;
+    clock_gettime(CLOCK_REALTIME, &ts);
+    ts.tv_sec += 60;
+    if (sem_timedwait(sem_, &ts) != 0) {
+      XPS_LOG(Warning) << "ipc-open lock timeout; proceeding unlocked";
+      timed_out_ = true;
+    }
+  }
+  void Unlock() {
+    if (sem_ != SEM_FAILED && !timed_out_) sem_post(sem_);
+    timed_out_ = false;
+  }
+
+ private:
+  sem_t* sem_ = SEM_FAILED;
+  bool timed_out_ = false;
+};
+IpcOpenLock g_ipc_lock;
 }  // namespace
 
 GpuPlane::GpuPlane(Postoffice* po, int device) : po_(po), device_(device) {
@@ -88,10 +122,17 @@ void GpuPlane::Stop() {
 
 void GpuPlane::OnPeer(const Node& peer) {
   if (peer.host_hash != my_host_hash_) return;
-  std::lock_guard<std::mutex> lk(peers_mu_);
-  auto& p = peers_[peer.id];
-  if (!p) p.reset(new Peer());
-  p->node = peer;
+  Peer* pp;
+  {
+    std::lock_guard<std::mutex> lk(peers_mu_);
+    auto& p = peers_[peer.id];
+    if (!p) p.reset(new Peer());
+    p->node = peer;
+    pp = p.get();
+  }
+  // eager import at bootstrap (serialized by the host-wide ipc lock):
+  // steady state then never touches hipIpcOpenMemHandle
+  if (peer.pool_capacity) PeerPoolBase(pp);
 }
 
 GpuPlane::Peer* GpuPlane::GetPeer(int id) {
@@ -136,7 +177,11 @@ void* GpuPlane::PeerPoolBase(Peer* p) {
         XPS_HIP_CHECK(hipSetDevice(device_));
         hipIpcMemHandle_t h;
         memcpy(&h, p->node.pool_handle, sizeof(h));
+        XPS_VLOG(3) << "ipc-open pool of peer " << p->node.id;
+        g_ipc_lock.Lock();
         hipError_t e = hipIpcOpenMemHandle(&base, h, hipIpcMemLazyEnablePeerAccess);
+        g_ipc_lock.Unlock();
+        XPS_VLOG(3) << "ipc-open pool of peer " << p->node.id << " -> " << base;
         if (e != hipSuccess) {
           XPS_LOG(Warning) << "hipIpcOpenMemHandle(peer " << p->node.id
                            << ") failed: " << hipGetErrorString(e);
@@ -229,6 +274,7 @@ bool GpuPlane::Serialize(const Message& msg, const std::vector<char>& by_ref, st
 }
 
 int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
+  XPS_VLOG(3) << "plane send -> " << peer_node.id << ": " << msg.DebugString();
   Peer* p = GetPeer(peer_node.id);
   {
     std::lock_guard<std::mutex> lk(p->mu);
@@ -303,6 +349,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     DeferSendInternal(p, peer_node.id, std::move(keepalive), std::move(payload), bytes);
   } else {
     if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
+    XPS_VLOG(3) << "plane send done -> " << peer_node.id;
   }
   return bytes;
 }
@@ -313,9 +360,13 @@ void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::s
   hipEvent_t ev = GetEvent();
   XPS_HIP_CHECK(hipSetDevice(device_));
   XPS_HIP_CHECK(hipEventRecord(ev, stream));
-  std::lock_guard<std::mutex> lk(pend_mu_);
-  pending_[peer_id].push_back(Pending{ev, peer_id, std::move(payload), std::move(keepalive), bytes});
+  {
+    std::lock_guard<std::mutex> lk(pend_mu_);
+    pending_[peer_id].push_back(
+        Pending{ev, peer_id, std::move(payload), std::move(keepalive), bytes});
+  }
   pending_count_.fetch_add(1);
+  XPS_VLOG(3) << "deferred send queued -> " << peer_id;
 }
 
 void GpuPlane::CompletionLoop() {
@@ -339,6 +390,9 @@ void GpuPlane::CompletionLoop() {
           Peer* peer = GetPeer(front.peer_id);
           if (peer && EnsureRing(peer)) {
             peer->ring.Push(front.payload.data(), static_cast<uint32_t>(front.payload.size()));
+            XPS_VLOG(3) << "deferred send done -> " << front.peer_id;
+          } else {
+            XPS_LOG(Warning) << "deferred send DROPPED -> " << front.peer_id;
           }
           PutEvent(front.ev);
           dq.pop_front();
@@ -403,6 +457,7 @@ void GpuPlane::RingPollLoop() {
       }
     }
     if (!ok) continue;
+    XPS_VLOG(3) << "ring recv: " << msg.DebugString();
     po_->van()->recv_bytes_ += n + ref_bytes;
     po_->van()->Deliver(std::move(msg));
   }

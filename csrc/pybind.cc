@@ -339,6 +339,19 @@ PYBIND11_MODULE(_core, m) {
 
   m.def("pool_init", [](int device, size_t capacity) { HbmPool::Get()->Init(device, capacity); },
         py::arg("device"), py::arg("capacity") = 0);
+  m.def("pool_ipc_handle", []() {
+    return py::bytes(HbmPool::Get()->ipc_handle(), kIpcHandleBytes);
+  });
+  m.def("ipc_open", [](py::bytes handle) {
+    std::string h = handle;
+    py::gil_scoped_release rel;
+    hipIpcMemHandle_t hh;
+    memcpy(&hh, h.data(), sizeof(hh));
+    void* ptr = nullptr;
+    hipError_t e = hipIpcOpenMemHandle(&ptr, hh, hipIpcMemLazyEnablePeerAccess);
+    XPS_CHECK(e == hipSuccess) << "hipIpcOpenMemHandle: " << hipGetErrorString(e);
+    return reinterpret_cast<uintptr_t>(ptr);
+  });
   m.def("pool_in_use", []() { return HbmPool::Get()->bytes_in_use(); });
   py::class_<PoolBuffer>(m, "PoolBuffer")
       .def_property_readonly("ptr", &PoolBuffer::ptr)
