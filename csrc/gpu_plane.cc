@@ -88,10 +88,17 @@ void GpuPlane::FillSelf(Node* self) {
     memcpy(self->pool_handle, pool->ipc_handle(), kIpcHandleBytes);
   }
   if (!started_) {
-    XPS_CHECK(in_ring_.Create(self->shm_uid)) << "cannot create shm ring";
+    auto* env = Environment::Get();
+    if (!env->GetInt("XPS_PROBE_NO_RING", 0)) {
+      XPS_CHECK(in_ring_.Create(self->shm_uid)) << "cannot create shm ring";
+    }
     started_ = true;
-    poll_thread_ = std::thread([this] { RingPollLoop(); });
-    comp_thread_ = std::thread([this] { CompletionLoop(); });
+    if (!env->GetInt("XPS_PROBE_NO_POLL", 0)) {
+      poll_thread_ = std::thread([this] { RingPollLoop(); });
+    }
+    if (!env->GetInt("XPS_PROBE_NO_COMP", 0)) {
+      comp_thread_ = std::thread([this] { CompletionLoop(); });
+    }
   }
 }
 
@@ -122,17 +129,26 @@ void GpuPlane::Stop() {
 
 void GpuPlane::OnPeer(const Node& peer) {
   if (peer.host_hash != my_host_hash_) return;
-  Peer* pp;
+  std::lock_guard<std::mutex> lk(peers_mu_);
+  auto& p = peers_[peer.id];
+  if (!p) p.reset(new Peer());
+  p->node = peer;
+}
+
+void GpuPlane::ImportPeers() {
+  // collect ids first (don't hold peers_mu_ across the import)
+  std::vector<int> ids;
   {
     std::lock_guard<std::mutex> lk(peers_mu_);
-    auto& p = peers_[peer.id];
-    if (!p) p.reset(new Peer());
-    p->node = peer;
-    pp = p.get();
+    for (auto& kv : peers_) ids.push_back(kv.first);
   }
-  // eager import at bootstrap (serialized by the host-wide ipc lock):
-  // steady state then never touches hipIpcOpenMemHandle
-  if (peer.pool_capacity) PeerPoolBase(pp);
+  for (int id : ids) {
+    Peer* p = GetPeer(id);
+    if (p->node.pool_capacity) {
+      void* base = PeerPoolBase(p);
+      XPS_VLOG(1) << "imported pool of peer " << id << " -> " << base;
+    }
+  }
 }
 
 GpuPlane::Peer* GpuPlane::GetPeer(int id) {

@@ -34,6 +34,7 @@ class GpuPlane : public DataPlane {
   ~GpuPlane() override;
 
   void FillSelf(Node* self) override;
+  void ImportPeers() override;
   bool CanSend(const Message& msg, const Node& peer) override;
   int64_t Send(Message& msg, const Node& peer) override;
   void OnPeer(const Node& peer) override;

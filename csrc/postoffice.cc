@@ -30,6 +30,19 @@ void Postoffice::Start(int customer_id, bool do_barrier) {
   if (do_barrier) {
     Barrier(customer_id, kScheduler | kServerGroup | kWorkerGroup);
   }
+  // Staged hipIpc bootstrap: exactly ONE instance cluster-wide imports
+  // its peers' pools at a time, from this (app) thread. Concurrent
+  // cross-process hipIpcOpenMemHandle — and imports issued from the
+  // data-plane recv threads — deadlock inside the ROCm runtime
+  // (measured; see gpu_plane.cc). Every server/worker instance walks the
+  // same barrier ladder so the sequencing is global; CPU-only instances
+  // just pass through the barriers.
+  if (do_barrier && !is_scheduler()) {
+    for (int id : GetNodeIDs(kServerGroup | kWorkerGroup)) {
+      if (id == node_id_ && van_->plane()) van_->plane()->ImportPeers();
+      Barrier(customer_id, kServerGroup | kWorkerGroup);
+    }
+  }
 }
 
 void Postoffice::Finalize(int customer_id, bool do_barrier) {

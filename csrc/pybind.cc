@@ -342,6 +342,27 @@ PYBIND11_MODULE(_core, m) {
   m.def("pool_ipc_handle", []() {
     return py::bytes(HbmPool::Get()->ipc_handle(), kIpcHandleBytes);
   });
+  m.def("make_stream_events", []() {
+    // mimic the plane's device state: nonblocking streams + events
+    py::gil_scoped_release rel;
+    hipStream_t s;
+    XPS_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking) == hipSuccess);
+    hipEvent_t ev;
+    XPS_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming) == hipSuccess);
+    XPS_CHECK(hipEventRecord(ev, s) == hipSuccess);
+    while (hipEventQuery(ev) != hipSuccess) {}
+    return reinterpret_cast<uintptr_t>(s);
+  });
+  m.def("kernel_on_stream", [](uintptr_t stream, uintptr_t dst, uintptr_t src, size_t n) {
+    py::gil_scoped_release rel;
+    kern::DenseSumF32(reinterpret_cast<float*>(dst), reinterpret_cast<float*>(src), n,
+                      reinterpret_cast<hipStream_t>(stream));
+    hipEvent_t ev;
+    XPS_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming) == hipSuccess);
+    XPS_CHECK(hipEventRecord(ev, reinterpret_cast<hipStream_t>(stream)) == hipSuccess);
+    while (hipEventQuery(ev) != hipSuccess) {}
+    hipEventDestroy(ev);
+  });
   m.def("ipc_open", [](py::bytes handle) {
     std::string h = handle;
     py::gil_scoped_release rel;

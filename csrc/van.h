@@ -34,6 +34,10 @@ class DataPlane {
   virtual void OnPeer(const Node& peer) {}  // called when a peer becomes known
   // fill my Node's pool/device/shm fields before ADD_NODE is sent
   virtual void FillSelf(Node* self) {}
+  // import every known peer's pool (called from the Start thread inside
+  // the staged bootstrap sequence — exactly one instance imports at a
+  // time cluster-wide; see Postoffice::Start)
+  virtual void ImportPeers() {}
   virtual void Stop() {}
 };
 
