@@ -85,7 +85,11 @@ void GpuPlane::FillSelf(Node* self) {
   auto* pool = HbmPool::Get();
   if (pool->initialized()) {
     self->pool_capacity = pool->capacity();
-    memcpy(self->pool_handle, pool->ipc_handle(), kIpcHandleBytes);
+    self->pool_slab_bytes = pool->slab_bytes();
+    self->pool_handles.resize(pool->slab_count());
+    for (size_t i = 0; i < pool->slab_count(); ++i) {
+      memcpy(self->pool_handles[i].data(), pool->slab_handle(i), kIpcHandleBytes);
+    }
   }
   if (!started_) {
     auto* env = Environment::Get();
@@ -145,8 +149,9 @@ void GpuPlane::ImportPeers() {
   for (int id : ids) {
     Peer* p = GetPeer(id);
     if (p->node.pool_capacity) {
-      void* base = PeerPoolBase(p);
-      XPS_VLOG(1) << "imported pool of peer " << id << " -> " << base;
+      bool ok = ImportPeerSlabs(p);
+      XPS_VLOG(1) << "imported pool of peer " << id << " (" << p->node.pool_handles.size()
+                  << " slabs): " << (ok ? "ok" : "FAILED");
     }
   }
 }
@@ -170,21 +175,25 @@ bool GpuPlane::EnsureRing(Peer* p) {
   return p->ring.Open(p->node.shm_uid);
 }
 
-void* GpuPlane::PeerPoolBase(Peer* p) {
+bool GpuPlane::ImportPeerSlabs(Peer* p) {
   {
     std::lock_guard<std::mutex> lk(p->mu);
-    if (p->pool_tried) return p->pool_base;
+    if (p->pool_tried) return !p->slab_bases.empty();
     p->pool_tried = true;
   }
-  void* base = nullptr;
-  if (p->node.pool_capacity) {
-    auto* pool = HbmPool::Get();
-    if (pool->initialized() &&
-        memcmp(p->node.pool_handle, pool->ipc_handle(), kIpcHandleBytes) == 0) {
-      base = pool->base();  // same process (joint): use the local mapping
+  std::vector<void*> bases;
+  auto* pool = HbmPool::Get();
+  size_t nslabs = p->node.pool_handles.size();
+  bool same_process =
+      pool->initialized() && nslabs == pool->slab_count() &&
+      memcmp(p->node.pool_handles[0].data(), pool->slab_handle(0), kIpcHandleBytes) == 0;
+  for (size_t i = 0; i < nslabs; ++i) {
+    void* base = nullptr;
+    if (same_process) {
+      base = pool->slab_base(i);  // joint process: use the local mapping
     } else {
       HandleKey key;
-      memcpy(key.h, p->node.pool_handle, kIpcHandleBytes);
+      memcpy(key.h, p->node.pool_handles[i].data(), kIpcHandleBytes);
       std::lock_guard<std::mutex> lk(g_map_mu);
       auto it = g_mapped.find(key);
       if (it != g_mapped.end()) {
@@ -192,24 +201,45 @@ void* GpuPlane::PeerPoolBase(Peer* p) {
       } else {
         XPS_HIP_CHECK(hipSetDevice(device_));
         hipIpcMemHandle_t h;
-        memcpy(&h, p->node.pool_handle, sizeof(h));
-        XPS_VLOG(3) << "ipc-open pool of peer " << p->node.id;
+        memcpy(&h, p->node.pool_handles[i].data(), sizeof(h));
+        XPS_VLOG(3) << "ipc-open slab " << i << " of peer " << p->node.id;
         g_ipc_lock.Lock();
         hipError_t e = hipIpcOpenMemHandle(&base, h, hipIpcMemLazyEnablePeerAccess);
         g_ipc_lock.Unlock();
-        XPS_VLOG(3) << "ipc-open pool of peer " << p->node.id << " -> " << base;
+        XPS_VLOG(3) << "ipc-open slab " << i << " of peer " << p->node.id << " -> " << base;
         if (e != hipSuccess) {
-          XPS_LOG(Warning) << "hipIpcOpenMemHandle(peer " << p->node.id
+          XPS_LOG(Warning) << "hipIpcOpenMemHandle(peer " << p->node.id << " slab " << i
                            << ") failed: " << hipGetErrorString(e);
           base = nullptr;
         }
         g_mapped[key] = base;
       }
     }
+    if (!base) {
+      std::lock_guard<std::mutex> lk(p->mu);
+      p->slab_bases.clear();
+      return false;
+    }
+    bases.push_back(base);
   }
   std::lock_guard<std::mutex> lk(p->mu);
-  p->pool_base = base;
-  return base;
+  p->slab_bases = std::move(bases);
+  return !p->slab_bases.empty();
+}
+
+char* GpuPlane::ResolvePeer(Peer* p, uint64_t global_off, uint64_t len) {
+  if (!ImportPeerSlabs(p)) return nullptr;
+  uint64_t slab_bytes = p->node.pool_slab_bytes;
+  if (slab_bytes == 0) return nullptr;
+  size_t idx = static_cast<size_t>(global_off / slab_bytes);
+  uint64_t local = global_off % slab_bytes;
+  std::lock_guard<std::mutex> lk(p->mu);
+  if (idx >= p->slab_bases.size() || local + len > slab_bytes) {
+    XPS_LOG(Warning) << "peer offset out of bounds: off=" << global_off << " len=" << len
+                     << " slab=" << idx;
+    return nullptr;
+  }
+  return static_cast<char*>(p->slab_bases[idx]) + local;
 }
 
 hipStream_t GpuPlane::StreamForPeer(int node_id) {
@@ -307,12 +337,12 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   // ---- pull response with device vals: one-sided xGMI write ----------
   if (response && msg.meta.pull && msg.data.size() > 1 && msg.data[1].on_device() &&
       (msg.meta.option & kOptPullAddr)) {
-    void* base = PeerPoolBase(p);
-    if (!base) return -1;
     SArray<char> vals = msg.data[1];
-    XPS_CHECK_LE(msg.meta.addr + vals.size(), p->node.pool_capacity)
-        << "in-place pull write out of peer pool bounds";
-    char* dst = static_cast<char*>(base) + msg.meta.addr;
+    char* dst = ResolvePeer(p, msg.meta.addr, vals.size());
+    if (!dst) {
+      hipStreamSynchronize(StreamForPeer(peer_node.id));
+      return -1;  // TCP fallback (stream drained first)
+    }
     hipStream_t stream = StreamForPeer(peer_node.id);
     XPS_HIP_CHECK(hipSetDevice(device_));
     XPS_HIP_CHECK(hipMemcpyAsync(dst, vals.data(), vals.size(), hipMemcpyDefault, stream));
@@ -460,15 +490,14 @@ void GpuPlane::RingPollLoop() {
         uint64_t off = r.U64();
         uint64_t len = r.U64();
         Peer* sender = GetPeer(msg.meta.sender);
-        void* base = sender ? PeerPoolBase(sender) : nullptr;
-        if (!base) {
+        char* ptr = sender ? ResolvePeer(sender, off, len) : nullptr;
+        if (!ptr) {
           XPS_LOG(Warning) << "dropping by-ref blob: sender pool not mapped (from "
                            << msg.meta.sender << ")";
           ok = false;
           break;
         }
-        msg.data.push_back(
-            SArray<char>(static_cast<char*>(base) + off, len, device_));
+        msg.data.push_back(SArray<char>(ptr, len, device_));
         ref_bytes += len;
       }
     }

@@ -51,7 +51,7 @@ class GpuPlane : public DataPlane {
     Node node;
     ShmRing ring;  // producer handle on the peer's inbound ring
     bool ring_tried = false;
-    void* pool_base = nullptr;  // peer pool mapped into our address space
+    std::vector<void*> slab_bases;  // peer pool slabs mapped into our space
     bool pool_tried = false;
     hipStream_t stream = nullptr;
     std::mutex mu;
@@ -67,7 +67,11 @@ class GpuPlane : public DataPlane {
 
   Peer* GetPeer(int id);
   bool EnsureRing(Peer* p);
-  void* PeerPoolBase(Peer* p);
+  // import every slab of the peer's pool (idempotent)
+  bool ImportPeerSlabs(Peer* p);
+  // translate a peer-pool GLOBAL offset range to a mapped pointer
+  // (nullptr if unmapped or the range is out of bounds / spans slabs)
+  char* ResolvePeer(Peer* p, uint64_t global_off, uint64_t len);
   // serialize msg (meta + blobs, by-ref where flagged) into `out`;
   // by_ref[i] true => blob i encoded as pool offset
   bool Serialize(const Message& msg, const std::vector<char>& by_ref, std::string* out);

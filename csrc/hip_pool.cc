@@ -21,79 +21,112 @@ HbmPool* HbmPool::Get() {
 
 void HbmPool::Init(int device, size_t capacity_bytes) {
   std::lock_guard<std::mutex> lk(mu_);
-  if (base_) {
+  if (!slabs_.empty()) {
     XPS_CHECK_EQ(device_, device) << "HbmPool already initialized on another device";
     return;
   }
+  auto* env = Environment::Get();
   if (capacity_bytes == 0) {
-    capacity_bytes = static_cast<size_t>(Environment::Get()->GetInt("XPS_POOL_GB", 8)) << 30;
+    capacity_bytes = static_cast<size_t>(env->GetInt("XPS_POOL_GB", 8)) << 30;
   }
+  slab_bytes_ = static_cast<size_t>(env->GetInt64("XPS_SLAB_BYTES", kDefaultSlabBytes));
+  XPS_CHECK_LT(slab_bytes_, 2ull << 30)
+      << "slabs must stay under 2 GiB (hipIpcOpenMemHandle hangs at >= 2 GiB)";
+  size_t nslabs = (capacity_bytes + slab_bytes_ - 1) / slab_bytes_;
   XPS_HIP_CHECK(hipSetDevice(device));
-  XPS_HIP_CHECK(hipMalloc(&base_, capacity_bytes));
   device_ = device;
-  capacity_ = capacity_bytes;
-  static_assert(sizeof(hipIpcMemHandle_t) <= sizeof(ipc_handle_), "ipc handle too large");
-  hipIpcMemHandle_t h;
-  hipError_t e = hipIpcGetMemHandle(&h, base_);
-  if (e == hipSuccess) {
-    memcpy(ipc_handle_, &h, sizeof(h));
-  } else {
-    XPS_LOG(Warning) << "hipIpcGetMemHandle failed (" << hipGetErrorString(e)
-                     << "); cross-process zero-copy disabled";
-    memset(ipc_handle_, 0, sizeof(ipc_handle_));
+  uint64_t global = 0;
+  for (size_t i = 0; i < nslabs; ++i) {
+    Slab s;
+    s.capacity = slab_bytes_;
+    XPS_HIP_CHECK(hipMalloc(&s.base, s.capacity));
+    s.global_begin = global;
+    global += s.capacity;
+    static_assert(sizeof(hipIpcMemHandle_t) <= sizeof(s.ipc_handle), "ipc handle too large");
+    hipIpcMemHandle_t h;
+    hipError_t e = hipIpcGetMemHandle(&h, s.base);
+    if (e == hipSuccess) {
+      memcpy(s.ipc_handle, &h, sizeof(h));
+    } else {
+      XPS_LOG(Warning) << "hipIpcGetMemHandle failed (" << hipGetErrorString(e)
+                       << "); cross-process zero-copy disabled for slab " << i;
+      memset(s.ipc_handle, 0, sizeof(s.ipc_handle));
+    }
+    s.free_[0] = s.capacity;
+    slabs_.push_back(std::move(s));
   }
-  free_[0] = capacity_;
-  XPS_VLOG(1) << "HbmPool: " << (capacity_ >> 20) << " MiB on device " << device_;
+  capacity_ = global;
+  XPS_VLOG(1) << "HbmPool: " << (capacity_ >> 20) << " MiB in " << slabs_.size()
+              << " slabs on device " << device_;
 }
 
 void* HbmPool::Alloc(size_t nbytes) {
-  XPS_CHECK(base_) << "HbmPool not initialized";
+  XPS_CHECK(!slabs_.empty()) << "HbmPool not initialized";
   nbytes = (nbytes + kAlign - 1) & ~(kAlign - 1);
+  XPS_CHECK_LE(nbytes, slab_bytes_)
+      << "single allocation exceeds the slab size (" << slab_bytes_
+      << " B); raise XPS_SLAB_BYTES (< 2 GiB) or split the buffer";
   std::lock_guard<std::mutex> lk(mu_);
-  for (auto it = free_.begin(); it != free_.end(); ++it) {
-    if (it->second >= nbytes) {
-      size_t off = it->first;
-      size_t rest = it->second - nbytes;
-      free_.erase(it);
-      if (rest) free_[off + nbytes] = rest;
-      used_[off] = nbytes;
-      return static_cast<char*>(base_) + off;
+  for (auto& slab : slabs_) {
+    for (auto it = slab.free_.begin(); it != slab.free_.end(); ++it) {
+      if (it->second >= nbytes) {
+        size_t off = it->first;
+        size_t rest = it->second - nbytes;
+        slab.free_.erase(it);
+        if (rest) slab.free_[off + nbytes] = rest;
+        slab.used_[off] = nbytes;
+        return static_cast<char*>(slab.base) + off;
+      }
     }
   }
-  XPS_LOG(Fatal) << "HbmPool exhausted: want " << nbytes << " bytes, capacity " << capacity_;
+  size_t in_use = 0;
+  for (auto& slab : slabs_) {
+    for (auto& kv : slab.used_) in_use += kv.second;
+  }
+  XPS_LOG(Fatal) << "HbmPool exhausted: want " << nbytes << " bytes, capacity " << capacity_
+                 << " (in use " << in_use << "); raise XPS_POOL_GB";
   return nullptr;
 }
 
 void HbmPool::Free(void* p) {
   std::lock_guard<std::mutex> lk(mu_);
-  size_t off = static_cast<char*>(p) - static_cast<char*>(base_);
-  auto it = used_.find(off);
-  XPS_CHECK(it != used_.end()) << "HbmPool::Free of unknown pointer";
-  size_t size = it->second;
-  used_.erase(it);
-  // coalesce with neighbors
-  auto next = free_.upper_bound(off);
-  if (next != free_.end() && off + size == next->first) {
-    size += next->second;
-    next = free_.erase(next);
-  }
-  if (next != free_.begin()) {
-    auto prev = std::prev(next);
-    if (prev->first + prev->second == off) {
-      prev->second += size;
-      return;
+  for (auto& slab : slabs_) {
+    const char* b = static_cast<const char*>(slab.base);
+    const char* c = static_cast<const char*>(p);
+    if (c < b || c >= b + slab.capacity) continue;
+    size_t off = c - b;
+    auto it = slab.used_.find(off);
+    XPS_CHECK(it != slab.used_.end()) << "HbmPool::Free of unknown pointer";
+    size_t size = it->second;
+    slab.used_.erase(it);
+    auto next = slab.free_.upper_bound(off);
+    if (next != slab.free_.end() && off + size == next->first) {
+      size += next->second;
+      next = slab.free_.erase(next);
     }
+    if (next != slab.free_.begin()) {
+      auto prev = std::prev(next);
+      if (prev->first + prev->second == off) {
+        prev->second += size;
+        return;
+      }
+    }
+    slab.free_[off] = size;
+    return;
   }
-  free_[off] = size;
+  XPS_LOG(Fatal) << "HbmPool::Free of pointer outside the pool";
 }
 
-bool HbmPool::OffsetOf(const void* p, uint64_t* off) const {
-  if (!base_) return false;
+bool HbmPool::OffsetOf(const void* p, uint64_t* global_off) const {
   const char* c = static_cast<const char*>(p);
-  const char* b = static_cast<const char*>(base_);
-  if (c < b || c >= b + capacity_) return false;
-  *off = static_cast<uint64_t>(c - b);
-  return true;
+  for (auto& slab : slabs_) {
+    const char* b = static_cast<const char*>(slab.base);
+    if (c >= b && c < b + slab.capacity) {
+      *global_off = slab.global_begin + static_cast<uint64_t>(c - b);
+      return true;
+    }
+  }
+  return false;
 }
 
 SArray<char> HbmPool::AllocArray(size_t nbytes) {
@@ -104,7 +137,9 @@ SArray<char> HbmPool::AllocArray(size_t nbytes) {
 size_t HbmPool::bytes_in_use() const {
   std::lock_guard<std::mutex> lk(mu_);
   size_t total = 0;
-  for (auto& kv : used_) total += kv.second;
+  for (auto& slab : slabs_) {
+    for (auto& kv : slab.used_) total += kv.second;
+  }
   return total;
 }
 

@@ -2,17 +2,24 @@
 //
 // MI355X-native replacement for ps-lite's lazy per-pointer ibv_reg_mr
 // cache (src/rdma_van.h:520-548) and the rdma_utils.h MemoryAllocator:
-// one hipMalloc slab per process, exported ONCE via hipIpcMemHandle at
-// bootstrap (it rides the ADD_NODE Node record), so steady-state
-// transfers carry only {offset, len} — no registration, no rendezvous.
-// Sized for 288 GB HBM3E per GPU (default pool 8 GiB, growable via
-// XPS_POOL_GB before Start).
+// hipMalloc slabs exported ONCE via hipIpcMemHandle at bootstrap (they
+// ride the ADD_NODE Node record), so steady-state transfers carry only
+// {global offset, len} — no registration, no rendezvous.
+//
+// The pool is MULTI-SLAB: hipIpcOpenMemHandle deadlocks inside the ROCm
+// runtime for allocations >= 2 GiB (measured on this stack: 2037 MiB
+// imports instantly, 2048 MiB hangs), so the arena is carved from
+// <= 1 GiB hipMalloc slabs, all allocated eagerly at Init so the slab
+// table is complete before it is broadcast. Addresses on the wire are
+// GLOBAL byte offsets over the concatenated slabs; a single allocation
+// never spans slabs.
 #pragma once
 
 #include <cstddef>
 #include <cstdint>
 #include <map>
 #include <mutex>
+#include <vector>
 
 #include "sarray.h"
 
@@ -20,20 +27,26 @@ namespace xps {
 
 class HbmPool {
  public:
+  static const size_t kDefaultSlabBytes = 1ull << 30;  // 1 GiB (< 2 GiB ipc limit)
+
   static HbmPool* Get();
 
-  // Allocate the slab on `device` (idempotent; first call wins).
+  // Allocate all slabs on `device` (idempotent; first call wins).
   void Init(int device, size_t capacity_bytes = 0);
-  bool initialized() const { return base_ != nullptr; }
+  bool initialized() const { return !slabs_.empty(); }
   int device() const { return device_; }
-  void* base() const { return base_; }
   size_t capacity() const { return capacity_; }
-  const char* ipc_handle() const { return ipc_handle_; }
+
+  size_t slab_count() const { return slabs_.size(); }
+  size_t slab_bytes() const { return slab_bytes_; }
+  uint64_t slab_capacity(size_t i) const { return slabs_[i].capacity; }
+  const char* slab_handle(size_t i) const { return slabs_[i].ipc_handle; }
+  void* slab_base(size_t i) const { return slabs_[i].base; }
 
   void* Alloc(size_t nbytes);
   void Free(void* p);
-  // true iff p lies inside the pool; fills byte offset from base
-  bool OffsetOf(const void* p, uint64_t* off) const;
+  // true iff p lies inside the pool; fills the GLOBAL byte offset
+  bool OffsetOf(const void* p, uint64_t* global_off) const;
 
   // SArray drawing from the pool (freed back on last release)
   SArray<char> AllocArray(size_t nbytes);
@@ -45,14 +58,21 @@ class HbmPool {
   size_t bytes_in_use() const;
 
  private:
+  struct Slab {
+    void* base = nullptr;
+    uint64_t capacity = 0;
+    uint64_t global_begin = 0;  // global offset of byte 0
+    char ipc_handle[64] = {0};
+    std::map<size_t, size_t> free_;  // local offset -> size
+    std::map<size_t, size_t> used_;
+  };
+
   HbmPool() = default;
   int device_ = -1;
-  void* base_ = nullptr;
   size_t capacity_ = 0;
-  char ipc_handle_[64] = {0};
+  size_t slab_bytes_ = kDefaultSlabBytes;
   mutable std::mutex mu_;
-  std::map<size_t, size_t> free_;   // offset -> size (coalesced free list)
-  std::map<size_t, size_t> used_;   // offset -> size
+  std::vector<Slab> slabs_;
 };
 
 }  // namespace xps

@@ -8,6 +8,7 @@
 // pool hipIpc handle that replaces per-buffer RDMA rendezvous.
 #pragma once
 
+#include <array>
 #include <sstream>
 #include <string>
 #include <vector>
@@ -33,8 +34,12 @@ struct Node {
   int dev_id = kCPU;           // HIP device ordinal owned by this node (-1 = cpu only)
   uint64_t host_hash = 0;      // same value <=> same physical host (for shm/ipc fast path)
   int is_recovery = 0;
-  uint64_t pool_capacity = 0;  // bytes in the exported HBM pool (0 = none)
-  char pool_handle[kIpcHandleBytes] = {0};  // hipIpcMemHandle of the pool base
+  // the exported HBM pool: uniform <= 1 GiB slabs (hipIpcOpenMemHandle
+  // deadlocks at >= 2 GiB), one ipc handle per slab; addresses on the
+  // wire are global offsets over the concatenated slabs
+  uint64_t pool_capacity = 0;  // total bytes (0 = no pool)
+  uint64_t pool_slab_bytes = 0;
+  std::vector<std::array<char, kIpcHandleBytes>> pool_handles;  // per slab
   uint64_t shm_uid = 0;        // uid for naming this node's shm data-plane segment
 
   std::string DebugString() const {

@@ -339,9 +339,9 @@ PYBIND11_MODULE(_core, m) {
 
   m.def("pool_init", [](int device, size_t capacity) { HbmPool::Get()->Init(device, capacity); },
         py::arg("device"), py::arg("capacity") = 0);
-  m.def("pool_ipc_handle", []() {
-    return py::bytes(HbmPool::Get()->ipc_handle(), kIpcHandleBytes);
-  });
+  m.def("pool_ipc_handle", [](size_t slab) {
+    return py::bytes(HbmPool::Get()->slab_handle(slab), kIpcHandleBytes);
+  }, py::arg("slab") = 0);
   m.def("make_stream_events", []() {
     // mimic the plane's device state: nonblocking streams + events
     py::gil_scoped_release rel;

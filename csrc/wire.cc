@@ -14,7 +14,9 @@ static void PackNode(const Node& n, ByteWriter* w) {
   w->U64(n.host_hash);
   w->I32(n.is_recovery);
   w->U64(n.pool_capacity);
-  w->Raw(n.pool_handle, kIpcHandleBytes);
+  w->U64(n.pool_slab_bytes);
+  w->I32(static_cast<int32_t>(n.pool_handles.size()));
+  for (auto& h : n.pool_handles) w->Raw(h.data(), kIpcHandleBytes);
   w->U64(n.shm_uid);
 }
 
@@ -28,7 +30,10 @@ static void UnpackNode(ByteReader* r, Node* n) {
   n->host_hash = r->U64();
   n->is_recovery = r->I32();
   n->pool_capacity = r->U64();
-  r->Raw(n->pool_handle, kIpcHandleBytes);
+  n->pool_slab_bytes = r->U64();
+  int nh = r->I32();
+  n->pool_handles.resize(nh);
+  for (int i = 0; i < nh; ++i) r->Raw(n->pool_handles[i].data(), kIpcHandleBytes);
   n->shm_uid = r->U64();
 }
 
