@@ -27,6 +27,9 @@ void Postoffice::Start(int customer_id, bool do_barrier) {
   EnsureVan();
   started_ = true;
   van_->Start(customer_id);
+  // a recovered node must not wait on the start barrier: the cluster is
+  // already past it (ps-lite is_recovery behavior)
+  if (van_->my_node().is_recovery) do_barrier = false;
   if (do_barrier) {
     Barrier(customer_id, kScheduler | kServerGroup | kWorkerGroup);
   }

@@ -259,17 +259,26 @@ void Van::RecvLoop(std::shared_ptr<TcpConn> conn) {
     int64_t n = conn->RecvFrame(&meta, &data);
     if (n < 0) {
       if (!stopping_.load()) XPS_VLOG(2) << "connection closed on " << my_node_.id;
+      // forget any route that used this connection
+      std::lock_guard<std::mutex> lk(conn_mu_);
+      for (auto it = conns_.begin(); it != conns_.end();) {
+        if (it->second == conn) {
+          it = conns_.erase(it);
+        } else {
+          ++it;
+        }
+      }
       return;
     }
     recv_bytes_ += n;
     Message msg;
     UnpackMeta(meta.data(), meta.size(), &msg.meta);
     msg.data = std::move(data);
-    // learn the reply route for this connection
+    // learn (or refresh) the reply route for this connection
     if (msg.meta.sender != kEmptyNodeID) {
       std::lock_guard<std::mutex> lk(conn_mu_);
       auto it = conns_.find(msg.meta.sender);
-      if (it == conns_.end()) conns_[msg.meta.sender] = conn;
+      if (it == conns_.end() || it->second->fd() < 0) conns_[msg.meta.sender] = conn;
     }
     if (!msg.meta.control.empty()) {
       ProcessControl(msg, conn);
@@ -340,6 +349,12 @@ void Van::ProcessControl(Message& msg, const std::shared_ptr<TcpConn>& conn) {
 }
 
 void Van::ProcessAddNodeAtScheduler(Message& msg, const std::shared_ptr<TcpConn>& conn) {
+  if (ready_.load()) {
+    // late registration = recovery: a restarted process re-joins and
+    // inherits a dead node's id (ps-lite van.cc:266-320 behavior)
+    ProcessRecoveryAtScheduler(msg, conn);
+    return;
+  }
   std::vector<std::pair<Node, std::shared_ptr<TcpConn>>> batch;
   {
     std::lock_guard<std::mutex> lk(sched_mu_);
@@ -410,10 +425,81 @@ void Van::ProcessAddNodeAtScheduler(Message& msg, const std::shared_ptr<TcpConn>
   ready_ = true;
 }
 
+void Van::ProcessRecoveryAtScheduler(Message& msg, const std::shared_ptr<TcpConn>& conn) {
+  XPS_CHECK_EQ(msg.meta.control.node.size(), 1u);
+  Node node = msg.meta.control.node[0];
+  int timeout = Environment::Get()->GetInt("PS_HEARTBEAT_TIMEOUT", 10);
+  std::vector<int> dead = po_->GetDeadNodes(timeout);
+  int assigned = kEmptyNodeID;
+  for (int id : dead) {
+    bool want_worker = node.role == Node::WORKER;
+    if (IsWorkerID(id) == want_worker) {
+      assigned = id;
+      break;
+    }
+  }
+  if (assigned == kEmptyNodeID) {
+    XPS_LOG(Warning) << "late ADD_NODE but no dead " << Node::RoleStr(node.role)
+                     << " to recover (" << node.DebugString() << ")";
+    return;
+  }
+  node.id = assigned;
+  node.is_recovery = 1;
+  XPS_VLOG(1) << "scheduler: recovering node " << assigned << " as " << node.DebugString();
+  {
+    std::lock_guard<std::mutex> lk(nodes_mu_);
+    nodes_[assigned] = node;
+  }
+  {
+    std::lock_guard<std::mutex> lk(conn_mu_);
+    conns_[assigned] = conn;
+  }
+  po_->UpdateHeartbeat(assigned, time(nullptr));
+  if (plane_) plane_->OnPeer(node);
+  // full node list to the recovered node; its updated record to survivors
+  Message reply;
+  reply.meta.control.cmd = Control::ADD_NODE;
+  reply.meta.request = false;
+  reply.meta.sender = kScheduler;
+  reply.meta.recver = assigned;
+  {
+    std::lock_guard<std::mutex> lk(nodes_mu_);
+    for (auto& kv : nodes_) reply.meta.control.node.push_back(kv.second);
+  }
+  std::string meta;
+  PackMeta(reply.meta, &meta);
+  conn->SendFrame(meta, {});
+  Message update;
+  update.meta.control.cmd = Control::ADD_NODE;
+  update.meta.request = false;
+  update.meta.sender = kScheduler;
+  update.meta.control.node.push_back(node);
+  for (int id : po_->GetNodeIDs(kServerGroup | kWorkerGroup)) {
+    if (id == assigned) continue;
+    update.meta.recver = id;
+    Message copy = update;
+    SendToNode(copy, id);
+  }
+}
+
 void Van::ProcessNodeListAssigned(Message& msg) {
+  bool was_ready = ready_.load();
   {
     std::lock_guard<std::mutex> lk(nodes_mu_);
     for (auto& n : msg.meta.control.node) {
+      auto it = nodes_.find(n.id);
+      if (it != nodes_.end() && it->second.shm_uid != 0 && n.shm_uid != 0 &&
+          it->second.shm_uid != n.shm_uid) {
+        // the id was re-assigned (recovery): drop the stale connection so
+        // the next send redials the new address. (shm_uid==0 means a
+        // pre-bootstrap stub record, NOT a re-assignment.)
+        std::lock_guard<std::mutex> lk2(conn_mu_);
+        auto cit = conns_.find(n.id);
+        if (cit != conns_.end()) {
+          cit->second->Close();
+          conns_.erase(cit);
+        }
+      }
       nodes_[n.id] = n;
       if (n.shm_uid == my_uid_ && n.role == my_node_.role) {
         my_node_ = n;
@@ -427,8 +513,17 @@ void Van::ProcessNodeListAssigned(Message& msg) {
       if (kv.second.id != my_node_.id) plane_->OnPeer(kv.second);
     }
   }
+  if (was_ready) {
+    XPS_VLOG(1) << "node list update processed (" << msg.meta.control.node.size() << " nodes)";
+    return;  // survivor receiving a recovery update
+  }
   XPS_CHECK_NE(my_node_.id, kEmptyNodeID) << "node list did not contain me";
-  XPS_VLOG(1) << "joined as " << my_node_.DebugString();
+  if (my_node_.is_recovery) {
+    XPS_VLOG(1) << "recovered as " << my_node_.DebugString();
+    if (plane_) plane_->ImportPeers();
+  } else {
+    XPS_VLOG(1) << "joined as " << my_node_.DebugString();
+  }
   ready_ = true;
 }
 

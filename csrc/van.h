@@ -75,6 +75,7 @@ class Van {
   void RecvLoop(std::shared_ptr<TcpConn> conn);
   void ProcessControl(Message& msg, const std::shared_ptr<TcpConn>& conn);
   void ProcessAddNodeAtScheduler(Message& msg, const std::shared_ptr<TcpConn>& conn);
+  void ProcessRecoveryAtScheduler(Message& msg, const std::shared_ptr<TcpConn>& conn);
   void ProcessNodeListAssigned(Message& msg);
   void ProcessBarrierAtScheduler(Message& msg);
   void ProcessHeartbeat(Message& msg);
