@@ -163,7 +163,7 @@ def main():
         server.set_gpu_sparse_handle(spec.rows_local(n), spec.width, accumulate=True,
                                      key_shift=spec.key_shift)
     else:
-        server.set_gpu_dense_handle(accumulate=(args.op == "sum"))
+        server.set_gpu_dense_handle(mode=("reduce" if args.mode == "rn50" else args.op))
     trace("server handler installed")
     worker = ps.KVWorker(0, 0)
     ps.barrier("worker", ps.WORKER_GROUP)

@@ -9,6 +9,7 @@
 #include "hip_util.h"
 #include "kernels.h"
 #include "kv_app.h"
+#include "kv_utils.h"
 #include "ps.h"
 #include "server_handlers.h"
 #include "simple_app.h"
@@ -188,10 +189,13 @@ class PyKVServer {
     });
   }
 
-  void SetGpuDenseHandle(bool accumulate) {
-    auto h = std::make_shared<GpuDenseHandler>(s_.postoffice(), accumulate);
-    s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
-      (*h)(m, kvs, srv);
+  void SetGpuDenseHandle(const std::string& mode) {
+    DenseMode m = mode == "sum" ? DenseMode::kSum
+                  : mode == "reduce" ? DenseMode::kReduce
+                                     : DenseMode::kAssign;
+    auto h = std::make_shared<GpuDenseHandler>(s_.postoffice(), m);
+    s_.set_request_handle([h](const KVMeta& m2, const KVPairs<float>& kvs, KVServer<float>* srv) {
+      (*h)(m2, kvs, srv);
     });
   }
 
@@ -407,11 +411,40 @@ PYBIND11_MODULE(_core, m) {
   py::class_<PyKVServer>(m, "KVServer")
       .def(py::init<int>(), py::arg("app_id") = 0)
       .def("set_default_handle", &PyKVServer::SetDefaultHandle)
-      .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("accumulate") = false)
+      .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("mode") = "assign")
       .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),
            py::arg("row_len"), py::arg("accumulate") = true, py::arg("key_shift") = 0)
       .def("sparse_table_ptr", &PyKVServer::SparseTablePtr)
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
+
+  // utility parity (ps-lite parallel_kv_match.h / parallel_sort.h)
+  m.def("parallel_ordered_match",
+        [](py::array_t<uint64_t> src_keys, py::array_t<float> src_vals,
+           py::array_t<uint64_t> dst_keys, size_t k, bool accumulate, int nthreads) {
+          SArray<Key> sk;
+          sk.CopyFrom(src_keys.data(), src_keys.size());
+          SArray<float> sv;
+          sv.CopyFrom(src_vals.data(), src_vals.size());
+          SArray<Key> dk;
+          dk.CopyFrom(dst_keys.data(), dst_keys.size());
+          SArray<float> dv;
+          size_t n = ParallelOrderedMatch(sk, sv, dk, &dv, k,
+                                          accumulate ? AssignOp::kPlus : AssignOp::kAssign,
+                                          nthreads);
+          py::array_t<float> out(dv.size());
+          std::copy(dv.begin(), dv.end(), out.mutable_data());
+          return py::make_tuple(n, out);
+        },
+        py::arg("src_keys"), py::arg("src_vals"), py::arg("dst_keys"), py::arg("k") = 1,
+        py::arg("accumulate") = false, py::arg("nthreads") = 4);
+  m.def("parallel_sort", [](py::array_t<uint64_t> keys, int nthreads) {
+    SArray<Key> k;
+    k.CopyFrom(keys.data(), keys.size());
+    ParallelSort(&k, nthreads);
+    py::array_t<uint64_t> out(k.size());
+    std::copy(k.begin(), k.end(), out.mutable_data());
+    return out;
+  }, py::arg("keys"), py::arg("nthreads") = 4);
 
   // raw kernel entry points (synchronous; for numerics tests)
   m.def("k_dense_assign", [](uintptr_t dst, uintptr_t src, size_t nbytes) {

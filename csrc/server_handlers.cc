@@ -23,9 +23,16 @@ static hipStream_t PeerStream(Postoffice* po, int sender, hipStream_t* fallback)
 
 // ------------------------------------------------------------------ dense
 
-GpuDenseHandler::GpuDenseHandler(Postoffice* po, bool default_sum)
-    : po_(po), default_sum_(default_sum) {
+GpuDenseHandler::GpuDenseHandler(Postoffice* po, DenseMode mode) : po_(po), mode_(mode) {
   XPS_CHECK(HbmPool::Get()->initialized()) << "GpuDenseHandler needs the HBM pool";
+  num_workers_ = std::max(1, po_->num_workers());
+}
+
+GpuDenseHandler::~GpuDenseHandler() {
+  for (auto& kv : store_) {
+    for (auto ev : kv.second.round_events) hipEventDestroy(ev);
+    for (auto ev : kv.second.pull_guard) hipEventDestroy(ev);
+  }
 }
 
 hipStream_t GpuDenseHandler::Stream(int sender) { return PeerStream(po_, sender, &fallback_stream_); }
@@ -41,11 +48,62 @@ void GpuDenseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
   }
 }
 
+void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& kvs,
+                                       KVServer<float>* server) {
+  XPS_CHECK_EQ(kvs.keys.size(), 1u) << "reduce mode is single-key-per-message";
+  size_t len = kvs.lens.empty() ? kvs.vals.nbytes()
+                                : static_cast<size_t>(kvs.lens[0]) * sizeof(float);
+  Entry* e;
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    e = &store_[kvs.keys[0]];
+  }
+  if (e->pushes >= num_workers_) {
+    // a fast worker started the next round before this round's pulls
+    // drained: defer (the KVPairs copy keeps the remote buffer alive)
+    e->waiting_pushes.emplace_back(req, kvs);
+    return;
+  }
+  if (e->buf.size() < len) {
+    e->buf = HbmPool::Get()->AllocArray(len);
+  }
+  hipStream_t stream = Stream(req.sender);
+  XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
+  XPS_CHECK(kvs.vals.on_device()) << "reduce mode needs device vals (pool buffers)";
+  if (e->pushes == 0) {
+    // the previous round's pull copies must finish before we overwrite
+    for (auto ev : e->pull_guard) {
+      XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev, 0));
+      hipEventDestroy(ev);
+    }
+    e->pull_guard.clear();
+    kern::DenseAssign(e->buf.data(), kvs.vals.data(), len, stream);
+  } else {
+    kern::DenseSumF32(reinterpret_cast<float*>(e->buf.data()),
+                      reinterpret_cast<const float*>(kvs.vals.data()), len / sizeof(float),
+                      stream);
+  }
+  e->pushes++;
+  hipEvent_t ev;
+  XPS_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  XPS_HIP_CHECK(hipEventRecord(ev, stream));
+  e->round_events.push_back(ev);
+  server->Response(req);
+  if (e->pushes >= num_workers_) {
+    std::vector<KVMeta> waiting;
+    waiting.swap(e->waiting_pulls);
+    for (auto& w : waiting) RespondPull(w, e, server);
+  }
+}
+
 void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
                                  KVServer<float>* server) {
   size_t n = kvs.keys.size();
   XPS_CHECK_GT(n, 0u);
-  bool sum = req.cmd == kCmdSum || (req.cmd == kCmdDefault && default_sum_);
+  if (mode_ == DenseMode::kReduce) {
+    HandleReducePush(req, kvs, server);
+    return;
+  }
   hipStream_t stream = Stream(req.sender);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
   size_t off = 0;  // bytes into vals
@@ -53,55 +111,96 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   for (size_t i = 0; i < n; ++i) {
     size_t len = kvs.lens.empty() ? kvs.vals.nbytes() / n
                                   : static_cast<size_t>(kvs.lens[i]) * sizeof(float);
-    SArray<char> entry;
+    Entry* e;
     {
       std::lock_guard<std::mutex> lk(mu_);
-      auto& e = store_[kvs.keys[i]];
-      if (e.size() < len) {
-        e = HbmPool::Get()->AllocArray(len);
+      e = &store_[kvs.keys[i]];
+      if (e->buf.size() < len) {
+        e->buf = HbmPool::Get()->AllocArray(len);
         // zero before publishing: another peer's stream may accumulate
         // into this entry concurrently with our first push
-        XPS_HIP_CHECK(hipMemset(e.data(), 0, len));
+        XPS_HIP_CHECK(hipMemset(e->buf.data(), 0, len));
       }
-      entry = e;
     }
+    bool sum = mode_ == DenseMode::kAssign ? req.cmd == kCmdSum : req.cmd != kCmdAssign;
     const char* src = reinterpret_cast<const char*>(kvs.vals.data()) + off;
     if (kvs.vals.on_device()) {
       if (sum) {
-        kern::DenseSumF32(reinterpret_cast<float*>(entry.data()),
+        kern::DenseSumF32(reinterpret_cast<float*>(e->buf.data()),
                           reinterpret_cast<const float*>(src), len / sizeof(float), stream);
       } else {
-        kern::DenseAssign(entry.data(), src, len, stream);
+        kern::DenseAssign(e->buf.data(), src, len, stream);
       }
     } else {
       // host vals (TCP-staged path): correctness-first synchronous route
       if (sum) {
         SArray<char> scratch = HbmPool::Get()->AllocArray(len);
         XPS_HIP_CHECK(hipMemcpy(scratch.data(), src, len, hipMemcpyHostToDevice));
-        kern::DenseSumF32(reinterpret_cast<float*>(entry.data()),
+        kern::DenseSumF32(reinterpret_cast<float*>(e->buf.data()),
                           reinterpret_cast<const float*>(scratch.data()), len / sizeof(float),
                           stream);
         XPS_HIP_CHECK(hipStreamSynchronize(stream));
         synced = true;
       } else {
-        XPS_HIP_CHECK(hipMemcpy(entry.data(), src, len, hipMemcpyHostToDevice));
+        XPS_HIP_CHECK(hipMemcpy(e->buf.data(), src, len, hipMemcpyHostToDevice));
       }
     }
     off += len;
   }
-  // if the response cannot ride the plane (deferred behind this stream),
-  // the worker may reuse its buffer as soon as the ack arrives — sync first
   auto* plane = po_->van() ? po_->van()->plane() : nullptr;
   if (!plane && !synced) XPS_HIP_CHECK(hipStreamSynchronize(stream));
   server->Response(req);
+}
+
+void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* server) {
+  hipStream_t stream = Stream(req.sender);
+  for (auto ev : e->round_events) {
+    XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev, 0));
+  }
+  KVPairs<float> res;
+  res.keys = SArray<Key>({req.key});
+  res.vals = SArray<float>::View(e->buf);
+  SArray<int> lens(1);
+  lens[0] = static_cast<int>(e->buf.size() / sizeof(float));
+  res.lens = lens;
+  if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
+  server->Response(req, res);  // the plane enqueues the copy on `stream` here
+  hipEvent_t pe;
+  XPS_HIP_CHECK(hipEventCreateWithFlags(&pe, hipEventDisableTiming));
+  XPS_HIP_CHECK(hipEventRecord(pe, stream));
+  e->pull_guard.push_back(pe);
+  e->pulls++;
+  if (e->pulls >= num_workers_) {
+    // round over: reset, then replay pushes deferred from the next round
+    e->pushes = 0;
+    e->pulls = 0;
+    for (auto ev : e->round_events) hipEventDestroy(ev);
+    e->round_events.clear();
+    std::vector<std::pair<KVMeta, KVPairs<float>>> deferred;
+    deferred.swap(e->waiting_pushes);
+    for (auto& d : deferred) HandleReducePush(d.first, d.second, server);
+  }
 }
 
 void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
                                  KVServer<float>* server) {
   size_t n = kvs.keys.size();
   XPS_CHECK_GT(n, 0u);
-  hipStream_t stream = Stream(req.sender);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
+  if (mode_ == DenseMode::kReduce) {
+    XPS_CHECK_EQ(n, 1u) << "reduce mode is single-key-per-message";
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = store_.find(kvs.keys[0]);
+    XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[0];
+    Entry* e = &it->second;
+    if (e->pushes < num_workers_) {
+      e->waiting_pulls.push_back(req);  // released by the round's last push
+      return;
+    }
+    RespondPull(req, e, server);
+    return;
+  }
+  hipStream_t stream = Stream(req.sender);
   KVPairs<float> res;
   res.keys = kvs.keys;
   SArray<int> lens(n);
@@ -111,7 +210,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
       std::lock_guard<std::mutex> lk(mu_);
       auto it = store_.find(kvs.keys[0]);
       XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[0];
-      entry = it->second;
+      entry = it->second.buf;
     }
     res.vals = SArray<float>::View(entry);  // zero-copy store view
     lens[0] = static_cast<int>(entry.size() / sizeof(float));
@@ -123,7 +222,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
       for (size_t i = 0; i < n; ++i) {
         auto it = store_.find(kvs.keys[i]);
         XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[i];
-        entries[i] = it->second;
+        entries[i] = it->second.buf;
         lens[i] = static_cast<int>(entries[i].size() / sizeof(float));
         total += entries[i].size();
       }

@@ -24,20 +24,47 @@ static const int kCmdDefault = 0;  // handler's configured default op
 static const int kCmdAssign = 1;
 static const int kCmdSum = 2;
 
+// handler modes
+enum class DenseMode {
+  kAssign,  // push overwrites the store (EmptyHandler / pure-goodput)
+  kSum,     // push accumulates forever (KVServerDefaultHandle)
+  kReduce,  // BytePS round semantics: first push of a round assigns, the
+            // rest accumulate; pulls are HELD until all num_workers
+            // pushes of the round arrived (cross-stream ordered via
+            // events), and the round resets after num_workers pulls.
+};
+
 class GpuDenseHandler {
  public:
-  GpuDenseHandler(Postoffice* po, bool default_sum);
+  GpuDenseHandler(Postoffice* po, DenseMode mode);
+  ~GpuDenseHandler();
   void operator()(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
 
  private:
+  struct Entry {
+    SArray<char> buf;
+    // reduce-mode round state (only the customer thread touches it)
+    int pushes = 0;
+    int pulls = 0;
+    std::vector<KVMeta> waiting_pulls;
+    // pushes of the NEXT round arriving before this round's pulls drain
+    // (KVPairs holds the remote buffer alive until we process + ack)
+    std::vector<std::pair<KVMeta, KVPairs<float>>> waiting_pushes;
+    std::vector<hipEvent_t> round_events;  // one per pusher stream
+    std::vector<hipEvent_t> pull_guard;    // pull copies the next round must wait on
+  };
+
   void HandlePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
+  void HandleReducePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   void HandlePull(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
+  void RespondPull(const KVMeta& req, Entry* e, KVServer<float>* server);
   hipStream_t Stream(int sender);
 
   Postoffice* po_;
-  bool default_sum_;
+  DenseMode mode_;
+  int num_workers_ = 1;
   std::mutex mu_;
-  std::unordered_map<Key, SArray<char>> store_;
+  std::unordered_map<Key, Entry> store_;
   hipStream_t fallback_stream_ = nullptr;
 };
 

@@ -86,7 +86,7 @@ def test_joint_push_pull_inproc_gpu():
     _boot_joint_inproc()
     try:
         server = ps.KVServer(0)
-        server.set_gpu_dense_handle(accumulate=False)
+        server.set_gpu_dense_handle(mode="assign")
         worker = ps.KVWorker(0, 0)
         n = 1 << 18  # 1 MiB of floats
         src = ps.pool_alloc(n * 4)
@@ -109,7 +109,7 @@ def test_joint_accumulate_gpu():
     _boot_joint_inproc()
     try:
         server = ps.KVServer(0)
-        server.set_gpu_dense_handle(accumulate=True)
+        server.set_gpu_dense_handle(mode="sum")
         worker = ps.KVWorker(0, 0)
         n = 4096
         src = ps.pool_alloc(n * 4)
@@ -153,7 +153,7 @@ def test_sparse_handler_gpu():
 
 def _gpu_worker_fn(ps_mod, rank):
     server = ps_mod.KVServer(0)
-    server.set_gpu_dense_handle(accumulate=True)
+    server.set_gpu_dense_handle(mode="sum")
     ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
     worker = ps_mod.KVWorker(0, 0)
     n = 1 << 16

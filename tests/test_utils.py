@@ -1,0 +1,40 @@
+"""Utility parity tests: parallel ordered match / parallel sort
+(ps-lite parallel_kv_match.h / parallel_sort.h)."""
+import numpy as np
+
+import ps_lite_amd as ps
+
+
+def test_parallel_ordered_match():
+    rng = np.random.default_rng(0)
+    src_keys = np.sort(rng.choice(10_000, size=500, replace=False)).astype(np.uint64)
+    k = 4
+    src_vals = rng.standard_normal(len(src_keys) * k).astype(np.float32)
+    dst_keys = np.sort(rng.choice(10_000, size=800, replace=False)).astype(np.uint64)
+    n, out = ps._core.parallel_ordered_match(src_keys, src_vals, dst_keys, k=k, nthreads=4)
+    # reference
+    ref = np.zeros(len(dst_keys) * k, dtype=np.float32)
+    src_map = {int(key): i for i, key in enumerate(src_keys)}
+    expect_n = 0
+    for i, key in enumerate(dst_keys):
+        if int(key) in src_map:
+            si = src_map[int(key)]
+            ref[i * k:(i + 1) * k] = src_vals[si * k:(si + 1) * k]
+            expect_n += 1
+    assert n == expect_n
+    assert np.allclose(out, ref)
+
+
+def test_parallel_ordered_match_accumulate():
+    keys = np.array([1, 2, 3], dtype=np.uint64)
+    vals = np.array([1.0, 2.0, 3.0], dtype=np.float32)
+    n, out = ps._core.parallel_ordered_match(keys, vals, keys, k=1, accumulate=True)
+    assert n == 3
+    assert np.allclose(out, vals)
+
+
+def test_parallel_sort():
+    rng = np.random.default_rng(1)
+    keys = rng.integers(0, 1 << 62, size=100_000, dtype=np.uint64)
+    out = ps._core.parallel_sort(keys, nthreads=4)
+    assert np.array_equal(out, np.sort(keys))
