@@ -169,16 +169,27 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* 
   XPS_HIP_CHECK(hipEventCreateWithFlags(&pe, hipEventDisableTiming));
   XPS_HIP_CHECK(hipEventRecord(pe, stream));
   e->pull_guard.push_back(pe);
+  e->pulled_senders.insert(req.sender);
   e->pulls++;
   if (e->pulls >= num_workers_) {
-    // round over: reset, then replay pushes deferred from the next round
+    // round over: reset, then replay deferred next-round pushes + pulls
     e->pushes = 0;
     e->pulls = 0;
+    e->pulled_senders.clear();
     for (auto ev : e->round_events) hipEventDestroy(ev);
     e->round_events.clear();
-    std::vector<std::pair<KVMeta, KVPairs<float>>> deferred;
-    deferred.swap(e->waiting_pushes);
-    for (auto& d : deferred) HandleReducePush(d.first, d.second, server);
+    std::vector<std::pair<KVMeta, KVPairs<float>>> dpush;
+    dpush.swap(e->waiting_pushes);
+    for (auto& d : dpush) HandleReducePush(d.first, d.second, server);
+    std::vector<KVMeta> dpull;
+    dpull.swap(e->waiting_next_pulls);
+    for (auto& d : dpull) {
+      if (e->pushes >= num_workers_) {
+        RespondPull(d, e, server);
+      } else {
+        e->waiting_pulls.push_back(d);
+      }
+    }
   }
 }
 
@@ -193,6 +204,11 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
     auto it = store_.find(kvs.keys[0]);
     XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[0];
     Entry* e = &it->second;
+    if (e->pulled_senders.count(req.sender)) {
+      // this sender already pulled the current round: a NEXT-round pull
+      e->waiting_next_pulls.push_back(req);
+      return;
+    }
     if (e->pushes < num_workers_) {
       e->waiting_pulls.push_back(req);  // released by the round's last push
       return;

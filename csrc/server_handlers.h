@@ -13,6 +13,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <set>
 #include <unordered_map>
 
 #include "kv_app.h"
@@ -50,6 +51,9 @@ class GpuDenseHandler {
     // pushes of the NEXT round arriving before this round's pulls drain
     // (KVPairs holds the remote buffer alive until we process + ack)
     std::vector<std::pair<KVMeta, KVPairs<float>>> waiting_pushes;
+    // pulls from senders that already pulled this round (next-round pulls)
+    std::vector<KVMeta> waiting_next_pulls;
+    std::set<int> pulled_senders;
     std::vector<hipEvent_t> round_events;  // one per pusher stream
     std::vector<hipEvent_t> pull_guard;    // pull copies the next round must wait on
   };

@@ -170,6 +170,35 @@ def _gpu_worker_fn(ps_mod, rank):
     return (float(out[0]), float(out[-1])), server
 
 
+def _reduce_worker_fn(ps_mod, rank):
+    server = ps_mod.KVServer(0)
+    server.set_gpu_dense_handle(mode="reduce")
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker = ps_mod.KVWorker(0, 0)
+    n = 1 << 14
+    src = ps_mod.pool_alloc(n * 4)
+    dst = ps_mod.pool_alloc(n * 4)
+    keys = np.array([9], dtype=np.uint64)
+    lens = np.array([n], dtype=np.int32)
+    outs = []
+    for step in range(4):  # multiple rounds exercise reset + deferred pushes
+        src.copy_from(np.full(n, float(rank + 1 + step), dtype=np.float32))
+        ts1 = worker.zpush_ptr(keys, src.ptr, n * 4, 0, lens)
+        ts2 = worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens)  # overlapped, no barrier
+        worker.wait(ts1)
+        worker.wait(ts2)
+        outs.append(float(dst.to_numpy_f32()[0]))
+    return outs, server
+
+
+def test_reduce_mode_two_joint_on_one_gpu():
+    results = launch_local(2, 2, _reduce_worker_fn, joint=True, devices={0: 0, 1: 0},
+                           env_extra={"XPS_POOL_GB": 4}, timeout=300)
+    # round k: workers push (1+k) and (2+k) -> both pull 3+2k
+    for rank, outs in results.items():
+        assert outs == [3.0 + 2 * k for k in range(4)], results
+
+
 def test_multiprocess_hipipc_two_joint_on_one_gpu():
     devices = {0: 0, 1: 0}
     results = launch_local(2, 2, _gpu_worker_fn, joint=True, devices=devices,
