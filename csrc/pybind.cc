@@ -184,9 +184,59 @@ class PyKVServer {
 
   void SetDefaultHandle() {
     auto h = std::make_shared<KVServerDefaultHandle<float>>();
+    default_ = h;
     s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
       (*h)(m, kvs, srv);
     });
+  }
+
+  // checkpoint/resume of the installed handler's server state
+  void SaveCheckpoint(const std::string& path) {
+    py::gil_scoped_release rel;
+    if (dense_) {
+      dense_->Save(path);
+    } else if (sparse_) {
+      sparse_->Save(path);
+    } else if (default_) {
+      FILE* f = fopen(path.c_str(), "wb");
+      XPS_CHECK(f) << "cannot open checkpoint " << path;
+      uint64_t n = default_->store.size();
+      fwrite(&n, 8, 1, f);
+      for (auto& kv : default_->store) {
+        uint64_t key = kv.first, len = kv.second.size() * sizeof(float);
+        fwrite(&key, 8, 1, f);
+        fwrite(&len, 8, 1, f);
+        fwrite(kv.second.data(), 1, len, f);
+      }
+      fclose(f);
+    } else {
+      XPS_LOG(Fatal) << "no checkpointable handler installed";
+    }
+  }
+
+  void LoadCheckpoint(const std::string& path) {
+    py::gil_scoped_release rel;
+    if (dense_) {
+      dense_->Load(path);
+    } else if (sparse_) {
+      sparse_->Load(path);
+    } else if (default_) {
+      FILE* f = fopen(path.c_str(), "rb");
+      XPS_CHECK(f) << "cannot open checkpoint " << path;
+      uint64_t n = 0;
+      XPS_CHECK_EQ(fread(&n, 8, 1, f), 1u);
+      for (uint64_t i = 0; i < n; ++i) {
+        uint64_t key, len;
+        XPS_CHECK_EQ(fread(&key, 8, 1, f), 1u);
+        XPS_CHECK_EQ(fread(&len, 8, 1, f), 1u);
+        auto& v = default_->store[key];
+        v.resize(len / sizeof(float));
+        XPS_CHECK_EQ(fread(v.data(), 1, len, f), len);
+      }
+      fclose(f);
+    } else {
+      XPS_LOG(Fatal) << "no checkpointable handler installed";
+    }
   }
 
   void SetGpuDenseHandle(const std::string& mode) {
@@ -194,6 +244,7 @@ class PyKVServer {
                   : mode == "reduce" ? DenseMode::kReduce
                                      : DenseMode::kAssign;
     auto h = std::make_shared<GpuDenseHandler>(s_.postoffice(), m);
+    dense_ = h;
     s_.set_request_handle([h](const KVMeta& m2, const KVPairs<float>& kvs, KVServer<float>* srv) {
       (*h)(m2, kvs, srv);
     });
@@ -250,6 +301,8 @@ class PyKVServer {
  private:
   KVServer<float> s_;
   std::shared_ptr<GpuSparseHandler> sparse_;
+  std::shared_ptr<GpuDenseHandler> dense_;
+  std::shared_ptr<KVServerDefaultHandle<float>> default_;
 };
 
 class PySimpleApp {
@@ -415,6 +468,8 @@ PYBIND11_MODULE(_core, m) {
       .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),
            py::arg("row_len"), py::arg("accumulate") = true, py::arg("key_shift") = 0)
       .def("sparse_table_ptr", &PyKVServer::SparseTablePtr)
+      .def("save_checkpoint", &PyKVServer::SaveCheckpoint)
+      .def("load_checkpoint", &PyKVServer::LoadCheckpoint)
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
 
   // utility parity (ps-lite parallel_kv_match.h / parallel_sort.h)

@@ -258,6 +258,44 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   server->Response(req, res);
 }
 
+void GpuDenseHandler::Save(const std::string& path) {
+  FILE* f = fopen(path.c_str(), "wb");
+  XPS_CHECK(f) << "cannot open checkpoint " << path;
+  std::lock_guard<std::mutex> lk(mu_);
+  uint64_t n = store_.size();
+  fwrite(&n, 8, 1, f);
+  std::vector<char> host;
+  for (auto& kv : store_) {
+    uint64_t key = kv.first, len = kv.second.buf.size();
+    fwrite(&key, 8, 1, f);
+    fwrite(&len, 8, 1, f);
+    host.resize(len);
+    XPS_HIP_CHECK(hipMemcpy(host.data(), kv.second.buf.data(), len, hipMemcpyDeviceToHost));
+    fwrite(host.data(), 1, len, f);
+  }
+  fclose(f);
+}
+
+void GpuDenseHandler::Load(const std::string& path) {
+  FILE* f = fopen(path.c_str(), "rb");
+  XPS_CHECK(f) << "cannot open checkpoint " << path;
+  std::lock_guard<std::mutex> lk(mu_);
+  uint64_t n = 0;
+  XPS_CHECK_EQ(fread(&n, 8, 1, f), 1u);
+  std::vector<char> host;
+  for (uint64_t i = 0; i < n; ++i) {
+    uint64_t key, len;
+    XPS_CHECK_EQ(fread(&key, 8, 1, f), 1u);
+    XPS_CHECK_EQ(fread(&len, 8, 1, f), 1u);
+    host.resize(len);
+    XPS_CHECK_EQ(fread(host.data(), 1, len, f), len);
+    auto& e = store_[key];
+    if (e.buf.size() < len) e.buf = HbmPool::Get()->AllocArray(len);
+    XPS_HIP_CHECK(hipMemcpy(e.buf.data(), host.data(), len, hipMemcpyHostToDevice));
+  }
+  fclose(f);
+}
+
 // ----------------------------------------------------------------- sparse
 
 GpuSparseHandler::GpuSparseHandler(Postoffice* po, size_t rows, size_t row_len, bool accumulate,
@@ -332,6 +370,34 @@ void GpuSparseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
     if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
     server->Response(req, res);
   }
+}
+
+}  // namespace xps
+
+namespace xps {
+
+void GpuSparseHandler::Save(const std::string& path) {
+  FILE* f = fopen(path.c_str(), "wb");
+  XPS_CHECK(f) << "cannot open checkpoint " << path;
+  uint64_t hdr[3] = {rows_, row_len_, static_cast<uint64_t>(key_shift_)};
+  fwrite(hdr, 8, 3, f);
+  std::vector<char> host(table_.size());
+  XPS_HIP_CHECK(hipMemcpy(host.data(), table_.data(), table_.size(), hipMemcpyDeviceToHost));
+  fwrite(host.data(), 1, host.size(), f);
+  fclose(f);
+}
+
+void GpuSparseHandler::Load(const std::string& path) {
+  FILE* f = fopen(path.c_str(), "rb");
+  XPS_CHECK(f) << "cannot open checkpoint " << path;
+  uint64_t hdr[3];
+  XPS_CHECK_EQ(fread(hdr, 8, 3, f), 3u);
+  XPS_CHECK_EQ(hdr[0], rows_);
+  XPS_CHECK_EQ(hdr[1], row_len_);
+  std::vector<char> host(table_.size());
+  XPS_CHECK_EQ(fread(host.data(), 1, host.size(), f), host.size());
+  XPS_HIP_CHECK(hipMemcpy(table_.data(), host.data(), table_.size(), hipMemcpyHostToDevice));
+  fclose(f);
 }
 
 }  // namespace xps

@@ -41,6 +41,11 @@ class GpuDenseHandler {
   ~GpuDenseHandler();
   void operator()(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
 
+  // checkpoint / resume of the server KV state (an aux subsystem the
+  // reference lacks entirely — SURVEY.md §5.4)
+  void Save(const std::string& path);
+  void Load(const std::string& path);
+
  private:
   struct Entry {
     SArray<char> buf;
@@ -82,6 +87,8 @@ class GpuSparseHandler {
                    int key_shift = 0);
   void operator()(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   uintptr_t table_ptr() const { return reinterpret_cast<uintptr_t>(table_.data()); }
+  void Save(const std::string& path);
+  void Load(const std::string& path);
 
  private:
   hipStream_t Stream(int sender);
