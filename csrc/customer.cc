@@ -46,19 +46,29 @@ void Customer::AddResponse(int ts, int num) {
   if (tracker_[ts].second >= tracker_[ts].first) cv_.notify_all();
 }
 
+void Customer::RunHandle(Message& msg) {
+  handle_(msg);
+  if (!msg.meta.request) {
+    std::lock_guard<std::mutex> lk(mu_);
+    tracker_[msg.meta.timestamp].second++;
+    if (tracker_[msg.meta.timestamp].second >= tracker_[msg.meta.timestamp].first) {
+      cv_.notify_all();
+    }
+  }
+}
+
+void Customer::ProcessInline(Message& msg) {
+  std::lock_guard<std::mutex> lk(handle_mu_);
+  RunHandle(msg);
+}
+
 void Customer::Receiving() {
   while (true) {
     Message msg;
     queue_.WaitAndPop(&msg);
     if (msg.meta.control.cmd == Control::TERMINATE) break;
-    handle_(msg);
-    if (!msg.meta.request) {
-      std::lock_guard<std::mutex> lk(mu_);
-      tracker_[msg.meta.timestamp].second++;
-      if (tracker_[msg.meta.timestamp].second >= tracker_[msg.meta.timestamp].first) {
-        cv_.notify_all();
-      }
-    }
+    std::lock_guard<std::mutex> lk(handle_mu_);
+    RunHandle(msg);
   }
 }
 

@@ -40,8 +40,15 @@ class Customer {
   // called by the Van for every data message addressed to this customer
   void Accept(Message msg) { queue_.Push(std::move(msg)); }
 
+  // Inline delivery (the data plane's single poll thread): runs the
+  // handler + tracker update on the CALLER's thread, serialized against
+  // the queue thread by handle_mu_ — saves the queue hop + cv wakeup on
+  // the hot path.
+  void ProcessInline(Message& msg);
+
  private:
   void Receiving();
+  void RunHandle(Message& msg);
 
   int app_id_;
   int customer_id_;
@@ -49,6 +56,7 @@ class Customer {
   Postoffice* po_;
   ThreadsafeQueue<Message> queue_;
   std::thread thread_;
+  std::mutex handle_mu_;
   std::mutex mu_;
   std::condition_variable cv_;
   std::vector<std::pair<int, int>> tracker_;  // (expected, received) per ts

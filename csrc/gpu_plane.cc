@@ -76,6 +76,7 @@ IpcOpenLock g_ipc_lock;
 
 GpuPlane::GpuPlane(Postoffice* po, int device) : po_(po), device_(device) {
   my_host_hash_ = HostHash();
+  inline_deliver_ = Environment::Get()->GetInt("XPS_INLINE_HANDLER", 1) != 0;
 }
 
 GpuPlane::~GpuPlane() { Stop(); }
@@ -504,7 +505,11 @@ void GpuPlane::RingPollLoop() {
     if (!ok) continue;
     XPS_VLOG(3) << "ring recv: " << msg.DebugString();
     po_->van()->recv_bytes_ += n + ref_bytes;
-    po_->van()->Deliver(std::move(msg));
+    if (inline_deliver_) {
+      po_->van()->DeliverInline(msg);
+    } else {
+      po_->van()->Deliver(std::move(msg));
+    }
   }
 }
 

@@ -45,6 +45,9 @@ class GpuPlane : public DataPlane {
   // runs on (gives the per-(key,peer) ordering guarantee)
   hipStream_t StreamForPeer(int node_id);
   int device() const { return device_; }
+  // pooled events (shared with the server handlers)
+  hipEvent_t GetEvent();
+  void PutEvent(hipEvent_t ev);
 
  private:
   struct Peer {
@@ -79,14 +82,13 @@ class GpuPlane : public DataPlane {
   void CompletionLoop();
   void DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::string payload,
                          int64_t bytes);
-  hipEvent_t GetEvent();
-  void PutEvent(hipEvent_t ev);
 
   Postoffice* po_;
   int device_;
   uint64_t my_host_hash_;
   ShmRing in_ring_;
   bool started_ = false;
+  bool inline_deliver_ = true;  // XPS_INLINE_HANDLER (default on)
 
   std::mutex peers_mu_;
   std::unordered_map<int, std::unique_ptr<Peer>> peers_;

@@ -12,13 +12,32 @@ namespace xps {
     XPS_CHECK(e_ == hipSuccess) << "HIP error: " << hipGetErrorString(e_) << " in " #cmd; \
   } while (0)
 
+static GpuPlane* ThePlane(Postoffice* po) {
+  return dynamic_cast<GpuPlane*>(po->van() ? po->van()->plane() : nullptr);
+}
+
 static hipStream_t PeerStream(Postoffice* po, int sender, hipStream_t* fallback) {
-  auto* plane = dynamic_cast<GpuPlane*>(po->van() ? po->van()->plane() : nullptr);
+  auto* plane = ThePlane(po);
   if (plane) return plane->StreamForPeer(sender);
   if (!*fallback) {
     XPS_HIP_CHECK(hipStreamCreateWithFlags(fallback, hipStreamNonBlocking));
   }
   return *fallback;
+}
+
+static hipEvent_t EventAlloc(Postoffice* po) {
+  if (auto* plane = ThePlane(po)) return plane->GetEvent();
+  hipEvent_t ev;
+  XPS_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  return ev;
+}
+
+static void EventFree(Postoffice* po, hipEvent_t ev) {
+  if (auto* plane = ThePlane(po)) {
+    plane->PutEvent(ev);
+  } else {
+    hipEventDestroy(ev);
+  }
 }
 
 // ------------------------------------------------------------------ dense
@@ -74,7 +93,7 @@ void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& 
     // the previous round's pull copies must finish before we overwrite
     for (auto ev : e->pull_guard) {
       XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev, 0));
-      hipEventDestroy(ev);
+      EventFree(po_, ev);
     }
     e->pull_guard.clear();
     kern::DenseAssign(e->buf.data(), kvs.vals.data(), len, stream);
@@ -84,8 +103,7 @@ void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& 
                       stream);
   }
   e->pushes++;
-  hipEvent_t ev;
-  XPS_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  hipEvent_t ev = EventAlloc(po_);
   XPS_HIP_CHECK(hipEventRecord(ev, stream));
   e->round_events.push_back(ev);
   server->Response(req);
@@ -165,8 +183,7 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* 
   res.lens = lens;
   if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
   server->Response(req, res);  // the plane enqueues the copy on `stream` here
-  hipEvent_t pe;
-  XPS_HIP_CHECK(hipEventCreateWithFlags(&pe, hipEventDisableTiming));
+  hipEvent_t pe = EventAlloc(po_);
   XPS_HIP_CHECK(hipEventRecord(pe, stream));
   e->pull_guard.push_back(pe);
   e->pulled_senders.insert(req.sender);
@@ -176,7 +193,7 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* 
     e->pushes = 0;
     e->pulls = 0;
     e->pulled_senders.clear();
-    for (auto ev : e->round_events) hipEventDestroy(ev);
+    for (auto ev : e->round_events) EventFree(po_, ev);
     e->round_events.clear();
     std::vector<std::pair<KVMeta, KVPairs<float>>> dpush;
     dpush.swap(e->waiting_pushes);

@@ -315,6 +315,19 @@ void Van::DeliverData(Message& msg) {
   c->Accept(std::move(msg));
 }
 
+void Van::DeliverInline(Message& msg) {
+  if (!msg.meta.control.empty()) {
+    ProcessControl(msg, nullptr);
+    return;
+  }
+  int app_id = msg.meta.app_id;
+  int customer_id = po_->is_worker() ? msg.meta.customer_id : app_id;
+  Customer* c = po_->GetCustomer(app_id, customer_id, 15);
+  XPS_CHECK(c) << "no customer (app=" << app_id << ", customer=" << customer_id << ") on node "
+               << my_node_.id << " for " << msg.DebugString();
+  c->ProcessInline(msg);
+}
+
 void Van::ProcessControl(Message& msg, const std::shared_ptr<TcpConn>& conn) {
   auto& ctrl = msg.meta.control;
   switch (ctrl.cmd) {

@@ -61,6 +61,9 @@ class Van {
 
   // Deliver a message produced locally (loopback or data-plane receive).
   void Deliver(Message msg);
+  // Data-plane fast delivery: run the customer handler inline on the
+  // caller (the single ring-poll thread) instead of the customer queue.
+  void DeliverInline(Message& msg);
 
   Postoffice* postoffice() const { return po_; }
 
