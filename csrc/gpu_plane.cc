@@ -276,6 +276,14 @@ void GpuPlane::PutEvent(hipEvent_t ev) {
 bool GpuPlane::CanSend(const Message& msg, const Node& peer) {
   if (!started_ || stop_.load()) return false;
   if (peer.host_hash != my_host_hash_ || peer.shm_uid == 0) return false;
+  // If the bootstrap import of this peer's pool failed, assume the
+  // reverse import failed too (same mechanism, same host) and keep
+  // everything on the TCP path — slow but never dropped.
+  if (peer.pool_capacity) {
+    Peer* p = GetPeer(peer.id);
+    std::lock_guard<std::mutex> lk(p->mu);
+    if (p->pool_tried && p->slab_bases.empty()) return false;
+  }
   size_t est = 160 + msg.meta.body.size();
   for (size_t i = 0; i < msg.data.size(); ++i) {
     const auto& d = msg.data[i];
