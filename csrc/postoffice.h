@@ -51,6 +51,9 @@ class Postoffice {
   int my_rank() const;
   int node_id() const { return node_id_; }
   void set_node_id(int id) { node_id_ = id; }
+  // preferred instance rank (DMLC_RANK expanded by DMLC_GROUP_SIZE)
+  int preferred_rank() const { return preferred_rank_; }
+  void set_preferred_rank(int r) { preferred_rank_ = r; }
 
   // ids of every node in a group mask (kScheduler|kServerGroup|kWorkerGroup)
   std::vector<int> GetNodeIDs(int group) const;
@@ -87,6 +90,7 @@ class Postoffice {
   int role_;
   int instance_idx_;
   int node_id_ = kEmptyNodeID;
+  int preferred_rank_ = -1;
   int num_workers_ = 0;
   int num_servers_ = 0;
   std::unique_ptr<Van> van_;

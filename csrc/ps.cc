@@ -27,8 +27,14 @@ static int ResolveDevice(int gpu_device) {
   return dev;
 }
 
-static void StartOne(const std::string& role, int customer_id, bool do_barrier, int device) {
-  Postoffice* po = GetPO(role);
+static void StartOne(const std::string& role, int customer_id, bool do_barrier, int device,
+                     int idx) {
+  Postoffice* po = GetPO(role, idx);
+  int base = Environment::Get()->GetInt("DMLC_RANK", -1);
+  if (base >= 0 && role != "scheduler") {
+    po->set_preferred_rank(base * std::max(1, Environment::Get()->GetInt("DMLC_GROUP_SIZE", 1)) +
+                           idx);
+  }
   po->EnsureVan();
   if (device >= 0 && role != "scheduler") {
     HbmPool::Get()->Init(device);
@@ -38,27 +44,44 @@ static void StartOne(const std::string& role, int customer_id, bool do_barrier, 
   po->Start(customer_id, do_barrier);
 }
 
+// DMLC_GROUP_SIZE instance groups (ps-lite ps.h:110-138 parity): one
+// logical rank expands into group_size PS instances per role in this
+// process, each with its own Van. All instances of this process share
+// one GPU (the per-GPU sharding is the joint-process-per-GPU layout).
+static int GroupSize() { return std::max(1, Environment::Get()->GetInt("DMLC_GROUP_SIZE", 1)); }
+
 void Start(int customer_id, const std::string& role, int rank, bool do_barrier, int gpu_device) {
   if (rank >= 0) Environment::Get()->Set("DMLC_RANK", std::to_string(rank));
   int device = ResolveDevice(gpu_device);
-  if (role == "joint") {
-    // server + worker instances must register concurrently
-    std::thread ts([&] { StartOne("server", customer_id, do_barrier, device); });
-    StartOne("worker", customer_id, do_barrier, device);
-    ts.join();
-  } else {
-    StartOne(role, customer_id, do_barrier, device);
+  int group = role == "scheduler" ? 1 : GroupSize();
+  std::vector<std::thread> threads;
+  for (int g = 0; g < group; ++g) {
+    if (role == "joint" || role == "server") {
+      threads.emplace_back([=] { StartOne("server", customer_id, do_barrier, device, g); });
+    }
+    if (role == "joint" || role == "worker") {
+      threads.emplace_back([=] { StartOne("worker", customer_id, do_barrier, device, g); });
+    }
   }
+  if (role == "scheduler") StartOne("scheduler", customer_id, do_barrier, device, 0);
+  for (auto& t : threads) t.join();
 }
 
 void Finalize(int customer_id, const std::string& role, bool do_barrier) {
-  if (role == "joint") {
-    std::thread ts([&] { Postoffice::GetServer()->Finalize(customer_id, do_barrier); });
-    Postoffice::GetWorker()->Finalize(customer_id, do_barrier);
-    ts.join();
-  } else {
-    GetPO(role)->Finalize(customer_id, do_barrier);
+  int group = role == "scheduler" ? 1 : GroupSize();
+  std::vector<std::thread> threads;
+  for (int g = 0; g < group; ++g) {
+    if (role == "joint" || role == "server") {
+      threads.emplace_back(
+          [=] { Postoffice::GetServer(g)->Finalize(customer_id, do_barrier); });
+    }
+    if (role == "joint" || role == "worker") {
+      threads.emplace_back(
+          [=] { Postoffice::GetWorker(g)->Finalize(customer_id, do_barrier); });
+    }
   }
+  if (role == "scheduler") Postoffice::GetScheduler()->Finalize(customer_id, do_barrier);
+  for (auto& t : threads) t.join();
 }
 
 int NumWorkers() { return Environment::Get()->GetInt("DMLC_NUM_WORKER", 0); }

@@ -41,7 +41,8 @@ struct PoolBuffer {
 
 class PyKVWorker {
  public:
-  PyKVWorker(int app_id, int customer_id) : w_(app_id, customer_id) {}
+  PyKVWorker(int app_id, int customer_id, int instance_idx)
+      : w_(app_id, customer_id, instance_idx) {}
 
   int Push(py::array_t<uint64_t> keys, py::array_t<float> vals, py::array_t<int> lens) {
     // copying variant (safe w.r.t. numpy lifetimes)
@@ -180,7 +181,7 @@ class PyKVWorker {
 
 class PyKVServer {
  public:
-  explicit PyKVServer(int app_id) : s_(app_id) {}
+  explicit PyKVServer(int app_id, int instance_idx) : s_(app_id, instance_idx) {}
 
   void SetDefaultHandle() {
     auto h = std::make_shared<KVServerDefaultHandle<float>>();
@@ -368,14 +369,16 @@ PYBIND11_MODULE(_core, m) {
         },
         py::arg("customer_id") = 0, py::arg("role") = "worker", py::arg("barrier") = true);
   m.def("clear_registry", []() { Postoffice::ClearRegistry(); });
-  m.def("barrier", [](const std::string& role, int group) {
+  m.def("barrier", [](const std::string& role, int group, int idx) {
     py::gil_scoped_release rel;
-    GetPO(role)->Barrier(0, group);
-  });
+    GetPO(role, idx)->Barrier(0, group);
+  }, py::arg("role"), py::arg("group"), py::arg("idx") = 0);
   m.def("num_workers", &NumWorkers);
   m.def("num_servers", &NumServers);
-  m.def("my_rank", [](const std::string& role) { return GetPO(role)->my_rank(); });
-  m.def("node_id", [](const std::string& role) { return GetPO(role)->node_id(); });
+  m.def("my_rank", [](const std::string& role, int idx) { return GetPO(role, idx)->my_rank(); },
+        py::arg("role"), py::arg("idx") = 0);
+  m.def("node_id", [](const std::string& role, int idx) { return GetPO(role, idx)->node_id(); },
+        py::arg("role"), py::arg("idx") = 0);
   m.def("send_bytes", [](const std::string& role) {
     auto* van = GetPO(role)->van();
     return van ? van->send_bytes_.load() : int64_t(0);
@@ -444,7 +447,8 @@ PYBIND11_MODULE(_core, m) {
   });
 
   py::class_<PyKVWorker>(m, "KVWorker")
-      .def(py::init<int, int>(), py::arg("app_id") = 0, py::arg("customer_id") = 0)
+      .def(py::init<int, int, int>(), py::arg("app_id") = 0, py::arg("customer_id") = 0,
+           py::arg("instance_idx") = 0)
       .def("push", &PyKVWorker::Push, py::arg("keys"), py::arg("vals"),
            py::arg("lens") = py::array_t<int>())
       .def("pull", &PyKVWorker::PullBlocking, py::arg("keys"))
@@ -462,7 +466,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("overlap") = true);
 
   py::class_<PyKVServer>(m, "KVServer")
-      .def(py::init<int>(), py::arg("app_id") = 0)
+      .def(py::init<int, int>(), py::arg("app_id") = 0, py::arg("instance_idx") = 0)
       .def("set_default_handle", &PyKVServer::SetDefaultHandle)
       .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("mode") = "assign")
       .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),

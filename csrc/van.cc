@@ -28,7 +28,8 @@ void Van::Start(int customer_id) {
   my_node_.role = po_->role();
   my_node_.hostname = LocalIP();
   my_node_.host_hash = HostHash();
-  my_node_.aux_id = env->GetInt("DMLC_RANK", -1);
+  my_node_.aux_id = po_->preferred_rank() >= 0 ? po_->preferred_rank()
+                                               : env->GetInt("DMLC_RANK", -1);
   std::random_device rd;
   my_uid_ = (static_cast<uint64_t>(rd()) << 32) ^ rd() ^ (static_cast<uint64_t>(getpid()) << 16);
   my_node_.shm_uid = my_uid_;
