@@ -62,6 +62,9 @@ def setup_env(num_workers, num_servers, root_uri="127.0.0.1", root_port=9100, **
         "DMLC_NUM_SERVER": str(num_servers),
         "DMLC_PS_ROOT_URI": root_uri,
         "DMLC_PS_ROOT_PORT": str(root_port),
+        # reset knobs that may linger from a previous cluster in this process
+        "DMLC_GROUP_SIZE": "1",
+        "DMLC_RANK": "-1",
     }
     env.update({k: str(v) for k, v in extra.items()})
     os.environ.update(env)
