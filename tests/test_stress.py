@@ -1,0 +1,41 @@
+"""Concurrency stress: multithreaded sessions hammering one KVWorker
+(the reference's test_benchmark_stress multithreaded-session pattern,
+SURVEY.md §4) plus mixed push/pull traffic across 2x2 processes."""
+import numpy as np
+
+from ps_lite_amd.parallel import launch_local
+
+
+def _threaded_worker(ps, rank):
+    import threading
+
+    worker = ps.KVWorker(0, 0)
+    nthreads, iters, width = 4, 15, 256
+    errors = []
+
+    def session(tid):
+        try:
+            key = np.array([1000 * (rank + 1) + tid], dtype=np.uint64)
+            lens = np.array([width], dtype=np.int32)
+            acc = np.zeros(width, dtype=np.float32)
+            for it in range(iters):
+                vals = np.full(width, float(tid + it + 1), dtype=np.float32)
+                worker.wait(worker.push(key, vals, lens))
+                acc += vals
+                out = worker.pull(key)
+                if not np.allclose(out, acc):
+                    errors.append((tid, it, out[:3].tolist(), acc[:3].tolist()))
+                    return
+        except Exception as e:  # noqa
+            errors.append((tid, repr(e)))
+
+    threads = [threading.Thread(target=session, args=(t,)) for t in range(nthreads)]
+    [t.start() for t in threads]
+    [t.join() for t in threads]
+    return errors
+
+
+def test_multithreaded_sessions():
+    results = launch_local(2, 2, _threaded_worker, timeout=300)
+    for rank, errors in results.items():
+        assert errors == [], (rank, errors)
