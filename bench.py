@@ -46,6 +46,8 @@ def parse_args():
                    help="dense message size per key (headline: 64 MiB)")
     p.add_argument("--keys-per-server", type=int, default=8)
     p.add_argument("--op", choices=["assign", "sum"], default="assign")
+    p.add_argument("--batch-keys", action="store_true",
+                   help="dense: one multi-key message per server per round")
     p.add_argument("--hot-keys", type=int, default=8192, help="sparse keys per step")
     p.add_argument("--emb-rows", type=int, default=1 << 20)
     p.add_argument("--emb-width", type=int, default=64)
@@ -208,6 +210,7 @@ def main():
         bytes_per_worker_step = 2.0 * vbytes
     else:
         keys_np = np.array(keys, dtype=np.uint64)
+        batch = args.batch_keys and args.mode == "dense" and device >= 0
         push_bufs, pull_bufs = [], []
         host_push, host_pull = [], []
         for sz in msg_sizes:
@@ -227,9 +230,26 @@ def main():
             push_ptrs = [a.ctypes.data for a in host_push]
             pull_ptrs = [a.ctypes.data for a in host_pull]
         uniform = len(set(msg_sizes)) == 1
+        if batch:
+            k = args.keys_per_server
+            srv_keys = [keys_np[s * k:(s + 1) * k] for s in range(n)]
+            blens = np.full(k, msg_sizes[0] // 4, dtype=np.int32)
+            bpush = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
+            bpull = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
+            for b in bpush:
+                b.copy_from(rng.standard_normal(k * msg_sizes[0] // 4).astype(np.float32))
 
         def one_step():
-            if uniform and not overlap_pull:
+            if batch:
+                tss = [worker.zpush_ptr(srv_keys[s], bpush[s].ptr, k * msg_sizes[0], device,
+                                        blens, cmd=cmd) for s in range(n)]
+                for ts in tss:
+                    worker.wait(ts)
+                tss = [worker.zpull_ptr(srv_keys[s], bpull[s].ptr, k * msg_sizes[0], device,
+                                        blens, cmd=cmd) for s in range(n)]
+                for ts in tss:
+                    worker.wait(ts)
+            elif uniform and not overlap_pull:
                 worker.round(keys_np, push_ptrs, msg_sizes[0], device, cmd, False)
                 worker.round(keys_np, pull_ptrs, msg_sizes[0], device, cmd, True)
             else:

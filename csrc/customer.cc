@@ -26,6 +26,12 @@ int Customer::NewRequest(int recver) {
 }
 
 void Customer::WaitRequest(int ts) {
+  // brief spin first: responses on the data plane land in ~10 µs and a
+  // cv sleep/wake costs ~5-10 µs per side
+  for (int i = 0; i < 4000; ++i) {
+    std::lock_guard<std::mutex> lk(mu_);
+    if (tracker_[ts].first == tracker_[ts].second) return;
+  }
   std::unique_lock<std::mutex> lk(mu_);
   cv_.wait(lk, [this, ts] { return tracker_[ts].first == tracker_[ts].second; });
 }

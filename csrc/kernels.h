@@ -37,5 +37,17 @@ void SparseScatterAssignF32(float* table, const uint64_t* rows_dev, size_t nrows
                             const float* src, hipStream_t s, int key_shift = 0,
                             uint64_t row_base = 0);
 
+// Batched segmented copy/accumulate: one launch serving up to
+// kMaxBatch (dst, src, nbytes) segments — the device-side slicing/merge
+// of SURVEY.md §2.6 item 3, used for multi-key messages.
+static const int kMaxBatch = 64;
+struct CopyDesc {
+  void* dst;
+  const void* src;
+  size_t nbytes;  // must be 16B-multiples for the batched kernels
+};
+void BatchedAssign(const CopyDesc* descs_host, int n, hipStream_t s);
+void BatchedSumF32(const CopyDesc* descs_host, int n, hipStream_t s);
+
 }  // namespace kern
 }  // namespace xps

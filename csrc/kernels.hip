@@ -8,6 +8,8 @@
 // the loads then ride xGMI.
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
+
 #include "kernels.h"
 
 namespace xps {
@@ -153,7 +155,66 @@ __global__ void scatter_add_rows_atomic_f32(float* __restrict__ table,
   }
 }
 
+struct DescArray {
+  CopyDesc d[kMaxBatch];
+  int n;
+};
+
+// every block strides over every segment's 16B chunks (grid sized for
+// the concatenated total, so all segments together fill the 8 XCDs)
+__global__ void batched_assign_kernel(DescArray da) {
+  for (int seg = 0; seg < da.n; ++seg) {
+    uint4* dst = reinterpret_cast<uint4*>(da.d[seg].dst);
+    const uint4* src = reinterpret_cast<const uint4*>(da.d[seg].src);
+    size_t n4 = da.d[seg].nbytes / 16;
+    size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+    size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+    for (; i < n4; i += stride) dst[i] = src[i];
+  }
+}
+
+__global__ void batched_sum_kernel_f32(DescArray da) {
+  for (int seg = 0; seg < da.n; ++seg) {
+    float4* dst = reinterpret_cast<float4*>(da.d[seg].dst);
+    const float4* src = reinterpret_cast<const float4*>(da.d[seg].src);
+    size_t n4 = da.d[seg].nbytes / 16;
+    size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+    size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+    for (; i < n4; i += stride) {
+      float4 d = dst[i];
+      float4 v = src[i];
+      d.x += v.x;
+      d.y += v.y;
+      d.z += v.z;
+      d.w += v.w;
+      dst[i] = d;
+    }
+  }
+}
+
 }  // namespace
+
+void BatchedAssign(const CopyDesc* descs_host, int n, hipStream_t s) {
+  size_t total = 0;
+  for (int i = 0; i < n; ++i) total += descs_host[i].nbytes;
+  DescArray da{};
+  for (int off = 0; off < n; off += kMaxBatch) {
+    da.n = std::min(n - off, kMaxBatch);
+    for (int i = 0; i < da.n; ++i) da.d[i] = descs_host[off + i];
+    hipLaunchKernelGGL(batched_assign_kernel, dim3(GridFor(total / 16)), dim3(kBlock), 0, s, da);
+  }
+}
+
+void BatchedSumF32(const CopyDesc* descs_host, int n, hipStream_t s) {
+  size_t total = 0;
+  for (int i = 0; i < n; ++i) total += descs_host[i].nbytes;
+  DescArray da{};
+  for (int off = 0; off < n; off += kMaxBatch) {
+    da.n = std::min(n - off, kMaxBatch);
+    for (int i = 0; i < da.n; ++i) da.d[i] = descs_host[off + i];
+    hipLaunchKernelGGL(batched_sum_kernel_f32, dim3(GridFor(total / 16)), dim3(kBlock), 0, s, da);
+  }
+}
 
 void DenseAssign(void* dst, const void* src, size_t nbytes, hipStream_t s) {
   size_t n4 = nbytes / 16;

@@ -124,6 +124,42 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   }
   hipStream_t stream = Stream(req.sender);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
+  bool sum_all = mode_ == DenseMode::kAssign ? req.cmd == kCmdSum : req.cmd != kCmdAssign;
+  // multi-key device push: one batched kernel launch for all segments
+  if (kvs.vals.on_device() && n > 1) {
+    std::vector<kern::CopyDesc> descs;
+    descs.reserve(n);
+    size_t boff = 0;
+    bool aligned = true;
+    for (size_t i = 0; i < n; ++i) {
+      size_t len = kvs.lens.empty() ? kvs.vals.nbytes() / n
+                                    : static_cast<size_t>(kvs.lens[i]) * sizeof(float);
+      Entry* e;
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        e = &store_[kvs.keys[i]];
+        if (e->buf.size() < len) {
+          e->buf = HbmPool::Get()->AllocArray(len);
+          XPS_HIP_CHECK(hipMemset(e->buf.data(), 0, len));
+        }
+      }
+      descs.push_back({e->buf.data(),
+                       reinterpret_cast<const char*>(kvs.vals.data()) + boff, len});
+      aligned = aligned && (len % 16 == 0) && (boff % 16 == 0);
+      boff += len;
+    }
+    if (aligned) {
+      if (sum_all) {
+        kern::BatchedSumF32(descs.data(), static_cast<int>(n), stream);
+      } else {
+        kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
+      }
+      auto* plane0 = po_->van() ? po_->van()->plane() : nullptr;
+      if (!plane0) XPS_HIP_CHECK(hipStreamSynchronize(stream));
+      server->Response(req);
+      return;
+    }
+  }
   size_t off = 0;  // bytes into vals
   bool synced = false;
   for (size_t i = 0; i < n; ++i) {
@@ -140,7 +176,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
         XPS_HIP_CHECK(hipMemset(e->buf.data(), 0, len));
       }
     }
-    bool sum = mode_ == DenseMode::kAssign ? req.cmd == kCmdSum : req.cmd != kCmdAssign;
+    bool sum = sum_all;
     const char* src = reinterpret_cast<const char*>(kvs.vals.data()) + off;
     if (kvs.vals.on_device()) {
       if (sum) {
@@ -262,9 +298,18 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
     }
     SArray<char> tmp = HbmPool::Get()->AllocArray(total);
     size_t off = 0;
+    std::vector<kern::CopyDesc> descs;
+    descs.reserve(n);
+    bool aligned = true;
     for (size_t i = 0; i < n; ++i) {
-      kern::DenseAssign(tmp.data() + off, entries[i].data(), entries[i].size(), stream);
+      descs.push_back({tmp.data() + off, entries[i].data(), entries[i].size()});
+      aligned = aligned && (entries[i].size() % 16 == 0) && (off % 16 == 0);
       off += entries[i].size();
+    }
+    if (aligned) {
+      kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
+    } else {
+      for (auto& d : descs) kern::DenseAssign(d.dst, d.src, d.nbytes, stream);
     }
     res.vals = SArray<float>::View(tmp);  // plane keeps it alive until sent
   }
