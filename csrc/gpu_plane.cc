@@ -116,19 +116,19 @@ void GpuPlane::Stop() {
     std::lock_guard<std::mutex> lk(pend_mu_);
     for (auto& kv : pending_) {
       for (auto& p : kv.second) {
-        hipEventSynchronize(p.ev);
+        (void)hipEventSynchronize(p.ev);
         Peer* peer = GetPeer(p.peer_id);
         if (peer && EnsureRing(peer)) {
           peer->ring.Push(p.payload.data(), static_cast<uint32_t>(p.payload.size()));
         }
-        hipEventDestroy(p.ev);
+        (void)hipEventDestroy(p.ev);
       }
     }
     pending_.clear();
   }
   in_ring_.CloseAndUnlink();
   std::lock_guard<std::mutex> lk(ev_mu_);
-  for (auto ev : event_pool_) hipEventDestroy(ev);
+  for (auto ev : event_pool_) (void)hipEventDestroy(ev);
   event_pool_.clear();
 }
 
@@ -339,7 +339,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   if (!EnsureRing(p)) {
     // TCP fallback for a response must not outrun handler kernels still
     // running on this peer's stream (the worker may reuse buffers on ack)
-    if (response) hipStreamSynchronize(StreamForPeer(peer_node.id));
+    if (response) (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
     return -1;
   }
 
@@ -349,7 +349,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     SArray<char> vals = msg.data[1];
     char* dst = ResolvePeer(p, msg.meta.addr, vals.size());
     if (!dst) {
-      hipStreamSynchronize(StreamForPeer(peer_node.id));
+      (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
       return -1;  // TCP fallback (stream drained first)
     }
     hipStream_t stream = StreamForPeer(peer_node.id);
@@ -394,7 +394,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   }
   std::string payload;
   if (!Serialize(msg, by_ref, &payload)) {
-    if (response) hipStreamSynchronize(StreamForPeer(peer_node.id));
+    if (response) (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
     return -1;
   }
   int64_t bytes = static_cast<int64_t>(payload.size()) + ref_bytes;

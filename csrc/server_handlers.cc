@@ -36,7 +36,7 @@ static void EventFree(Postoffice* po, hipEvent_t ev) {
   if (auto* plane = ThePlane(po)) {
     plane->PutEvent(ev);
   } else {
-    hipEventDestroy(ev);
+    (void)hipEventDestroy(ev);
   }
 }
 
@@ -49,8 +49,8 @@ GpuDenseHandler::GpuDenseHandler(Postoffice* po, DenseMode mode) : po_(po), mode
 
 GpuDenseHandler::~GpuDenseHandler() {
   for (auto& kv : store_) {
-    for (auto ev : kv.second.round_events) hipEventDestroy(ev);
-    for (auto ev : kv.second.pull_guard) hipEventDestroy(ev);
+    for (auto ev : kv.second.round_events) (void)hipEventDestroy(ev);
+    for (auto ev : kv.second.pull_guard) (void)hipEventDestroy(ev);
   }
 }
 

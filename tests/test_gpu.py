@@ -126,6 +126,30 @@ def test_joint_accumulate_gpu():
         _down_joint()
 
 
+def test_multikey_message_gpu():
+    """Multi-key push/pull in ONE message exercises the batched
+    segmented kernels (BatchedAssign/BatchedSumF32)."""
+    _boot_joint_inproc()
+    try:
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(mode="sum")
+        worker = ps.KVWorker(0, 0)
+        nk, n = 3, 4096
+        src = ps.pool_alloc(nk * n * 4)
+        dst = ps.pool_alloc(nk * n * 4)
+        vals = np.concatenate([np.full(n, float(i + 1), dtype=np.float32) for i in range(nk)])
+        src.copy_from(vals)
+        keys = np.array([10, 20, 30], dtype=np.uint64)
+        lens = np.full(nk, n, dtype=np.int32)
+        for _ in range(2):
+            worker.wait(worker.zpush_ptr(keys, src.ptr, nk * n * 4, 0, lens, cmd=2))
+        worker.wait(worker.zpull_ptr(keys, dst.ptr, nk * n * 4, 0, lens))
+        out = dst.to_numpy_f32()
+        assert np.allclose(out, 2 * vals), out[::n]
+    finally:
+        _down_joint()
+
+
 def test_sparse_handler_gpu():
     _boot_joint_inproc()
     try:
