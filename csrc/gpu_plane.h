@@ -48,6 +48,11 @@ class GpuPlane : public DataPlane {
   // pooled events (shared with the server handlers)
   hipEvent_t GetEvent();
   void PutEvent(hipEvent_t ev);
+  // resolve a peer-pool global offset to a locally mapped pointer
+  // (handlers use this to write pull responses in place themselves)
+  char* PeerDst(int peer_id, uint64_t global_off, uint64_t len) {
+    return ResolvePeer(GetPeer(peer_id), global_off, len);
+  }
 
  private:
   struct Peer {

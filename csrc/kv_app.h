@@ -392,10 +392,11 @@ class KVServer : public SimpleApp {
     msg.meta.key = req.key;
     msg.meta.addr = req.addr;         // in-place pull destination (pool offset)
     msg.meta.option = req.option;
+    msg.meta.val_len = req.val_len;   // in-place responses keep the request's byte count
     msg.meta.src_dev = res.vals.device();
     msg.meta.dst_dev = req.dst_dev;
     if (req.pull && (!res.vals.empty() || !res.keys.empty())) {
-      msg.meta.val_len = static_cast<int64_t>(res.vals.nbytes());
+      if (!res.vals.empty()) msg.meta.val_len = static_cast<int64_t>(res.vals.nbytes());
       msg.AddData(res.keys);  // may be empty (sparse: meta.key suffices)
       msg.AddData(res.vals);
       if (!res.lens.empty()) msg.AddData(res.lens);

@@ -270,6 +270,43 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
     return;
   }
   hipStream_t stream = Stream(req.sender);
+  // multi-key pull with an in-place destination: batched copy of every
+  // store entry straight into the requester's mapped pool (no staging
+  // buffer, meta-only response)
+  if (n > 1 && (req.option & kOptPullAddr)) {
+    if (auto* plane = ThePlane(po_)) {
+      std::vector<kern::CopyDesc> descs;
+      SArray<int> lens2(n);
+      uint64_t off = 0;
+      bool ok = true;
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        for (size_t i = 0; i < n && ok; ++i) {
+          auto it = store_.find(kvs.keys[i]);
+          XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[i];
+          size_t len = it->second.buf.size();
+          char* dst = plane->PeerDst(req.sender, req.addr + off, len);
+          ok = dst != nullptr && len % 16 == 0 && off % 16 == 0;
+          if (ok) {
+            descs.push_back({dst, it->second.buf.data(), len});
+            lens2[i] = static_cast<int>(len / sizeof(float));
+            off += len;
+          }
+        }
+      }
+      if (ok) {
+        kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
+        KVMeta r2 = req;
+        r2.option |= kOptInPlace;
+        r2.val_len = static_cast<int64_t>(off);
+        KVPairs<float> res2;
+        res2.keys = kvs.keys;
+        res2.lens = lens2;
+        server->Response(r2, res2);  // meta+keys/lens only; plane defers on `stream`
+        return;
+      }
+    }
+  }
   KVPairs<float> res;
   res.keys = kvs.keys;
   SArray<int> lens(n);
