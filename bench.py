@@ -48,8 +48,10 @@ def parse_args():
     p.add_argument("--op", choices=["assign", "sum"], default="assign")
     p.add_argument("--batch-keys", action="store_true",
                    help="dense: one multi-key message per server per round")
-    p.add_argument("--overlap", action="store_true",
-                   help="dense: issue each key's pull right behind its push")
+    p.add_argument("--no-overlap", action="store_true",
+                   help="dense: separate push and pull phases (the reference "
+                        "PUSH_PULL loop issues ZPush+ZPull per key together, "
+                        "so overlap is the default)")
     p.add_argument("--hot-keys", type=int, default=8192, help="sparse keys per step")
     p.add_argument("--emb-rows", type=int, default=1 << 20)
     p.add_argument("--emb-width", type=int, default=64)
@@ -131,7 +133,7 @@ def main():
         msg_sizes = [size] * (args.keys_per_server * n)
         keys = [s * step_range + i for s in range(n) for i in range(args.keys_per_server)]
         model_name = "dense push+pull (ps-lite test_benchmark PUSH_PULL)"
-        overlap_pull = args.overlap
+        overlap_pull = not args.no_overlap and not args.cpu and not args.batch_keys
     elif args.mode == "rn50":
         buckets = resnet50_grad_buckets()
         if args.smoke:
@@ -335,7 +337,7 @@ def main():
                 "msg_bytes": max(msg_sizes) if msg_sizes else args.hot_keys * args.emb_width * 4,
                 "keys_per_server": args.keys_per_server,
                 "batch_keys": args.batch_keys,
-                "overlap": getattr(args, "overlap", False),
+                "overlap": overlap_pull,
                 "op": args.op,
                 "gbs_per_worker": round(gbs_per_worker, 3),
                 "p50_rtt_us": {k: round(v, 1) for k, v in rtts.items()},
