@@ -10,6 +10,7 @@
 #include <cstring>
 
 #include "hip_pool.h"
+#include "kernels.h"
 #include "postoffice.h"
 #include "wire.h"
 
@@ -354,7 +355,9 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     }
     hipStream_t stream = StreamForPeer(peer_node.id);
     XPS_HIP_CHECK(hipSetDevice(device_));
-    XPS_HIP_CHECK(hipMemcpyAsync(dst, vals.data(), vals.size(), hipMemcpyDefault, stream));
+    // copy KERNEL instead of hipMemcpyAsync: measured 5.2 vs 4.85 TB/s
+    // same-device, and it reads/writes hipIpc-mapped peer memory alike
+    kern::DenseAssign(dst, vals.data(), vals.size(), stream);
     Message meta_msg;
     meta_msg.meta = msg.meta;
     meta_msg.meta.option |= kOptInPlace;
