@@ -26,19 +26,13 @@ inline int GridFor(size_t work_items, int cap = 8192) {
   return static_cast<int>(g);
 }
 
-// clang ext_vector (not HIP_vector_type) so the nontemporal builtins apply
-using u32x4 = __attribute__((ext_vector_type(4))) unsigned int;
-
+// NOTE: nontemporal loads/stores were measured 3 % SLOWER end-to-end
+// here — the 256 MiB Infinity Cache serves the overlapped pull's re-read
+// of the just-written store, and nt stores bypass it. Keep plain float4.
 __global__ void assign_kernel(uint4* __restrict__ dst, const uint4* __restrict__ src, size_t n4) {
   size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
-  u32x4* __restrict__ d = reinterpret_cast<u32x4*>(dst);
-  const u32x4* __restrict__ s = reinterpret_cast<const u32x4*>(src);
-  // nontemporal: the payload is written once and read later by another
-  // op — skip cache pollution on both sides of the stream
-  for (; i < n4; i += stride) {
-    __builtin_nontemporal_store(__builtin_nontemporal_load(s + i), d + i);
-  }
+  for (; i < n4; i += stride) dst[i] = src[i];
 }
 
 __global__ void assign_tail_kernel(char* __restrict__ dst, const char* __restrict__ src,
