@@ -12,6 +12,11 @@ PYINC := $(shell python3 -m pybind11 --includes)
 TORCHLIB := $(shell python3 -c "import torch, os; print(os.path.join(os.path.dirname(torch.__file__), 'lib'))" 2>/dev/null)
 CXXFLAGS := -O3 -std=c++17 -fPIC -Wall -Wno-unused-function --offload-arch=$(ARCH) $(PYINC)
 LDFLAGS := -shared -fPIC
+# `make ASAN=1` — address-sanitized build (ps-lite Makefile:54-56 parity)
+ifeq ($(ASAN),1)
+CXXFLAGS += -fsanitize=address -fno-omit-frame-pointer -g
+LDFLAGS += -fsanitize=address
+endif
 ifneq ($(TORCHLIB),)
 LDFLAGS += -L$(TORCHLIB) -Wl,-rpath,$(TORCHLIB)
 endif
