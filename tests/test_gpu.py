@@ -82,6 +82,32 @@ def test_kernel_sparse_gather_scatter_vs_torch():
     assert torch.allclose(table.cpu(), ref2.cpu(), atol=1e-6)
 
 
+def test_ops_wrappers_vs_torch():
+    import torch
+
+    from ps_lite_amd import ops
+
+    torch.manual_seed(2)
+    a = torch.randn(1 << 16, device="cuda:0")
+    b = torch.randn(1 << 16, device="cuda:0")
+    ref = a + b
+    ops.dense_sum(a, b)
+    assert torch.allclose(a, ref)
+    c = torch.empty_like(a)
+    ops.dense_assign(c, a)
+    assert torch.equal(c, a)
+
+    table = torch.randn(1024, 32, device="cuda:0")
+    rows = torch.randperm(1024, device="cuda:0")[:100]
+    got = ops.sparse_gather(table, rows)
+    assert torch.allclose(got, table[rows])
+    grad = torch.randn(100, 32, device="cuda:0")
+    ref2 = table.clone()
+    ref2[rows] += grad
+    ops.sparse_scatter_add(table, rows, grad)
+    assert torch.allclose(table, ref2, atol=1e-6)
+
+
 def test_joint_push_pull_inproc_gpu():
     _boot_joint_inproc()
     try:
