@@ -113,10 +113,17 @@ bool ShmRing::Push(const void* payload, uint32_t len) {
   if (len > MaxPayload() || !mem_) return false;
   uint64_t pos = Hdr(mem_)->head.fetch_add(1, std::memory_order_relaxed);
   RingSlot* s = Slot(mem_, pos);
-  // wait for the consumer to free this slot (seq == pos)
+  // wait for the consumer to free this slot (seq == pos); bounded so a
+  // crashed consumer cannot wedge this producer forever (~10 s)
   int spins = 0;
   while (s->seq.load(std::memory_order_acquire) != pos) {
-    if (++spins > 1000) std::this_thread::yield();
+    if (++spins > 1000) {
+      std::this_thread::yield();
+      if (spins > 10 * 1000 * 1000) {
+        XPS_LOG(Warning) << "shm ring full for too long (consumer dead?); dropping message";
+        return false;
+      }
+    }
   }
   memcpy(s->payload, payload, len);
   s->len = len;

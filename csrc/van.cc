@@ -161,7 +161,6 @@ int64_t Van::Send(Message& msg) {
 
 int64_t Van::SendToNode(Message& msg, int id) {
   MaybeTrace(msg, /*recv=*/false);
-  if (resender_ && msg.meta.control.empty()) resender_->AddOutgoing(msg);
   if (id == my_node_.id) {
     // loopback without touching the wire
     Message copy = msg;
@@ -189,6 +188,9 @@ int64_t Van::SendToNode(Message& msg, int id) {
 }
 
 int64_t Van::SendOverTcp(Message& msg, int id) {
+  // retransmission covers the TCP path only: shm/xGMI plane messages are
+  // lossless by construction and their recv path sends no ACKs
+  if (resender_ && msg.meta.control.empty()) resender_->AddOutgoing(msg);
   auto conn = GetOrDial(id);
   if (!conn) {
     XPS_LOG(Warning) << "no route to node " << id << " from " << my_node_.id;
