@@ -198,6 +198,34 @@ def test_sparse_handler_gpu():
         _down_joint()
 
 
+def test_dp_gradsync_gpu():
+    """PSGradSync on GPU: reduce-mode handler, pool staging buffers."""
+    import torch
+
+    from ps_lite_amd.parallel.dp import PSGradSync
+
+    _boot_joint_inproc()
+    try:
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(mode="reduce")
+        worker = ps.KVWorker(0, 0)
+        torch.manual_seed(0)
+        model = torch.nn.Linear(64, 32).cuda()
+        sync = PSGradSync(ps, worker, model.parameters(), num_workers=1, device=0)
+        for step in range(3):
+            x = torch.randn(16, 64, device="cuda:0")
+            model.zero_grad()
+            model(x).sum().backward()
+            before = [p.grad.clone() for p in model.parameters()]
+            sync.allreduce()
+            torch.cuda.synchronize()
+            # single worker: the averaged grad equals the local grad
+            for b, p in zip(before, model.parameters()):
+                assert torch.allclose(b, p.grad, atol=1e-6)
+    finally:
+        _down_joint()
+
+
 # ------- cross-process hipIpc on a single GPU (config #2 layout x2) -------
 
 
