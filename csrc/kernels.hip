@@ -82,33 +82,40 @@ __global__ void sum_multi_f32(float4* __restrict__ dst, SrcList srcs, int nsrc, 
   }
 }
 
-// one wave per row-chunk: rows are contiguous float runs; row_len4 is the
-// row length in float4. Each block strides over rows; lanes stride the row.
+// flat indexing: thread i handles element i of the CONCATENATED rows
+// (row = i / row_len4), so narrow rows (e.g. 64 floats = 16 float4)
+// still use every lane — a row-per-block mapping left 94 % of the block
+// idle at width 64
 __global__ void gather_rows_f32(const float4* __restrict__ table, const uint64_t* __restrict__ rows,
                                 size_t nrows, size_t row_len4, float4* __restrict__ out,
                                 int shift, uint64_t base) {
-  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    const float4* src = table + ((rows[r] >> shift) - base) * row_len4;
-    float4* dst = out + r * row_len4;
-    for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) dst[c] = src[c];
+  size_t total = nrows * row_len4;
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (; i < total; i += stride) {
+    size_t r = i / row_len4;
+    size_t c = i - r * row_len4;
+    out[i] = table[((rows[r] >> shift) - base) * row_len4 + c];
   }
 }
 
 __global__ void scatter_add_rows_f32(float4* __restrict__ table, const uint64_t* __restrict__ rows,
                                      size_t nrows, size_t row_len4,
                                      const float4* __restrict__ src, int shift, uint64_t base) {
-  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float4* dst = table + ((rows[r] >> shift) - base) * row_len4;
-    const float4* s = src + r * row_len4;
-    for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) {
-      float4 d = dst[c];
-      float4 v = s[c];
-      d.x += v.x;
-      d.y += v.y;
-      d.z += v.z;
-      d.w += v.w;
-      dst[c] = d;
-    }
+  size_t total = nrows * row_len4;
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (; i < total; i += stride) {
+    size_t r = i / row_len4;
+    size_t c = i - r * row_len4;
+    float4* dst = table + ((rows[r] >> shift) - base) * row_len4 + c;
+    float4 d = *dst;
+    float4 v = src[i];
+    d.x += v.x;
+    d.y += v.y;
+    d.z += v.z;
+    d.w += v.w;
+    *dst = d;
   }
 }
 
@@ -127,10 +134,13 @@ __global__ void scatter_assign_rows_f32(float4* __restrict__ table,
                                         const uint64_t* __restrict__ rows, size_t nrows,
                                         size_t row_len4, const float4* __restrict__ src,
                                         int shift, uint64_t base) {
-  for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float4* dst = table + ((rows[r] >> shift) - base) * row_len4;
-    const float4* s = src + r * row_len4;
-    for (size_t c = threadIdx.x; c < row_len4; c += blockDim.x) dst[c] = s[c];
+  size_t total = nrows * row_len4;
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (; i < total; i += stride) {
+    size_t r = i / row_len4;
+    size_t c = i - r * row_len4;
+    table[((rows[r] >> shift) - base) * row_len4 + c] = src[i];
   }
 }
 
@@ -165,6 +175,7 @@ struct DescArray {
 
 // every block strides over every segment's 16B chunks (grid sized for
 // the concatenated total, so all segments together fill the 8 XCDs)
+// (segment loop per block; fine for <= kMaxBatch segments)
 __global__ void batched_assign_kernel(DescArray da) {
   for (int seg = 0; seg < da.n; ++seg) {
     uint4* dst = reinterpret_cast<uint4*>(da.d[seg].dst);
