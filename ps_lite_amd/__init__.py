@@ -55,6 +55,38 @@ from ._core import (  # noqa: F401
 __version__ = "0.1.0"
 
 
+class _CudaArrayView:
+    """Exports a PoolBuffer as __cuda_array_interface__ so torch can wrap it."""
+
+    def __init__(self, buf, shape, typestr, itemsize):
+        self._buf = buf
+        self.__cuda_array_interface__ = {
+            "shape": tuple(shape),
+            "typestr": typestr,
+            "data": (buf.ptr, False),
+            "version": 2,
+            "strides": None,
+        }
+
+
+def pool_tensor(shape, dtype="float32"):
+    """A torch CUDA tensor backed by HBM-pool memory — ZPush/ZPull on its
+    data_ptr() ride the zero-copy hipIpc fast path (pool membership is
+    what the plane checks). The pool allocation is pinned to the tensor.
+    """
+    import numpy as np
+    import torch
+
+    shape = (shape,) if isinstance(shape, int) else tuple(shape)
+    np_dtype = np.dtype(dtype)
+    nbytes = int(np.prod(shape)) * np_dtype.itemsize
+    buf = pool_alloc(nbytes)
+    view = _CudaArrayView(buf, shape, np_dtype.str, np_dtype.itemsize)
+    t = torch.as_tensor(view, device="cuda")
+    t._xps_pool_buf = buf  # keep the pool allocation alive with the tensor
+    return t
+
+
 def setup_env(num_workers, num_servers, root_uri="127.0.0.1", root_port=9100, **extra):
     """Set the DMLC_* environment both for this process and for children."""
     env = {

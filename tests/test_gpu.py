@@ -198,6 +198,29 @@ def test_sparse_handler_gpu():
         _down_joint()
 
 
+def test_pool_tensor_fast_path():
+    """torch tensors backed by pool memory ride the zero-copy plane."""
+    import torch
+
+    _boot_joint_inproc()
+    try:
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(mode="assign")
+        worker = ps.KVWorker(0, 0)
+        n = 1 << 16
+        src = ps.pool_tensor(n)
+        dst = ps.pool_tensor(n)
+        src.copy_(torch.randn(n, device="cuda:0"))
+        torch.cuda.synchronize()
+        keys = np.array([77], dtype=np.uint64)
+        lens = np.array([n], dtype=np.int32)
+        worker.wait(worker.zpush_ptr(keys, src.data_ptr(), n * 4, 0, lens, cmd=1))
+        worker.wait(worker.zpull_ptr(keys, dst.data_ptr(), n * 4, 0, lens))
+        assert torch.allclose(dst, src)
+    finally:
+        _down_joint()
+
+
 def test_dp_gradsync_gpu():
     """PSGradSync on GPU: reduce-mode handler, pool staging buffers."""
     import torch
