@@ -9,7 +9,10 @@
 // delivered in place carry kOptInPlace and skip the worker-side merge.
 #pragma once
 
+#include <unistd.h>
+
 #include <algorithm>
+#include <atomic>
 #include <functional>
 #include <unordered_map>
 #include <vector>
@@ -377,7 +380,10 @@ class KVServer : public SimpleApp {
     obj_.reset(new Customer(app_id, app_id, [this](const Message& m) { Process(m); }, po_));
   }
 
-  void set_request_handle(ReqHandle h) { request_handle2_ = std::move(h); }
+  void set_request_handle(ReqHandle h) {
+    request_handle2_ = std::move(h);
+    handle_set_.store(true, std::memory_order_release);
+  }
 
   void Response(const KVMeta& req, const KVPairs<V>& res = KVPairs<V>()) {
     Message msg;
@@ -432,11 +438,19 @@ class KVServer : public SimpleApp {
     } else if (msg.meta.key || msg.meta.val_len) {
       kvs.keys = SArray<Key>({msg.meta.key});
     }
-    XPS_CHECK(request_handle2_) << "KVServer has no request handle";
+    // a worker's first request can arrive between construction and
+    // set_request_handle — wait for the app to install it
+    if (!handle_set_.load(std::memory_order_acquire)) {
+      for (int i = 0; i < 30000 && !handle_set_.load(std::memory_order_acquire); ++i) {
+        usleep(1000);
+      }
+    }
+    XPS_CHECK(handle_set_.load()) << "KVServer has no request handle";
     request_handle2_(meta, kvs, this);
   }
 
   ReqHandle request_handle2_;
+  std::atomic<bool> handle_set_{false};
 };
 
 // Default CPU handle: store[key] op= vals; pull echoes the store
