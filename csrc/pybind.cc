@@ -421,7 +421,7 @@ PYBIND11_MODULE(_core, m) {
     XPS_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming) == hipSuccess);
     XPS_CHECK(hipEventRecord(ev, reinterpret_cast<hipStream_t>(stream)) == hipSuccess);
     while (hipEventQuery(ev) != hipSuccess) {}
-    hipEventDestroy(ev);
+    (void)hipEventDestroy(ev);
   });
   m.def("ipc_open", [](py::bytes handle) {
     std::string h = handle;
