@@ -44,7 +44,11 @@ def parse_args():
     p.add_argument("--cpu", action="store_true", help="config #1: TCP van, host buffers")
     p.add_argument("--size-mb", type=float, default=64.0,
                    help="dense message size per key (headline: 64 MiB)")
-    p.add_argument("--keys-per-server", type=int, default=8)
+    p.add_argument("--keys-per-server", type=int, default=8,
+                   help="with --gpus 8 this is exact; the TOTAL key count is "
+                        "keys_per_server*8 and stays FIXED as N varies (the "
+                        "model does not grow with the cluster -> honest weak "
+                        "scaling of per-worker work)")
     p.add_argument("--op", choices=["assign", "sum"], default="assign")
     p.add_argument("--batch-keys", action="store_true",
                    help="dense: one multi-key message per server per round")
@@ -130,8 +134,13 @@ def main():
     cmd = 2 if args.op == "sum" else 1
     if args.mode == "dense":
         size = int(args.size_mb * (1 << 20))
-        msg_sizes = [size] * (args.keys_per_server * n)
-        keys = [s * step_range + i for s in range(n) for i in range(args.keys_per_server)]
+        # FIXED model: keys_per_server*8 keys total, sharded over the n
+        # servers (each worker pushes+pulls the whole model every step,
+        # BytePS semantics) — per-worker work does not grow with n
+        total_keys = args.keys_per_server * 8
+        assert total_keys % n == 0
+        msg_sizes = [size] * total_keys
+        keys = [(j % n) * step_range + (j // n) for j in range(total_keys)]
         model_name = "dense push+pull (ps-lite test_benchmark PUSH_PULL)"
         overlap_pull = not args.no_overlap and not args.cpu and not args.batch_keys
     elif args.mode == "rn50":
@@ -235,8 +244,9 @@ def main():
             pull_ptrs = [a.ctypes.data for a in host_pull]
         uniform = len(set(msg_sizes)) == 1
         if batch:
-            k = args.keys_per_server
-            srv_keys = [keys_np[s * k:(s + 1) * k] for s in range(n)]
+            k = len(keys) // n
+            srv_keys = [np.sort(keys_np[keys_np // np.uint64(step_range) == s])
+                        for s in range(n)]
             blens = np.full(k, msg_sizes[0] // 4, dtype=np.int32)
             bpush = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
             bpull = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
