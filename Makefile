@@ -22,7 +22,7 @@ LDFLAGS += -L$(TORCHLIB) -Wl,-rpath,$(TORCHLIB)
 endif
 
 SRCS := csrc/env.cc csrc/wire.cc csrc/tcp.cc csrc/van.cc csrc/postoffice.cc \
-        csrc/customer.cc csrc/resender.cc csrc/hip_util.cc csrc/hip_pool.cc \
+        csrc/customer.cc csrc/resender.cc csrc/hip_util.cc csrc/hip_pool.cc csrc/host_pool.cc \
         csrc/gpu_plane.cc csrc/shm_ring.cc csrc/server_handlers.cc csrc/ps.cc csrc/pybind.cc
 HIPSRCS := $(wildcard csrc/*.hip)
 OBJS := $(SRCS:%.cc=build/%.o) $(HIPSRCS:%.hip=build/%.hip.o)

@@ -225,23 +225,16 @@ def main():
         keys_np = np.array(keys, dtype=np.uint64)
         batch = args.batch_keys and args.mode == "dense" and device >= 0
         push_bufs, pull_bufs = [], []
-        host_push, host_pull = [], []
+        alloc = ps.pool_alloc if device >= 0 else ps.host_alloc
         for sz in msg_sizes:
-            if device >= 0:
-                b = ps.pool_alloc(sz)
-                b.copy_from(rng.standard_normal(sz // 4).astype(np.float32))
-                push_bufs.append(b)
-                pull_bufs.append(ps.pool_alloc(sz))
-            else:
-                a = rng.standard_normal(sz // 4).astype(np.float32)
-                host_push.append(a)
-                host_pull.append(np.zeros(sz // 4, dtype=np.float32))
-        if device >= 0:
-            push_ptrs = [b.ptr for b in push_bufs]
-            pull_ptrs = [b.ptr for b in pull_bufs]
-        else:
-            push_ptrs = [a.ctypes.data for a in host_push]
-            pull_ptrs = [a.ctypes.data for a in host_pull]
+            # host buffers come from the shm pool so the CPU config rides
+            # the same-host zero-copy plane (not TCP)
+            b = alloc(sz)
+            b.copy_from(rng.standard_normal(sz // 4).astype(np.float32))
+            push_bufs.append(b)
+            pull_bufs.append(alloc(sz))
+        push_ptrs = [b.ptr for b in push_bufs]
+        pull_ptrs = [b.ptr for b in pull_bufs]
         uniform = len(set(msg_sizes)) == 1
         if batch:
             k = len(keys) // n

@@ -10,11 +10,14 @@
 #include <cstring>
 
 #include "hip_pool.h"
+#include "host_pool.h"
 #include "kernels.h"
 #include "postoffice.h"
 #include "wire.h"
 
 namespace xps {
+
+static const size_t kInlineMax = 4096;  // host blobs above this ride the shm pool
 
 #define XPS_HIP_CHECK(cmd)                                                            \
   do {                                                                                \
@@ -85,7 +88,7 @@ GpuPlane::~GpuPlane() { Stop(); }
 void GpuPlane::FillSelf(Node* self) {
   self->dev_id = device_;
   auto* pool = HbmPool::Get();
-  if (pool->initialized()) {
+  if (device_ >= 0 && pool->initialized()) {
     self->pool_capacity = pool->capacity();
     self->pool_slab_bytes = pool->slab_bytes();
     self->pool_handles.resize(pool->slab_count());
@@ -93,6 +96,12 @@ void GpuPlane::FillSelf(Node* self) {
       memcpy(self->pool_handles[i].data(), pool->slab_handle(i), kIpcHandleBytes);
     }
   }
+  // host-shm arena: zero-copy HOST payloads for same-host peers (one per
+  // process; both nodes of a joint process advertise the same uid)
+  auto* hpool = HostShmPool::Get();
+  hpool->Init(self->shm_uid);
+  self->host_pool_uid = hpool->uid();
+  self->host_pool_capacity = hpool->capacity();
   if (!started_) {
     auto* env = Environment::Get();
     if (!env->GetInt("XPS_PROBE_NO_RING", 0)) {
@@ -102,7 +111,7 @@ void GpuPlane::FillSelf(Node* self) {
     if (!env->GetInt("XPS_PROBE_NO_POLL", 0)) {
       poll_thread_ = std::thread([this] { RingPollLoop(); });
     }
-    if (!env->GetInt("XPS_PROBE_NO_COMP", 0)) {
+    if (device_ >= 0 && !env->GetInt("XPS_PROBE_NO_COMP", 0)) {
       comp_thread_ = std::thread([this] { CompletionLoop(); });
     }
   }
@@ -128,6 +137,7 @@ void GpuPlane::Stop() {
     pending_.clear();
   }
   in_ring_.CloseAndUnlink();
+  HostShmPool::Get()->Unlink();  // idempotent; mappings stay valid
   std::lock_guard<std::mutex> lk(ev_mu_);
   for (auto ev : event_pool_) (void)hipEventDestroy(ev);
   event_pool_.clear();
@@ -292,10 +302,17 @@ bool GpuPlane::CanSend(const Message& msg, const Node& peer) {
       uint64_t off;
       if (!HbmPool::Get()->OffsetOf(d.data(), &off)) return false;  // not our pool -> TCP
       est += 24;
-      // device vals in a pull response need an in-place destination
+      // device vals in a pull response need an HBM in-place destination
       if (!msg.meta.request && msg.meta.pull && i == 1) {
-        if (!(msg.meta.option & kOptPullAddr) || peer.pool_capacity == 0) return false;
+        if (!(msg.meta.option & kOptPullAddr) || (msg.meta.option & kOptHostAddr) ||
+            peer.pool_capacity == 0) {
+          return false;
+        }
       }
+    } else if (d.size() > kInlineMax) {
+      uint64_t off;
+      if (!HostShmPool::Get()->OffsetOf(d.data(), &off)) return false;  // big + unpooled -> TCP
+      est += 24;
     } else {
       est += 24 + d.size();
     }
@@ -314,10 +331,16 @@ bool GpuPlane::Serialize(const Message& msg, const std::vector<char>& by_ref, st
   w.U8(static_cast<uint8_t>(msg.data.size()));
   for (size_t i = 0; i < msg.data.size(); ++i) {
     const auto& d = msg.data[i];
-    if (by_ref[i]) {
+    if (by_ref[i] == 1) {  // HBM pool ref
       uint64_t off = 0;
       XPS_CHECK(HbmPool::Get()->OffsetOf(d.data(), &off));
       w.U8(1);
+      w.U64(off);
+      w.U64(d.size());
+    } else if (by_ref[i] == 2) {  // host shm pool ref
+      uint64_t off = 0;
+      XPS_CHECK(HostShmPool::Get()->OffsetOf(d.data(), &off));
+      w.U8(2);
       w.U64(off);
       w.U64(d.size());
     } else {
@@ -340,8 +363,39 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   if (!EnsureRing(p)) {
     // TCP fallback for a response must not outrun handler kernels still
     // running on this peer's stream (the worker may reuse buffers on ack)
-    if (response) (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
+    if (response && device_ >= 0) (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
     return -1;
+  }
+
+  // ---- pull response with HOST vals + host destination: memcpy now --
+  if (response && msg.meta.pull && msg.data.size() > 1 && !msg.data[1].on_device() &&
+      msg.data[1].size() > 0 && (msg.meta.option & kOptPullAddr) &&
+      (msg.meta.option & kOptHostAddr)) {
+    void* base = HostShmPool::MapPeer(p->node.host_pool_uid, p->node.host_pool_capacity);
+    SArray<char> vals = msg.data[1];
+    if (base && msg.meta.addr + vals.size() <= p->node.host_pool_capacity) {
+      memcpy(static_cast<char*>(base) + msg.meta.addr, vals.data(), vals.size());
+      Message meta_msg;
+      meta_msg.meta = msg.meta;
+      meta_msg.meta.option |= kOptInPlace;
+      meta_msg.meta.val_len = static_cast<int64_t>(vals.size());
+      for (size_t i = 0; i < msg.data.size(); ++i) {
+        if (i == 1 || msg.data[i].on_device() || msg.data[i].size() > kInlineMax) continue;
+        meta_msg.data.push_back(msg.data[i]);
+        meta_msg.meta.data_type.push_back(msg.meta.data_type[i]);
+      }
+      std::vector<char> br(meta_msg.data.size(), 0);
+      std::string payload;
+      if (!Serialize(meta_msg, br, &payload)) {
+        meta_msg.data.clear();
+        meta_msg.meta.data_type.clear();
+        XPS_CHECK(Serialize(meta_msg, {}, &payload));
+      }
+      int64_t bytes = static_cast<int64_t>(vals.size() + payload.size());
+      if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
+      return bytes;
+    }
+    return -1;  // cannot map the destination: TCP fallback
   }
 
   // ---- pull response with device vals: one-sided xGMI write ----------
@@ -385,7 +439,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     return bytes;
   }
 
-  // ---- general path: inline blobs + by-ref device blobs --------------
+  // ---- general path: inline blobs + by-ref device/host-pool blobs ----
   std::vector<char> by_ref(msg.data.size(), 0);
   int64_t ref_bytes = 0;
   for (size_t i = 0; i < msg.data.size(); ++i) {
@@ -393,19 +447,24 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
       by_ref[i] = 1;
       ref_bytes += msg.data[i].size();
       if (i == 1) msg.meta.option |= kOptValsByRef;
+    } else if (msg.data[i].size() > kInlineMax) {
+      by_ref[i] = 2;  // CanSend verified host-pool membership
+      ref_bytes += msg.data[i].size();
     }
   }
   std::string payload;
   if (!Serialize(msg, by_ref, &payload)) {
-    if (response) (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
+    if (response && device_ >= 0) (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
     return -1;
   }
   int64_t bytes = static_cast<int64_t>(payload.size()) + ref_bytes;
-  if (response) {
+  if (response && device_ >= 0) {
     // order behind any handler kernels on this peer's stream
     Message keepalive = msg;
     DeferSendInternal(p, peer_node.id, std::move(keepalive), std::move(payload), bytes);
   } else {
+    // requests — and every send of a host-only plane — go out now (host
+    // responses were produced synchronously; nothing to wait for)
     if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
     XPS_VLOG(3) << "plane send done -> " << peer_node.id;
   }
@@ -414,6 +473,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
 
 void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::string payload,
                                  int64_t bytes) {
+  XPS_CHECK_GE(device_, 0) << "deferred sends are a GPU-plane feature";
   hipStream_t stream = StreamForPeer(peer_id);
   hipEvent_t ev = GetEvent();
   XPS_HIP_CHECK(hipSetDevice(device_));
@@ -468,7 +528,7 @@ void GpuPlane::CompletionLoop() {
 }
 
 void GpuPlane::RingPollLoop() {
-  XPS_HIP_CHECK(hipSetDevice(device_));
+  if (device_ >= 0) XPS_HIP_CHECK(hipSetDevice(device_));
   prctl(PR_SET_TIMERSLACK, 1000);
   const int kSpin = Environment::Get()->GetInt("XPS_POLL_SPIN", 200000);
   std::vector<char> buf(ShmRing::MaxPayload());
@@ -498,7 +558,7 @@ void GpuPlane::RingPollLoop() {
         SArray<char> d(len);
         r.Raw(d.data(), len);
         msg.data.push_back(d);
-      } else {
+      } else if (kind == 1) {
         uint64_t off = r.U64();
         uint64_t len = r.U64();
         Peer* sender = GetPeer(msg.meta.sender);
@@ -510,6 +570,21 @@ void GpuPlane::RingPollLoop() {
           break;
         }
         msg.data.push_back(SArray<char>(ptr, len, device_));
+        ref_bytes += len;
+      } else {  // kind 2: host shm pool
+        uint64_t off = r.U64();
+        uint64_t len = r.U64();
+        Peer* sender = GetPeer(msg.meta.sender);
+        void* base = sender ? HostShmPool::MapPeer(sender->node.host_pool_uid,
+                                                   sender->node.host_pool_capacity)
+                            : nullptr;
+        if (!base || off + len > sender->node.host_pool_capacity) {
+          XPS_LOG(Warning) << "dropping host-ref blob: sender host pool not mapped (from "
+                           << msg.meta.sender << ")";
+          ok = false;
+          break;
+        }
+        msg.data.push_back(SArray<char>(static_cast<char*>(base) + off, len, kCPU));
         ref_bytes += len;
       }
     }
@@ -525,7 +600,7 @@ void GpuPlane::RingPollLoop() {
 }
 
 std::shared_ptr<DataPlane> CreateGpuPlane(Postoffice* po, int device) {
-  if (device < 0) return nullptr;
+  // device < 0 builds the host-only variant (shm rings + host pool)
   return std::make_shared<GpuPlane>(po, device);
 }
 

@@ -19,9 +19,15 @@
 
 #include "hip_pool.h"
 #include "hip_util.h"
+#include "host_pool.h"
 #include "simple_app.h"
 
 namespace xps {
+
+// worker-side cmd values (meta.head) understood by the built-in handlers
+static const int kCmdDefault = 0;  // handler's configured default op
+static const int kCmdAssign = 1;
+static const int kCmdSum = 2;
 
 template <typename V>
 struct KVPairs {
@@ -222,11 +228,16 @@ class KVWorker : public SimpleApp {
       } else {
         // pull request: keys (+lens geometry) only; advertise destination
         msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
-        if (!s.vals.empty() && s.vals.on_device()) {
+        if (!s.vals.empty()) {
           uint64_t off = 0;
-          if (HbmPool::Get()->OffsetOf(s.vals.data(), &off)) {
+          if (s.vals.on_device()) {
+            if (HbmPool::Get()->OffsetOf(s.vals.data(), &off)) {
+              msg.meta.addr = off;
+              msg.meta.option |= kOptPullAddr;  // in-place HBM response
+            }
+          } else if (HostShmPool::Get()->OffsetOf(s.vals.data(), &off)) {
             msg.meta.addr = off;
-            msg.meta.option |= kOptPullAddr;  // server may write the response in place
+            msg.meta.option |= kOptPullAddr | kOptHostAddr;  // in-place host response
           }
         }
         if (!s.lens.empty()) msg.AddData(s.lens);
@@ -453,8 +464,9 @@ class KVServer : public SimpleApp {
   std::atomic<bool> handle_set_{false};
 };
 
-// Default CPU handle: store[key] op= vals; pull echoes the store
-// (ps-lite kv_app.h:431-452 behavior).
+// Default CPU handle: store[key] op= vals (sum by default, memcpy for
+// cmd=kCmdAssign); pull echoes the store (ps-lite kv_app.h:431-452
+// behavior, plus the assign op the EmptyHandler benchmark semantics use).
 template <typename V>
 struct KVServerDefaultHandle {
   void operator()(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
@@ -467,23 +479,38 @@ struct KVServerDefaultHandle {
         size_t len = kvs.lens.empty() ? kvs.vals.size() / n : kvs.lens[i];
         auto& entry = store[kvs.keys[i]];
         if (entry.size() < len) entry.resize(len, V(0));
-        for (size_t j = 0; j < len; ++j) entry[j] += kvs.vals[off + j];
+        if (req.cmd == kCmdAssign) {
+          memcpy(entry.data(), kvs.vals.data() + off, len * sizeof(V));
+        } else {
+          V* __restrict__ e = entry.data();
+          const V* __restrict__ v = kvs.vals.data() + off;
+          for (size_t j = 0; j < len; ++j) e[j] += v[j];
+        }
         off += len;
       }
     } else if (req.pull) {
       res.keys = kvs.keys;
-      size_t total = 0;
-      for (size_t i = 0; i < n; ++i) total += store[kvs.keys[i]].size();
-      res.vals.Resize(total);
-      SArray<int> lens(n);
-      size_t off = 0;
-      for (size_t i = 0; i < n; ++i) {
-        auto& entry = store[kvs.keys[i]];
-        std::copy(entry.begin(), entry.end(), res.vals.data() + off);
-        off += entry.size();
-        lens[i] = static_cast<int>(entry.size());
+      if (n == 1) {
+        // zero-copy view of the store entry (stable: map nodes don't move)
+        auto& entry = store[kvs.keys[0]];
+        res.vals = SArray<V>(entry.data(), entry.size(), kCPU);
+        SArray<int> lens(1);
+        lens[0] = static_cast<int>(entry.size());
+        res.lens = lens;
+      } else {
+        size_t total = 0;
+        for (size_t i = 0; i < n; ++i) total += store[kvs.keys[i]].size();
+        res.vals.Resize(total);
+        SArray<int> lens(n);
+        size_t off = 0;
+        for (size_t i = 0; i < n; ++i) {
+          auto& entry = store[kvs.keys[i]];
+          std::copy(entry.begin(), entry.end(), res.vals.data() + off);
+          off += entry.size();
+          lens[i] = static_cast<int>(entry.size());
+        }
+        res.lens = lens;
       }
-      res.lens = lens;
     }
     server->Response(req, res);
   }

@@ -41,6 +41,9 @@ struct Node {
   uint64_t pool_slab_bytes = 0;
   std::vector<std::array<char, kIpcHandleBytes>> pool_handles;  // per slab
   uint64_t shm_uid = 0;        // uid for naming this node's shm data-plane segment
+  // host-shm arena (zero-copy HOST payloads on the same host)
+  uint64_t host_pool_uid = 0;  // names the segment (per process, shared by its nodes)
+  uint64_t host_pool_capacity = 0;
 
   std::string DebugString() const {
     std::ostringstream os;
@@ -74,6 +77,7 @@ struct Control {
 static const int kOptInPlace = 1;   // pull response already written into dst buffer
 static const int kOptValsByRef = 2; // vals travel as {pool offset, len}, not bytes
 static const int kOptPullAddr = 4;  // meta.addr holds a valid pull-destination offset
+static const int kOptHostAddr = 8;  // meta.addr is a HOST-shm-pool offset (else HBM pool)
 
 enum DataType : int { kChar = 0, kInt32, kInt64, kUint64, kFloat32, kFloat64, kUint8 };
 inline size_t DataTypeSize(int t) {

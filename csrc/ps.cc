@@ -36,10 +36,14 @@ static void StartOne(const std::string& role, int customer_id, bool do_barrier, 
                            idx);
   }
   po->EnsureVan();
-  if (device >= 0 && role != "scheduler") {
-    HbmPool::Get()->Init(device);
-    auto plane = CreateGpuPlane(po, device);
-    if (plane) po->van()->SetDataPlane(plane);
+  if (role != "scheduler") {
+    if (device >= 0) HbmPool::Get()->Init(device);
+    // device < 0 still gets the plane: same-host HOST payloads ride the
+    // shm rings + host pool zero-copy (XPS_HOST_PLANE=0 forces pure TCP)
+    if (device >= 0 || Environment::Get()->GetInt("XPS_HOST_PLANE", 1)) {
+      auto plane = CreateGpuPlane(po, device);
+      if (plane) po->van()->SetDataPlane(plane);
+    }
   }
   po->Start(customer_id, do_barrier);
 }

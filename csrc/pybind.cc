@@ -7,6 +7,7 @@
 #include "gpu_plane.h"
 #include "hip_pool.h"
 #include "hip_util.h"
+#include "host_pool.h"
 #include "kernels.h"
 #include "kv_app.h"
 #include "kv_utils.h"
@@ -29,12 +30,20 @@ struct PoolBuffer {
     py::buffer_info info = src.request();
     size_t n = static_cast<size_t>(info.size) * info.itemsize;
     XPS_CHECK_LE(n, arr.size());
-    gpu::CopyHostToDevice(arr.data(), info.ptr, n, arr.device());
+    if (arr.on_device()) {
+      gpu::CopyHostToDevice(arr.data(), info.ptr, n, arr.device());
+    } else {
+      memcpy(arr.data(), info.ptr, n);
+    }
   }
   py::array_t<float> to_numpy_f32() const {
     size_t n = arr.size() / sizeof(float);
     py::array_t<float> out(n);
-    gpu::CopyDeviceToHost(out.mutable_data(), arr.data(), n * sizeof(float));
+    if (arr.on_device()) {
+      gpu::CopyDeviceToHost(out.mutable_data(), arr.data(), n * sizeof(float));
+    } else {
+      memcpy(out.mutable_data(), arr.data(), n * sizeof(float));
+    }
     return out;
   }
 };
@@ -443,6 +452,12 @@ PYBIND11_MODULE(_core, m) {
   m.def("pool_alloc", [](size_t nbytes) {
     PoolBuffer b;
     b.arr = HbmPool::Get()->AllocArray(nbytes);
+    return b;
+  });
+  // host-shm pool buffer: same-host zero-copy for CPU payloads
+  m.def("host_alloc", [](size_t nbytes) {
+    PoolBuffer b;
+    b.arr = HostShmPool::Get()->AllocArray(nbytes);
     return b;
   });
 

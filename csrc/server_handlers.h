@@ -20,11 +20,6 @@
 
 namespace xps {
 
-// worker-side cmd values (meta.head)
-static const int kCmdDefault = 0;  // handler's configured default op
-static const int kCmdAssign = 1;
-static const int kCmdSum = 2;
-
 // handler modes
 enum class DenseMode {
   kAssign,  // push overwrites the store (EmptyHandler / pure-goodput)

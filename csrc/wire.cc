@@ -18,6 +18,8 @@ static void PackNode(const Node& n, ByteWriter* w) {
   w->I32(static_cast<int32_t>(n.pool_handles.size()));
   for (auto& h : n.pool_handles) w->Raw(h.data(), kIpcHandleBytes);
   w->U64(n.shm_uid);
+  w->U64(n.host_pool_uid);
+  w->U64(n.host_pool_capacity);
 }
 
 static void UnpackNode(ByteReader* r, Node* n) {
@@ -35,6 +37,8 @@ static void UnpackNode(ByteReader* r, Node* n) {
   n->pool_handles.resize(nh);
   for (int i = 0; i < nh; ++i) r->Raw(n->pool_handles[i].data(), kIpcHandleBytes);
   n->shm_uid = r->U64();
+  n->host_pool_uid = r->U64();
+  n->host_pool_capacity = r->U64();
 }
 
 void PackMeta(const Meta& m, std::string* out) {

@@ -39,6 +39,7 @@ from ._core import (  # noqa: F401
     device_sync,
     finalize,
     gpu_count,
+    host_alloc,
     init_env,
     my_rank,
     node_id,
