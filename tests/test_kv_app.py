@@ -156,6 +156,30 @@ def test_multiprocess_joint_mode():
         assert np.allclose(np.array(out), expect)
 
 
+def _host_pool_worker(ps_mod, rank):
+    # large host payloads from the host-shm pool: by-ref push + in-place
+    # pull across BOTH servers (exercises per-slice addr offsets)
+    worker = ps_mod.KVWorker(0, 0)
+    n = 1 << 16  # 256 KiB per key, well above the inline threshold
+    step = (1 << 64) // 2
+    keys = np.array([9, step + 9], dtype=np.uint64)
+    lens = np.array([n, n], dtype=np.int32)
+    src = ps_mod.host_alloc(2 * n * 4)
+    dst = ps_mod.host_alloc(2 * n * 4)
+    vals = np.concatenate([np.full(n, 3.0, dtype=np.float32),
+                           np.full(n, 5.0, dtype=np.float32)])
+    src.copy_from(vals)
+    worker.wait(worker.zpush_ptr(keys, src.ptr, 2 * n * 4, -1, lens, cmd=1))
+    worker.wait(worker.zpull_ptr(keys, dst.ptr, 2 * n * 4, -1, lens, cmd=1))
+    out = dst.to_numpy_f32()
+    return [float(out[0]), float(out[n - 1]), float(out[n]), float(out[-1])]
+
+
+def test_host_pool_zero_copy_two_servers():
+    results = launch_local(1, 2, _host_pool_worker, timeout=180)
+    assert results[0] == [3.0, 3.0, 5.0, 5.0], results
+
+
 def test_resend_with_drop():
     env = {"PS_RESEND": "1", "PS_RESEND_TIMEOUT": "200", "PS_DROP_MSG": "10"}
     results = launch_local(1, 1, _worker_single, env_extra=env, timeout=180)
