@@ -45,4 +45,12 @@ $(TARGET): $(OBJS)
 clean:
 	rm -rf build $(TARGET)
 
-.PHONY: all clean
+# pure-C++ benchmark (no Python): build/bench_kv
+CORE_OBJS := $(filter-out build/csrc/pybind.o,$(OBJS))
+build/bench_kv.o: examples/cpp/bench_kv.cc csrc/*.h
+	$(HIPCC) $(CXXFLAGS) -c examples/cpp/bench_kv.cc -o build/bench_kv.o
+cppbench: $(CORE_OBJS) build/bench_kv.o
+	$(HIPCC) -O3 --offload-arch=$(ARCH) build/bench_kv.o $(CORE_OBJS) \
+	    -o build/bench_kv -lpthread
+
+.PHONY: all clean cppbench

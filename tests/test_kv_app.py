@@ -181,6 +181,9 @@ def test_host_pool_zero_copy_two_servers():
 
 
 def test_resend_with_drop():
-    env = {"PS_RESEND": "1", "PS_RESEND_TIMEOUT": "200", "PS_DROP_MSG": "10"}
+    # force pure TCP: drop injection + retransmission are TCP-path
+    # features (the shm plane is lossless and would bypass both)
+    env = {"PS_RESEND": "1", "PS_RESEND_TIMEOUT": "200", "PS_DROP_MSG": "10",
+           "XPS_HOST_PLANE": "0"}
     results = launch_local(1, 1, _worker_single, env_extra=env, timeout=180)
     assert np.allclose(np.array(results[0]), np.ones(64))
