@@ -15,6 +15,7 @@
 #include "server_handlers.h"
 #include "simple_app.h"
 #include "van.h"
+#include "wire.h"
 
 namespace py = pybind11;
 using namespace xps;
@@ -490,6 +491,58 @@ PYBIND11_MODULE(_core, m) {
       .def("save_checkpoint", &PyKVServer::SaveCheckpoint)
       .def("load_checkpoint", &PyKVServer::LoadCheckpoint)
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
+
+  // wire-format roundtrip (unit-test hook)
+  m.def("_test_meta_roundtrip", []() {
+    Meta m;
+    m.app_id = 7;
+    m.customer_id = 3;
+    m.timestamp = 42;
+    m.sender = 9;
+    m.recver = 8;
+    m.request = true;
+    m.push = true;
+    m.head = 2;
+    m.body = "hello\x00world";
+    m.key = 0xDEADBEEFCAFEull;
+    m.addr = 1234567;
+    m.val_len = 1 << 20;
+    m.option = kOptPullAddr | kOptHostAddr;
+    m.data_type = {kUint64, kFloat32, kInt32};
+    m.control.cmd = Control::ADD_NODE;
+    Node n;
+    n.role = Node::WORKER;
+    n.id = 11;
+    n.hostname = "10.0.0.5";
+    n.port = 1234;
+    n.shm_uid = 0x1122334455667788ull;
+    n.host_pool_uid = 77;
+    n.host_pool_capacity = 1 << 30;
+    n.pool_capacity = 2ull << 30;
+    n.pool_slab_bytes = 1ull << 30;
+    n.pool_handles.resize(2);
+    n.pool_handles[0][0] = 'x';
+    n.pool_handles[1][63] = 'y';
+    m.control.node.push_back(n);
+    std::string buf;
+    PackMeta(m, &buf);
+    Meta out;
+    UnpackMeta(buf.data(), buf.size(), &out);
+    bool ok = out.app_id == m.app_id && out.customer_id == m.customer_id &&
+              out.timestamp == m.timestamp && out.sender == m.sender &&
+              out.recver == m.recver && out.request == m.request && out.push == m.push &&
+              out.head == m.head && out.body == m.body && out.key == m.key &&
+              out.addr == m.addr && out.val_len == m.val_len && out.option == m.option &&
+              out.data_type == m.data_type && out.control.cmd == m.control.cmd &&
+              out.control.node.size() == 1;
+    auto& on = out.control.node[0];
+    ok = ok && on.role == n.role && on.id == n.id && on.hostname == n.hostname &&
+         on.port == n.port && on.shm_uid == n.shm_uid && on.host_pool_uid == n.host_pool_uid &&
+         on.host_pool_capacity == n.host_pool_capacity && on.pool_capacity == n.pool_capacity &&
+         on.pool_slab_bytes == n.pool_slab_bytes && on.pool_handles.size() == 2 &&
+         on.pool_handles[0][0] == 'x' && on.pool_handles[1][63] == 'y';
+    return ok;
+  });
 
   // utility parity (ps-lite parallel_kv_match.h / parallel_sort.h)
   m.def("parallel_ordered_match",

@@ -38,3 +38,7 @@ def test_parallel_sort():
     keys = rng.integers(0, 1 << 62, size=100_000, dtype=np.uint64)
     out = ps._core.parallel_sort(keys, nthreads=4)
     assert np.array_equal(out, np.sort(keys))
+
+
+def test_wire_meta_roundtrip():
+    assert ps._core._test_meta_roundtrip()
