@@ -44,7 +44,7 @@ def parse_args():
     p.add_argument("--cpu", action="store_true", help="config #1: TCP van, host buffers")
     p.add_argument("--size-mb", type=float, default=64.0,
                    help="dense message size per key (headline: 64 MiB)")
-    p.add_argument("--keys-per-server", type=int, default=8,
+    p.add_argument("--keys-per-server", type=int, default=16,
                    help="with --gpus 8 this is exact; the TOTAL key count is "
                         "keys_per_server*8 and stays FIXED as N varies (the "
                         "model does not grow with the cluster -> honest weak "
