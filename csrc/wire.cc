@@ -34,6 +34,7 @@ static void UnpackNode(ByteReader* r, Node* n) {
   n->pool_capacity = r->U64();
   n->pool_slab_bytes = r->U64();
   int nh = r->I32();
+  XPS_CHECK(nh >= 0 && nh <= 4096) << "wire: bad pool handle count " << nh;
   n->pool_handles.resize(nh);
   for (int i = 0; i < nh; ++i) r->Raw(n->pool_handles[i].data(), kIpcHandleBytes);
   n->shm_uid = r->U64();
@@ -109,6 +110,7 @@ void UnpackMeta(const char* buf, size_t len, Meta* m) {
     m->control.instance_barrier = r.U8();
     m->control.msg_sig = r.U64();
     int nn = r.I32();
+    XPS_CHECK(nn >= 0 && nn <= 65536) << "wire: bad node count " << nn;
     m->control.node.resize(nn);
     for (int i = 0; i < nn; ++i) UnpackNode(&r, &m->control.node[i]);
   }

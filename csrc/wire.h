@@ -39,13 +39,15 @@ class ByteReader {
   int64_t I64() { int64_t v; Raw(&v, 8); return v; }
   std::string Str() {
     uint64_t n = U64();
-    XPS_CHECK_LE(p_ + n, end_) << "wire: truncated string";
+    // compare against the remaining length, never p_ + n (which can wrap
+    // for a hostile/corrupt length and defeat the bounds check)
+    XPS_CHECK_LE(n, static_cast<uint64_t>(end_ - p_)) << "wire: truncated string";
     std::string s(p_, n);
     p_ += n;
     return s;
   }
   void Raw(void* out, size_t n) {
-    XPS_CHECK_LE(p_ + n, end_) << "wire: truncated field";
+    XPS_CHECK_LE(n, static_cast<size_t>(end_ - p_)) << "wire: truncated field";
     memcpy(out, p_, n);
     p_ += n;
   }
