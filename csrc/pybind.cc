@@ -544,6 +544,29 @@ PYBIND11_MODULE(_core, m) {
     return ok;
   });
 
+  // raw wire access for robustness tests: pack a representative meta,
+  // and unpack arbitrary bytes (XPS_CHECK-aborts on malformed input, so
+  // callers exercise it from a subprocess)
+  m.def("_pack_meta_sample", []() {
+    Meta m;
+    m.app_id = 1;
+    m.body = "abc";
+    m.control.cmd = Control::ADD_NODE;
+    Node n;
+    n.hostname = "127.0.0.1";
+    n.pool_handles.resize(1);
+    m.control.node.push_back(n);
+    std::string buf;
+    PackMeta(m, &buf);
+    return py::bytes(buf);
+  });
+  m.def("_unpack_meta_raw", [](py::bytes b) {
+    std::string s = b;
+    Meta out;
+    UnpackMeta(s.data(), s.size(), &out);
+    return true;
+  });
+
   // utility parity (ps-lite parallel_kv_match.h / parallel_sort.h)
   m.def("parallel_ordered_match",
         [](py::array_t<uint64_t> src_keys, py::array_t<float> src_vals,
