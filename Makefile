@@ -49,8 +49,12 @@ clean:
 CORE_OBJS := $(filter-out build/csrc/pybind.o,$(OBJS))
 build/bench_kv.o: examples/cpp/bench_kv.cc csrc/*.h
 	$(HIPCC) $(CXXFLAGS) -c examples/cpp/bench_kv.cc -o build/bench_kv.o
+CPPBENCH_LD := -O3 --offload-arch=$(ARCH)
+ifeq ($(ASAN),1)
+CPPBENCH_LD += -fsanitize=address
+endif
 cppbench: $(CORE_OBJS) build/bench_kv.o
-	$(HIPCC) -O3 --offload-arch=$(ARCH) build/bench_kv.o $(CORE_OBJS) \
+	$(HIPCC) $(CPPBENCH_LD) build/bench_kv.o $(CORE_OBJS) \
 	    -o build/bench_kv -lpthread
 
 .PHONY: all clean cppbench
