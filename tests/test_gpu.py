@@ -307,3 +307,37 @@ def test_multiprocess_hipipc_two_joint_on_one_gpu():
     # both workers pushed (1+2) with accumulate -> 3 everywhere
     for rank, (first, last) in results.items():
         assert first == 3.0 and last == 3.0, results
+
+
+def _gpu_scale_worker_fn(ps_mod, rank):
+    """8 joint procs sharing one GPU: each worker reduces a distinct key
+    set with all 8 servers (stress the concurrent hipIpc import ladder
+    the multi-GPU bootstrap uses)."""
+    server = ps_mod.KVServer(0)
+    server.set_gpu_dense_handle(mode="sum")
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker = ps_mod.KVWorker(0, 0)
+    n = 1 << 14
+    nsrv = 8
+    step = (1 << 64) // nsrv
+    keys = np.array(sorted(s * step + 3 for s in range(nsrv)), dtype=np.uint64)
+    lens = np.full(nsrv, n, dtype=np.int32)
+    src = ps_mod.pool_alloc(nsrv * n * 4)
+    dst = ps_mod.pool_alloc(nsrv * n * 4)
+    src.copy_from(np.full(nsrv * n, float(rank + 1), dtype=np.float32))
+    worker.wait(worker.zpush_ptr(keys, src.ptr, nsrv * n * 4, 0, lens, cmd=2))
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker.wait(worker.zpull_ptr(keys, dst.ptr, nsrv * n * 4, 0, lens))
+    out = dst.to_numpy_f32()
+    return (float(out[0]), float(out[-1])), server
+
+
+@pytest.mark.gpu
+def test_multiprocess_hipipc_eight_joint_on_one_gpu():
+    devices = {r: 0 for r in range(8)}
+    results = launch_local(8, 8, _gpu_scale_worker_fn, joint=True, devices=devices,
+                           env_extra={"XPS_POOL_GB": 2}, timeout=420)
+    # all 8 workers pushed rank+1 with accumulate -> sum(1..8) = 36
+    assert len(results) == 8
+    for rank, (first, last) in results.items():
+        assert first == 36.0 and last == 36.0, results
