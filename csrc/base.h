@@ -6,6 +6,8 @@
 // include/ps/internal/env.h (Environment singleton). See SURVEY.md §L1.
 #pragma once
 
+#include <atomic>
+#include <chrono>
 #include <cstdint>
 #include <cstdlib>
 #include <cstring>
@@ -117,5 +119,42 @@ class Environment {
 };
 
 inline int GetEnvInt(const char* k, int dflt) { return Environment::Get()->GetInt(k, dflt); }
+
+// ------------------------------------------------------------ stage timing
+// XPS_TIMING=1 accumulates wall time per named hot-path stage (send,
+// serialize, poll parse, handler, deferred release, ...) and prints one
+// table per process at plane shutdown. Off by default: a single branch
+// on a bool per scope.
+struct StageStat {
+  const char* name;
+  std::atomic<uint64_t> ns{0};
+  std::atomic<uint64_t> n{0};
+  explicit StageStat(const char* nm);
+};
+bool TimingEnabled();
+void PrintStageStats(const char* tag);
+
+class StageScope {
+ public:
+  explicit StageScope(StageStat* s) : s_(TimingEnabled() ? s : nullptr) {
+    if (s_) t0_ = std::chrono::steady_clock::now();
+  }
+  ~StageScope() {
+    if (s_) {
+      auto dt = std::chrono::steady_clock::now() - t0_;
+      s_->ns.fetch_add(std::chrono::duration_cast<std::chrono::nanoseconds>(dt).count(),
+                       std::memory_order_relaxed);
+      s_->n.fetch_add(1, std::memory_order_relaxed);
+    }
+  }
+
+ private:
+  StageStat* s_;
+  std::chrono::steady_clock::time_point t0_;
+};
+
+#define XPS_STAGE(nm)                         \
+  static ::xps::StageStat xps_stage_##nm(#nm); \
+  ::xps::StageScope xps_scope_##nm(&xps_stage_##nm)
 
 }  // namespace xps

@@ -64,6 +64,7 @@ void Customer::RunHandle(Message& msg) {
 }
 
 void Customer::ProcessInline(Message& msg) {
+  XPS_STAGE(customer_process);
   std::lock_guard<std::mutex> lk(handle_mu_);
   RunHandle(msg);
 }

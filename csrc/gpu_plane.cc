@@ -119,6 +119,7 @@ void GpuPlane::FillSelf(Node* self) {
 
 void GpuPlane::Stop() {
   if (stop_.exchange(true)) return;
+  PrintStageStats(device_ >= 0 ? "gpu plane" : "host plane");
   if (poll_thread_.joinable()) poll_thread_.join();
   if (comp_thread_.joinable()) comp_thread_.join();
   // drain pending sends synchronously so responses are not lost on stop
@@ -321,6 +322,7 @@ bool GpuPlane::CanSend(const Message& msg, const Node& peer) {
 }
 
 bool GpuPlane::Serialize(const Message& msg, const std::vector<char>& by_ref, std::string* out) {
+  XPS_STAGE(serialize);
   std::string meta;
   PackMeta(msg.meta, &meta);
   out->clear();
@@ -353,6 +355,7 @@ bool GpuPlane::Serialize(const Message& msg, const std::vector<char>& by_ref, st
 }
 
 int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
+  XPS_STAGE(plane_send);
   XPS_VLOG(3) << "plane send -> " << peer_node.id << ": " << msg.DebugString();
   Peer* p = GetPeer(peer_node.id);
   {
@@ -473,6 +476,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
 
 void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::string payload,
                                  int64_t bytes) {
+  XPS_STAGE(defer_queue);
   XPS_CHECK_GE(device_, 0) << "deferred sends are a GPU-plane feature";
   hipStream_t stream = StreamForPeer(peer_id);
   hipEvent_t ev = GetEvent();
@@ -504,6 +508,7 @@ void GpuPlane::CompletionLoop() {
           Pending& front = dq.front();
           hipError_t e = hipEventQuery(front.ev);
           if (e == hipErrorNotReady) break;
+          XPS_STAGE(defer_release);
           XPS_CHECK(e == hipSuccess) << "hipEventQuery: " << hipGetErrorString(e);
           Peer* peer = GetPeer(front.peer_id);
           if (peer && EnsureRing(peer)) {
@@ -540,6 +545,7 @@ void GpuPlane::RingPollLoop() {
       continue;
     }
     idle = 0;
+    XPS_STAGE(ring_handle);
     ByteReader r(buf.data(), n);
     uint64_t meta_len = r.U64();
     Message msg;

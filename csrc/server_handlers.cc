@@ -69,6 +69,7 @@ void GpuDenseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
 
 void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& kvs,
                                        KVServer<float>* server) {
+  XPS_STAGE(reduce_push);
   XPS_CHECK_EQ(kvs.keys.size(), 1u) << "reduce mode is single-key-per-message";
   size_t len = kvs.lens.empty() ? kvs.vals.nbytes()
                                 : static_cast<size_t>(kvs.lens[0]) * sizeof(float);
@@ -116,6 +117,7 @@ void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& 
 
 void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
                                  KVServer<float>* server) {
+  XPS_STAGE(dense_push);
   size_t n = kvs.keys.size();
   XPS_CHECK_GT(n, 0u);
   if (mode_ == DenseMode::kReduce) {
@@ -207,6 +209,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
 }
 
 void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* server) {
+  XPS_STAGE(reduce_respond_pull);
   hipStream_t stream = Stream(req.sender);
   for (auto ev : e->round_events) {
     XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev, 0));
@@ -248,6 +251,7 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* 
 
 void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
                                  KVServer<float>* server) {
+  XPS_STAGE(dense_pull);
   size_t n = kvs.keys.size();
   XPS_CHECK_GT(n, 0u);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));

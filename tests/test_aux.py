@@ -63,3 +63,21 @@ def test_profiling_trace_file(tmp_path):
     assert traces, "no trace files written"
     content = "".join(open(t).read() for t in traces)
     assert "van_send_push" in content or "van_recv_push" in content, content[:500]
+
+
+def _timing_worker(ps_mod, rank):
+    w = ps_mod.KVWorker(0, 0)
+    keys = np.array([3], dtype=np.uint64)
+    vals = np.ones(512, dtype=np.float32)
+    for _ in range(5):
+        w.wait(w.push(keys, vals, np.array([512], dtype=np.int32)))
+    return True
+
+
+def test_stage_timing_table(capfd):
+    """XPS_TIMING=1 prints the per-stage table at plane shutdown."""
+    results = launch_local(1, 1, _timing_worker,
+                           env_extra={"XPS_TIMING": "1"}, timeout=180)
+    assert results[0] is True
+    err = capfd.readouterr().err
+    assert "stage timing" in err and "plane_send" in err, err[-2000:]
