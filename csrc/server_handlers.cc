@@ -42,16 +42,25 @@ static void EventFree(Postoffice* po, hipEvent_t ev) {
 
 // ------------------------------------------------------------------ dense
 
+static EventRef MakeEventRef(Postoffice* po, hipStream_t s) {
+  hipEvent_t ev = EventAlloc(po);
+  XPS_HIP_CHECK(hipEventRecord(ev, s));
+  return EventRef(ev, [po](hipEvent_t e) { EventFree(po, e); });
+}
+
 GpuDenseHandler::GpuDenseHandler(Postoffice* po, DenseMode mode) : po_(po), mode_(mode) {
   XPS_CHECK(HbmPool::Get()->initialized()) << "GpuDenseHandler needs the HBM pool";
   num_workers_ = std::max(1, po_->num_workers());
+  // one worker process = one peer stream = stream-ordered kernels; with
+  // more, same-key kernels land on different peer streams and must be
+  // chained through the entry's last_ev
+  chain_ = num_workers_ > 1;
 }
 
-GpuDenseHandler::~GpuDenseHandler() {
-  for (auto& kv : store_) {
-    for (auto ev : kv.second.round_events) (void)hipEventDestroy(ev);
-    for (auto ev : kv.second.pull_guard) (void)hipEventDestroy(ev);
-  }
+GpuDenseHandler::~GpuDenseHandler() = default;
+
+void GpuDenseHandler::OrderAfter(Entry* e, hipStream_t s) {
+  if (e->last_ev) XPS_HIP_CHECK(hipStreamWaitEvent(s, e->last_ev.get(), 0));
 }
 
 hipStream_t GpuDenseHandler::Stream(int sender) { return PeerStream(po_, sender, &fallback_stream_); }
@@ -90,11 +99,11 @@ void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& 
   hipStream_t stream = Stream(req.sender);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
   XPS_CHECK(kvs.vals.on_device()) << "reduce mode needs device vals (pool buffers)";
+  OrderAfter(e, stream);  // chain behind the previous kernel on this key
   if (e->pushes == 0) {
     // the previous round's pull copies must finish before we overwrite
-    for (auto ev : e->pull_guard) {
-      XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev, 0));
-      EventFree(po_, ev);
+    for (auto& ev : e->pull_guard) {
+      XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev.get(), 0));
     }
     e->pull_guard.clear();
     kern::DenseAssign(e->buf.data(), kvs.vals.data(), len, stream);
@@ -104,9 +113,9 @@ void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& 
                       stream);
   }
   e->pushes++;
-  hipEvent_t ev = EventAlloc(po_);
-  XPS_HIP_CHECK(hipEventRecord(ev, stream));
+  EventRef ev = MakeEventRef(po_, stream);
   e->round_events.push_back(ev);
+  e->last_ev = ev;
   server->Response(req);
   if (e->pushes >= num_workers_) {
     std::vector<KVMeta> waiting;
@@ -130,7 +139,9 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   // multi-key device push: one batched kernel launch for all segments
   if (kvs.vals.on_device() && n > 1) {
     std::vector<kern::CopyDesc> descs;
+    std::vector<Entry*> ents;
     descs.reserve(n);
+    ents.reserve(n);
     size_t boff = 0;
     bool aligned = true;
     for (size_t i = 0; i < n; ++i) {
@@ -145,16 +156,22 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
           XPS_HIP_CHECK(hipMemset(e->buf.data(), 0, len));
         }
       }
+      ents.push_back(e);
       descs.push_back({e->buf.data(),
                        reinterpret_cast<const char*>(kvs.vals.data()) + boff, len});
       aligned = aligned && (len % 16 == 0) && (boff % 16 == 0);
       boff += len;
     }
     if (aligned) {
+      for (Entry* e : ents) OrderAfter(e, stream);
       if (sum_all) {
         kern::BatchedSumF32(descs.data(), static_cast<int>(n), stream);
       } else {
         kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
+      }
+      if (chain_) {
+        EventRef ev = MakeEventRef(po_, stream);  // one event covers the batch
+        for (Entry* e : ents) e->last_ev = ev;
       }
       auto* plane0 = po_->van() ? po_->van()->plane() : nullptr;
       if (!plane0) XPS_HIP_CHECK(hipStreamSynchronize(stream));
@@ -181,13 +198,21 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
     bool sum = sum_all;
     const char* src = reinterpret_cast<const char*>(kvs.vals.data()) + off;
     if (kvs.vals.on_device()) {
+      OrderAfter(e, stream);
       if (sum) {
         kern::DenseSumF32(reinterpret_cast<float*>(e->buf.data()),
                           reinterpret_cast<const float*>(src), len / sizeof(float), stream);
       } else {
         kern::DenseAssign(e->buf.data(), src, len, stream);
       }
+      if (chain_) e->last_ev = MakeEventRef(po_, stream);
     } else {
+      // host vals land via synchronous copies below: drain the entry's
+      // outstanding cross-stream kernel first
+      if (chain_ && e->last_ev) {
+        XPS_HIP_CHECK(hipEventSynchronize(e->last_ev.get()));
+        e->last_ev.reset();
+      }
       // host vals (TCP-staged path): correctness-first synchronous route
       if (sum) {
         SArray<char> scratch = HbmPool::Get()->AllocArray(len);
@@ -211,8 +236,8 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
 void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* server) {
   XPS_STAGE(reduce_respond_pull);
   hipStream_t stream = Stream(req.sender);
-  for (auto ev : e->round_events) {
-    XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev, 0));
+  for (auto& ev : e->round_events) {
+    XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev.get(), 0));
   }
   KVPairs<float> res;
   res.keys = SArray<Key>({req.key});
@@ -222,9 +247,9 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* 
   res.lens = lens;
   if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
   server->Response(req, res);  // the plane enqueues the copy on `stream` here
-  hipEvent_t pe = EventAlloc(po_);
-  XPS_HIP_CHECK(hipEventRecord(pe, stream));
+  EventRef pe = MakeEventRef(po_, stream);
   e->pull_guard.push_back(pe);
+  e->last_ev = pe;
   e->pulled_senders.insert(req.sender);
   e->pulls++;
   if (e->pulls >= num_workers_) {
@@ -232,8 +257,7 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* 
     e->pushes = 0;
     e->pulls = 0;
     e->pulled_senders.clear();
-    for (auto ev : e->round_events) EventFree(po_, ev);
-    e->round_events.clear();
+    e->round_events.clear();  // refs drop back to the event pool
     std::vector<std::pair<KVMeta, KVPairs<float>>> dpush;
     dpush.swap(e->waiting_pushes);
     for (auto& d : dpush) HandleReducePush(d.first, d.second, server);
@@ -280,6 +304,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   if (n > 1 && (req.option & kOptPullAddr)) {
     if (auto* plane = ThePlane(po_)) {
       std::vector<kern::CopyDesc> descs;
+      std::vector<Entry*> ents;
       SArray<int> lens2(n);
       uint64_t off = 0;
       bool ok = true;
@@ -292,6 +317,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
           char* dst = plane->PeerDst(req.sender, req.addr + off, len);
           ok = dst != nullptr && len % 16 == 0 && off % 16 == 0;
           if (ok) {
+            ents.push_back(&it->second);
             descs.push_back({dst, it->second.buf.data(), len});
             lens2[i] = static_cast<int>(len / sizeof(float));
             off += len;
@@ -299,6 +325,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
         }
       }
       if (ok) {
+        for (Entry* e : ents) OrderAfter(e, stream);
         kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
         KVMeta r2 = req;
         r2.option |= kOptInPlace;
@@ -307,6 +334,10 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
         res2.keys = kvs.keys;
         res2.lens = lens2;
         server->Response(r2, res2);  // meta+keys/lens only; plane defers on `stream`
+        if (chain_) {
+          EventRef ev = MakeEventRef(po_, stream);  // pushes must wait these reads
+          for (Entry* e : ents) e->last_ev = ev;
+        }
         return;
       }
     }
@@ -314,6 +345,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   KVPairs<float> res;
   res.keys = kvs.keys;
   SArray<int> lens(n);
+  std::vector<Entry*> touched;
   if (n == 1) {
     SArray<char> entry;
     {
@@ -321,7 +353,11 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
       auto it = store_.find(kvs.keys[0]);
       XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[0];
       entry = it->second.buf;
+      touched.push_back(&it->second);
     }
+    // the plane's in-place write (or the sync below) reads buf on
+    // `stream`: order it behind the last writer
+    OrderAfter(touched[0], stream);
     res.vals = SArray<float>::View(entry);  // zero-copy store view
     lens[0] = static_cast<int>(entry.size() / sizeof(float));
   } else {
@@ -333,10 +369,12 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
         auto it = store_.find(kvs.keys[i]);
         XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[i];
         entries[i] = it->second.buf;
+        touched.push_back(&it->second);
         lens[i] = static_cast<int>(entries[i].size() / sizeof(float));
         total += entries[i].size();
       }
     }
+    for (Entry* e : touched) OrderAfter(e, stream);
     SArray<char> tmp = HbmPool::Get()->AllocArray(total);
     size_t off = 0;
     std::vector<kern::CopyDesc> descs;
@@ -358,7 +396,11 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   // TCP fallback (no in-place destination): the staging D2H copy below in
   // the van is stream-unaware — drain our stream first
   if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
-  server->Response(req, res);
+  server->Response(req, res);  // plane enqueues the in-place read on `stream`
+  if (chain_) {
+    EventRef ev = MakeEventRef(po_, stream);  // pushes must wait these reads
+    for (Entry* e : touched) e->last_ev = ev;
+  }
 }
 
 void GpuDenseHandler::Save(const std::string& path) {
@@ -406,6 +448,9 @@ GpuSparseHandler::GpuSparseHandler(Postoffice* po, size_t rows, size_t row_len, 
     : po_(po), rows_(rows), row_len_(row_len), accumulate_(accumulate), key_shift_(key_shift) {
   auto* pool = HbmPool::Get();
   XPS_CHECK(pool->initialized()) << "GpuSparseHandler needs the HBM pool";
+  // concurrent workers scatter on different peer streams: accumulate
+  // must be element-atomic or overlapping rows lose updates
+  atomic_ = po_->num_workers() > 1;
   if (key_shift_ > 0) {
     int rank = po_->my_rank();
     row_base_ = po_->GetServerKeyRanges()[rank].begin >> key_shift_;
@@ -450,7 +495,7 @@ void GpuSparseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
     XPS_CHECK_EQ(kvs.vals.size(), n * row_len_);
     const uint64_t* rows = DeviceKeys(kvs.keys, req.sender, stream);
     if (accumulate_ || req.cmd == kCmdSum) {
-      kern::SparseScatterAddF32(table, rows, n, row_len_, kvs.vals.data(), /*atomic=*/false,
+      kern::SparseScatterAddF32(table, rows, n, row_len_, kvs.vals.data(), atomic_,
                                 stream, key_shift_, row_base_);
     } else {
       kern::SparseScatterAssignF32(table, rows, n, row_len_, kvs.vals.data(), stream, key_shift_,

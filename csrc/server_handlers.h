@@ -13,12 +13,20 @@
 
 #include <hip/hip_runtime.h>
 
+#include <memory>
 #include <set>
+#include <type_traits>
 #include <unordered_map>
 
 #include "kv_app.h"
 
 namespace xps {
+
+// A pooled hipEvent_t with shared ownership: freed back to the plane's
+// event pool (or destroyed) when the last reference drops. Shared
+// ownership lets one recorded event serve as the ordering point for
+// every key touched by a batched kernel.
+using EventRef = std::shared_ptr<std::remove_pointer<hipEvent_t>::type>;
 
 // handler modes
 enum class DenseMode {
@@ -54,8 +62,14 @@ class GpuDenseHandler {
     // pulls from senders that already pulled this round (next-round pulls)
     std::vector<KVMeta> waiting_next_pulls;
     std::set<int> pulled_senders;
-    std::vector<hipEvent_t> round_events;  // one per pusher stream
-    std::vector<hipEvent_t> pull_guard;    // pull copies the next round must wait on
+    std::vector<EventRef> round_events;  // one per pusher stream
+    std::vector<EventRef> pull_guard;    // pull copies the next round must wait on
+    // last kernel that touched buf, on WHICHEVER peer stream: with >1
+    // workers, concurrent senders use different streams, so every
+    // kernel touching this entry waits on last_ev and publishes a new
+    // one — otherwise two senders' sum kernels race (lost updates) and
+    // a pull can read a half-written store
+    EventRef last_ev;
   };
 
   void HandlePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
@@ -63,10 +77,12 @@ class GpuDenseHandler {
   void HandlePull(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   void RespondPull(const KVMeta& req, Entry* e, KVServer<float>* server);
   hipStream_t Stream(int sender);
+  void OrderAfter(Entry* e, hipStream_t s);  // wait the entry's last_ev
 
   Postoffice* po_;
   DenseMode mode_;
   int num_workers_ = 1;
+  bool chain_ = false;  // >1 workers: cross-stream same-key ordering needed
   std::mutex mu_;
   std::unordered_map<Key, Entry> store_;
   hipStream_t fallback_stream_ = nullptr;
@@ -95,6 +111,7 @@ class GpuSparseHandler {
   size_t rows_;
   size_t row_len_;
   bool accumulate_;
+  bool atomic_ = false;  // >1 workers: concurrent scatters need atomics
   int key_shift_ = 0;
   uint64_t row_base_ = 0;
   SArray<char> table_;
