@@ -23,20 +23,29 @@ TcpConn::TcpConn(int fd) : fd_(fd) {
   setsockopt(fd_, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
 }
 
-TcpConn::~TcpConn() { Close(); }
+TcpConn::~TcpConn() {
+  // the only close(): by now every reader holding this conn has exited
+  // (they keep a shared_ptr), so the fd number cannot be reused under a
+  // blocked recv()
+  int fd = fd_.exchange(-1);
+  if (fd >= 0) {
+    shutdown(fd, SHUT_RDWR);
+    close(fd);
+  }
+}
 
 void TcpConn::Close() {
-  if (fd_ >= 0) {
-    shutdown(fd_, SHUT_RDWR);
-    close(fd_);
-    fd_ = -1;
-  }
+  // wake any thread blocked in recv()/send() on this conn; the fd stays
+  // open (and poisoned) until the destructor so the kernel cannot hand
+  // the same number to an unrelated connection while a reader races
+  int fd = fd_.load(std::memory_order_relaxed);
+  if (fd >= 0) shutdown(fd, SHUT_RDWR);
 }
 
 bool TcpConn::SendAll(const void* p, size_t n) {
   const char* c = static_cast<const char*>(p);
   while (n > 0) {
-    ssize_t w = send(fd_, c, n, MSG_NOSIGNAL);
+    ssize_t w = send(fd_.load(std::memory_order_relaxed), c, n, MSG_NOSIGNAL);
     if (w < 0) {
       if (errno == EINTR) continue;
       return false;
@@ -50,7 +59,7 @@ bool TcpConn::SendAll(const void* p, size_t n) {
 bool TcpConn::RecvAll(void* p, size_t n) {
   char* c = static_cast<char*>(p);
   while (n > 0) {
-    ssize_t r = recv(fd_, c, n, 0);
+    ssize_t r = recv(fd_.load(std::memory_order_relaxed), c, n, 0);
     if (r < 0) {
       if (errno == EINTR) continue;
       return false;
@@ -64,7 +73,7 @@ bool TcpConn::RecvAll(void* p, size_t n) {
 
 int64_t TcpConn::SendFrame(const std::string& meta, const std::vector<SArray<char>>& data) {
   std::lock_guard<std::mutex> lk(send_mu_);
-  if (fd_ < 0) return -1;
+  if (fd_.load(std::memory_order_relaxed) < 0) return -1;
   uint32_t ndata = static_cast<uint32_t>(data.size());
   std::string hdr;
   hdr.reserve(12 + 8 * ndata);
@@ -139,21 +148,22 @@ int TcpConnect(const std::string& host, int port, int retries, int retry_ms) {
 }
 
 int TcpListener::Bind(int port, int retries) {
-  listen_fd_ = socket(AF_INET, SOCK_STREAM, 0);
-  if (listen_fd_ < 0) return -1;
+  int fd = socket(AF_INET, SOCK_STREAM, 0);
+  listen_fd_ = fd;
+  if (fd < 0) return -1;
   int one = 1;
-  setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
   struct sockaddr_in addr;
   memset(&addr, 0, sizeof(addr));
   addr.sin_family = AF_INET;
   addr.sin_addr.s_addr = htonl(INADDR_ANY);
   for (int attempt = 0; attempt < retries; ++attempt) {
     addr.sin_port = htons(static_cast<uint16_t>(port));
-    if (bind(listen_fd_, reinterpret_cast<struct sockaddr*>(&addr), sizeof(addr)) == 0) {
+    if (::bind(fd, reinterpret_cast<struct sockaddr*>(&addr), sizeof(addr)) == 0) {
       socklen_t len = sizeof(addr);
-      getsockname(listen_fd_, reinterpret_cast<struct sockaddr*>(&addr), &len);
+      getsockname(fd, reinterpret_cast<struct sockaddr*>(&addr), &len);
       port_ = ntohs(addr.sin_port);
-      if (listen(listen_fd_, 128) == 0) return port_;
+      if (listen(fd, 128) == 0) return port_;
       return -1;
     }
     if (port != 0) port += 1;  // probe next port like ps-lite's bind retry
@@ -180,12 +190,13 @@ void TcpListener::StartAccepting(std::function<void(int)> cb) {
 
 void TcpListener::Stop() {
   if (stop_.exchange(true)) return;
-  if (listen_fd_ >= 0) {
-    shutdown(listen_fd_, SHUT_RDWR);
-    close(listen_fd_);
-    listen_fd_ = -1;
-  }
+  // shutdown wakes the blocked accept(); close only after the accept
+  // thread joined (no accept can race a reused fd number)
+  int fd = listen_fd_.load(std::memory_order_relaxed);
+  if (fd >= 0) shutdown(fd, SHUT_RDWR);
   if (accept_thread_.joinable()) accept_thread_.join();
+  fd = listen_fd_.exchange(-1);
+  if (fd >= 0) close(fd);
 }
 
 uint64_t HostHash() {

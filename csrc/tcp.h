@@ -34,12 +34,15 @@ class TcpConn {
   int64_t RecvFrame(std::string* meta, std::vector<SArray<char>>* data);
 
   void Close();
-  int fd() const { return fd_; }
+  int fd() const { return fd_.load(std::memory_order_relaxed); }
 
  private:
   bool SendAll(const void* p, size_t n);
   bool RecvAll(void* p, size_t n);
-  int fd_;
+  // atomic: Close() runs from the van's Stop path to WAKE a reader
+  // thread blocked in recv() on this same fd (shutdown-to-unblock);
+  // exchange also makes destructor-vs-Stop close idempotent
+  std::atomic<int> fd_;
   std::mutex send_mu_;
 };
 
@@ -57,7 +60,7 @@ class TcpListener {
   ~TcpListener() { Stop(); }
 
  private:
-  int listen_fd_ = -1;
+  std::atomic<int> listen_fd_{-1};  // Stop() closes it under the accept loop
   int port_ = -1;
   std::atomic<bool> stop_{false};
   std::thread accept_thread_;
