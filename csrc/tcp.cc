@@ -38,6 +38,7 @@ void TcpConn::Close() {
   // wake any thread blocked in recv()/send() on this conn; the fd stays
   // open (and poisoned) until the destructor so the kernel cannot hand
   // the same number to an unrelated connection while a reader races
+  closed_.store(true, std::memory_order_relaxed);
   int fd = fd_.load(std::memory_order_relaxed);
   if (fd >= 0) shutdown(fd, SHUT_RDWR);
 }

@@ -35,6 +35,9 @@ class TcpConn {
 
   void Close();
   int fd() const { return fd_.load(std::memory_order_relaxed); }
+  // true once Close() ran: the fd stays open (deferred close in the
+  // destructor) but the connection is dead for routing purposes
+  bool closed() const { return closed_.load(std::memory_order_relaxed); }
 
  private:
   bool SendAll(const void* p, size_t n);
@@ -43,6 +46,7 @@ class TcpConn {
   // thread blocked in recv() on this same fd (shutdown-to-unblock);
   // exchange also makes destructor-vs-Stop close idempotent
   std::atomic<int> fd_;
+  std::atomic<bool> closed_{false};
   std::mutex send_mu_;
 };
 

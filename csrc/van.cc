@@ -281,7 +281,7 @@ void Van::RecvLoop(std::shared_ptr<TcpConn> conn) {
     if (msg.meta.sender != kEmptyNodeID) {
       std::lock_guard<std::mutex> lk(conn_mu_);
       auto it = conns_.find(msg.meta.sender);
-      if (it == conns_.end() || it->second->fd() < 0) conns_[msg.meta.sender] = conn;
+      if (it == conns_.end() || it->second->closed()) conns_[msg.meta.sender] = conn;
     }
     if (!msg.meta.control.empty()) {
       ProcessControl(msg, conn);
