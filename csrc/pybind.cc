@@ -4,6 +4,9 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <atomic>
+#include <thread>
+
 #include "gpu_plane.h"
 #include "hip_pool.h"
 #include "hip_util.h"
@@ -13,6 +16,7 @@
 #include "kv_utils.h"
 #include "ps.h"
 #include "server_handlers.h"
+#include "shm_ring.h"
 #include "simple_app.h"
 #include "van.h"
 #include "wire.h"
@@ -493,6 +497,55 @@ PYBIND11_MODULE(_core, m) {
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
 
   // wire-format roundtrip (unit-test hook)
+  // shm-ring stress hook: P producer threads push `per` tagged messages
+  // each through one ring while a consumer drains; returns (received,
+  // payload checksum ok). Deliberately overruns kSlots so the full-ring
+  // CAS backoff path runs (the single consumer must never wedge).
+  m.def("_ring_stress", [](int producers, int per, int payload) {
+    ShmRing ring;
+    XPS_CHECK(ring.Create(0xabcdef12345ull + getpid()));
+    std::atomic<long> received{0}, sum{0}, pushed{0};
+    std::atomic<bool> done{false};
+    std::thread consumer([&]() {
+      std::vector<char> buf(ShmRing::MaxPayload());
+      while (true) {
+        uint32_t n = ring.Pop(buf.data());
+        if (n == 0) {
+          if (done.load() && received.load() >= pushed.load()) break;
+          std::this_thread::yield();
+          continue;
+        }
+        long v = 0;
+        for (uint32_t i = 0; i < n; ++i) v += static_cast<unsigned char>(buf[i]);
+        sum.fetch_add(v);
+        received.fetch_add(1);
+      }
+    });
+    std::vector<std::thread> ps;
+    std::atomic<long> expect{0};
+    for (int p = 0; p < producers; ++p) {
+      ps.emplace_back([&, p]() {
+        std::vector<char> msg(payload);
+        for (int i = 0; i < per; ++i) {
+          for (int b = 0; b < payload; ++b) msg[b] = static_cast<char>((p + i + b) & 0x7f);
+          long v = 0;
+          for (int b = 0; b < payload; ++b) v += static_cast<unsigned char>(msg[b]);
+          if (ring.Push(msg.data(), payload)) {
+            expect.fetch_add(v);
+            pushed.fetch_add(1);
+          }
+        }
+      });
+    }
+    for (auto& t : ps) t.join();
+    done.store(true);
+    consumer.join();
+    ring.CloseAndUnlink();
+    return py::make_tuple(static_cast<long>(received.load()),
+                          static_cast<long>(pushed.load()),
+                          sum.load() == expect.load());
+  });
+
   m.def("_test_meta_roundtrip", []() {
     Meta m;
     m.app_id = 7;

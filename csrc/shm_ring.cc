@@ -111,18 +111,34 @@ void ShmRing::CloseAndUnlink() {
 
 bool ShmRing::Push(const void* payload, uint32_t len) {
   if (len > MaxPayload() || !mem_) return false;
-  uint64_t pos = Hdr(mem_)->head.fetch_add(1, std::memory_order_relaxed);
-  RingSlot* s = Slot(mem_, pos);
-  // wait for the consumer to free this slot (seq == pos); bounded so a
-  // crashed consumer cannot wedge this producer forever (~10 s)
+  // Vyukov MPMC enqueue: reserve a position ONLY once its slot is free
+  // (seq == pos, then CAS head). A ticket fetch_add cannot back out on a
+  // full ring: dropping would leave a hole the single consumer waits on
+  // forever. With CAS, a bounded wait (~10 s) on a full ring can give up
+  // without consuming a sequence number — the ring stays usable.
+  RingHeader* h = Hdr(mem_);
+  uint64_t pos = h->head.load(std::memory_order_relaxed);
   int spins = 0;
-  while (s->seq.load(std::memory_order_acquire) != pos) {
-    if (++spins > 1000) {
-      std::this_thread::yield();
-      if (spins > 10 * 1000 * 1000) {
-        XPS_LOG(Warning) << "shm ring full for too long (consumer dead?); dropping message";
-        return false;
+  RingSlot* s;
+  for (;;) {
+    s = Slot(mem_, pos);
+    uint64_t seq = s->seq.load(std::memory_order_acquire);
+    int64_t dif = static_cast<int64_t>(seq) - static_cast<int64_t>(pos);
+    if (dif == 0) {
+      if (h->head.compare_exchange_weak(pos, pos + 1, std::memory_order_relaxed)) break;
+      // pos was reloaded by the failed CAS; retry
+    } else if (dif < 0) {
+      // ring full: the consumer has not freed this slot yet
+      if (++spins > 1000) {
+        std::this_thread::yield();
+        if (spins > 10 * 1000 * 1000) {
+          XPS_LOG(Warning) << "shm ring full for too long (consumer dead?); dropping message";
+          return false;
+        }
       }
+      pos = h->head.load(std::memory_order_relaxed);
+    } else {
+      pos = h->head.load(std::memory_order_relaxed);
     }
   }
   memcpy(s->payload, payload, len);

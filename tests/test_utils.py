@@ -42,3 +42,12 @@ def test_parallel_sort():
 
 def test_wire_meta_roundtrip():
     assert ps._core._test_meta_roundtrip()
+
+
+def test_shm_ring_stress_full_ring():
+    """8 producers x 2000 msgs through a 1024-slot ring: the full-ring
+    backoff path runs constantly; nothing may be lost or corrupted and
+    the consumer must never wedge."""
+    received, pushed, checksum_ok = ps._core._ring_stress(8, 2000, 512)
+    assert received == pushed == 8 * 2000
+    assert checksum_ok
