@@ -86,6 +86,9 @@ GpuPlane::GpuPlane(Postoffice* po, int device) : po_(po), device_(device) {
 GpuPlane::~GpuPlane() { Stop(); }
 
 void GpuPlane::FillSelf(Node* self) {
+  // reap segments left by crashed/killed runs before creating ours
+  static std::once_flag gc_once;
+  std::call_once(gc_once, [] { HostShmPool::GcStaleSegments(); });
   self->dev_id = device_;
   auto* pool = HbmPool::Get();
   if (device_ >= 0 && pool->initialized()) {

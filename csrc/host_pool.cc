@@ -1,6 +1,8 @@
 #include "host_pool.h"
 
+#include <dirent.h>
 #include <fcntl.h>
+#include <signal.h>
 #include <sys/mman.h>
 #include <unistd.h>
 
@@ -12,6 +14,11 @@
 namespace xps {
 
 static const size_t kAlign = 64;
+// 64-byte arena header: {magic, owner pid}; allocations start at offset
+// 64 so peer-visible offsets stay consistent on both sides
+static const size_t kHeaderBytes = 64;
+static const uint64_t kPoolMagic = 0x587053686d506f31ull;  // "XpShmPo1"
+
 
 static std::string PoolName(uint64_t uid) {
   char buf[64];
@@ -40,7 +47,10 @@ void HostShmPool::Init(uint64_t uid, size_t capacity_bytes) {
   XPS_CHECK(base_ != MAP_FAILED) << "mmap host pool";
   uid_ = uid;
   capacity_ = capacity_bytes;
-  free_[0] = capacity_;
+  uint64_t* hdr = static_cast<uint64_t*>(base_);
+  hdr[1] = static_cast<uint64_t>(getpid());
+  __atomic_store_n(&hdr[0], kPoolMagic, __ATOMIC_RELEASE);
+  free_[kHeaderBytes] = capacity_ - kHeaderBytes;
   XPS_VLOG(1) << "HostShmPool: " << (capacity_ >> 20) << " MiB (" << name << ")";
 }
 
@@ -120,6 +130,36 @@ void* HostShmPool::MapPeer(uint64_t uid, size_t capacity) {
   if (base == MAP_FAILED) base = nullptr;
   mapped[uid] = base;
   return base;
+}
+
+void HostShmPool::GcStaleSegments() {
+  DIR* d = opendir("/dev/shm");
+  if (!d) return;
+  // ring header layout (shm_ring.cc): head@0, tail@64, magic@128, pid@192
+  static const uint64_t kRingMagic = 0x587052696e673166ull;
+  struct dirent* ent;
+  while ((ent = readdir(d)) != nullptr) {
+    bool pool = strncmp(ent->d_name, "xps_hostpool_", 13) == 0;
+    bool ring = strncmp(ent->d_name, "xps_ring_", 9) == 0;
+    if (!pool && !ring) continue;
+    std::string name = std::string("/") + ent->d_name;
+    int fd = shm_open(name.c_str(), O_RDONLY, 0600);
+    if (fd < 0) continue;
+    void* m = mmap(nullptr, 4096, PROT_READ, MAP_SHARED, fd, 0);
+    close(fd);
+    if (m == MAP_FAILED) continue;
+    const uint64_t* w = static_cast<const uint64_t*>(m);
+    uint64_t magic = pool ? w[0] : w[16];   // ring magic at byte 128
+    uint64_t pid = pool ? w[1] : w[24];     // ring pid at byte 192
+    bool dead = magic == (pool ? kPoolMagic : kRingMagic) && pid > 0 &&
+                kill(static_cast<pid_t>(pid), 0) == -1 && errno == ESRCH;
+    munmap(m, 4096);
+    if (dead) {
+      XPS_VLOG(1) << "GC stale shm segment " << name << " (owner " << pid << " dead)";
+      shm_unlink(name.c_str());
+    }
+  }
+  closedir(d);
 }
 
 void HostShmPool::Unlink() {

@@ -34,6 +34,11 @@ class HostShmPool {
 
   void Unlink();
 
+  // unlink any /dev/shm xps segment (host pool or ring) whose recorded
+  // owner process is dead — crashed runs (SIGKILL, test teardown) leak
+  // segments that no teardown path can reach
+  static void GcStaleSegments();
+
  private:
   HostShmPool() = default;
   uint64_t uid_ = 0;

@@ -497,6 +497,12 @@ PYBIND11_MODULE(_core, m) {
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
 
   // wire-format roundtrip (unit-test hook)
+  // stale-shm GC hooks (tested in test_utils.py)
+  m.def("_host_pool_init_for_test", [](uint64_t uid, size_t bytes) {
+    HostShmPool::Get()->Init(uid, bytes);
+  });
+  m.def("_gc_stale_shm", []() { HostShmPool::GcStaleSegments(); });
+
   // shm-ring stress hook: P producer threads push `per` tagged messages
   // each through one ring while a consumer drains; returns (received,
   // payload checksum ok). Deliberately overruns kSlots so the full-ring

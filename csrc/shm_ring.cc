@@ -16,9 +16,10 @@ namespace xps {
 namespace {
 
 struct RingHeader {
-  alignas(64) std::atomic<uint64_t> head;  // producer ticket
+  alignas(64) std::atomic<uint64_t> head;  // producer position
   alignas(64) std::atomic<uint64_t> tail;  // consumer position
   alignas(64) uint64_t magic;
+  alignas(64) uint64_t owner_pid;  // for stale-segment GC (host_pool.cc)
 };
 static const uint64_t kRingMagic = 0x587052696e673166ull;  // "XpRing1f"
 
@@ -71,6 +72,7 @@ bool ShmRing::Create(uint64_t uid) {
   for (uint64_t i = 0; i < kSlots; ++i) Slot(mem_, i)->seq.store(i, std::memory_order_relaxed);
   Hdr(mem_)->head.store(0);
   Hdr(mem_)->tail.store(0);
+  Hdr(mem_)->owner_pid = static_cast<uint64_t>(getpid());
   __atomic_store_n(&Hdr(mem_)->magic, kRingMagic, __ATOMIC_RELEASE);
   owner_ = true;
   return true;

@@ -51,3 +51,21 @@ def test_shm_ring_stress_full_ring():
     received, pushed, checksum_ok = ps._core._ring_stress(8, 2000, 512)
     assert received == pushed == 8 * 2000
     assert checksum_ok
+
+
+def test_shm_gc_reaps_dead_owner_segments(tmp_path):
+    """A process that dies without teardown leaks its host-shm arena; the
+    next plane bring-up GCs it (owner pid recorded in the arena header)."""
+    import os
+    import subprocess
+    import sys
+    uid = 0xfeed0000 | os.getpid()
+    seg = f"/dev/shm/xps_hostpool_{uid:016x}"
+    code = (f"import ps_lite_amd as ps, os; "
+            f"ps._core._host_pool_init_for_test({uid}, 1 << 20); os._exit(0)")
+    env = dict(os.environ)
+    env.setdefault("PYTHONPATH", os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    subprocess.run([sys.executable, "-c", code], check=True, timeout=60, env=env)
+    assert os.path.exists(seg), "leaked segment expected"
+    ps._core._gc_stale_shm()
+    assert not os.path.exists(seg), "GC did not reap the dead-owner segment"
