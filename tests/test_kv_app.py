@@ -187,3 +187,33 @@ def test_resend_with_drop():
            "XPS_HOST_PLANE": "0"}
     results = launch_local(1, 1, _worker_single, env_extra=env, timeout=180)
     assert np.allclose(np.array(results[0]), np.ones(64))
+
+
+def _random_shard_worker(ps_mod, rank):
+    """Random keys + non-uniform lens across 4 servers: exercises the
+    slicer boundaries and the worker-side pull merge (re-sorting slices,
+    lens bookkeeping) against a local reference."""
+    server = None  # plain worker (separate server procs use default fn)
+    w = ps_mod.KVWorker(0, 0)
+    rng = np.random.default_rng(4242)
+    nkeys = 57
+    keys = np.sort(rng.choice(1 << 62, size=nkeys, replace=False)).astype(np.uint64)
+    lens = rng.integers(1, 96, size=nkeys).astype(np.int32)
+    vals = rng.standard_normal(int(lens.sum())).astype(np.float32)
+    w.wait(w.push(keys, vals, lens))
+    got = w.pull(keys)
+    assert got.shape == vals.shape
+    assert np.allclose(got, vals, atol=1e-6)
+    # partial pull of a random subset must return that subset's slices
+    idx = np.sort(rng.choice(nkeys, size=13, replace=False))
+    sub_keys = keys[idx]
+    expect = np.concatenate([
+        vals[int(lens[:i].sum()):int(lens[:i].sum() + lens[i])] for i in idx])
+    got2 = w.pull(sub_keys)
+    assert np.allclose(got2, expect, atol=1e-6)
+    return True
+
+
+def test_random_sharding_four_servers():
+    results = launch_local(1, 4, _random_shard_worker, timeout=240)
+    assert results[0] is True
