@@ -39,3 +39,31 @@ def test_multithreaded_sessions():
     results = launch_local(2, 2, _threaded_worker, timeout=300)
     for rank, errors in results.items():
         assert errors == [], (rank, errors)
+
+
+def _dense_reduce_worker(ps, rank):
+    """Both workers accumulate into the SAME key each step and read back
+    the cluster-wide sum (test_benchmark_stress DenseReduce shape)."""
+    w = ps.KVWorker(0, 0)
+    key = np.array([77], dtype=np.uint64)
+    width = 512
+    lens = np.array([width], dtype=np.int32)
+    expect = np.zeros(width, dtype=np.float32)
+    bad = []
+    for step in range(8):
+        mine = np.full(width, float(rank + 1 + step), dtype=np.float32)
+        other = np.full(width, float((1 - rank) + 1 + step), dtype=np.float32)
+        w.wait(w.push(key, mine, lens))
+        expect += mine + other
+        ps.barrier("worker", ps.WORKER_GROUP)  # both pushes landed
+        out = w.pull(key)
+        if not np.allclose(out, expect):
+            bad.append((step, out[:2].tolist(), expect[:2].tolist()))
+        ps.barrier("worker", ps.WORKER_GROUP)  # reads done before next round
+    return bad
+
+
+def test_dense_reduce_two_workers():
+    results = launch_local(2, 2, _dense_reduce_worker, timeout=300)
+    for rank, bad in results.items():
+        assert bad == [], (rank, bad)
