@@ -497,6 +497,26 @@ PYBIND11_MODULE(_core, m) {
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
 
   // wire-format roundtrip (unit-test hook)
+  // rank-ordering policy hook (tested in test_utils.py): takes
+  // (role, host, port) tuples, returns them in rank-assignment order
+  m.def("_order_nodes", [](std::vector<std::tuple<std::string, std::string, int>> in) {
+    std::vector<Node> nodes;
+    for (auto& t : in) {
+      Node n;
+      n.role = std::get<0>(t) == "worker" ? Node::WORKER : Node::SERVER;
+      n.hostname = std::get<1>(t);
+      n.port = std::get<2>(t);
+      nodes.push_back(n);
+    }
+    auto less = NodeRankOrder(nodes);
+    std::sort(nodes.begin(), nodes.end(), less);
+    std::vector<std::tuple<std::string, std::string, int>> out;
+    for (auto& n : nodes) {
+      out.emplace_back(n.role == Node::WORKER ? "worker" : "server", n.hostname, n.port);
+    }
+    return out;
+  });
+
   // stale-shm GC hooks (tested in test_utils.py)
   m.def("_host_pool_init_for_test", [](uint64_t uid, size_t bytes) {
     HostShmPool::Get()->Init(uid, bytes);

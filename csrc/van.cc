@@ -1,3 +1,5 @@
+#include <tuple>
+
 #include "van.h"
 
 #include <unistd.h>
@@ -364,6 +366,37 @@ void Van::ProcessControl(Message& msg, const std::shared_ptr<TcpConn>& conn) {
   }
 }
 
+std::function<bool(const Node&, const Node&)> NodeRankOrder(const std::vector<Node>& batch) {
+  auto* env = Environment::Get();
+  std::map<std::string, int> host_idx;  // BYTEPS_ORDERED_HOSTS position
+  {
+    std::string hosts = env->GetStr("BYTEPS_ORDERED_HOSTS", "");
+    int idx = 0;
+    size_t start = 0;
+    while (start <= hosts.size() && !hosts.empty()) {
+      size_t comma = hosts.find(',', start);
+      std::string h = hosts.substr(start, comma == std::string::npos ? comma : comma - start);
+      if (!h.empty()) host_idx[h] = idx++;
+      if (comma == std::string::npos) break;
+      start = comma + 1;
+    }
+  }
+  bool mixed = env->GetInt("BYTEPS_ENABLE_MIXED_MODE", 0) != 0;
+  std::set<std::string> worker_hosts;
+  if (mixed) {
+    for (auto& n : batch)
+      if (n.role == Node::WORKER) worker_hosts.insert(n.hostname);
+  }
+  auto key = [host_idx, mixed, worker_hosts](const Node& n) {
+    auto it = host_idx.find(n.hostname);
+    int order = it == host_idx.end() ? (1 << 30) : it->second;
+    int coloc = 0;
+    if (mixed && n.role == Node::SERVER) coloc = worker_hosts.count(n.hostname) ? 1 : 0;
+    return std::make_tuple(order, coloc, n.hostname, n.port);
+  };
+  return [key](const Node& a, const Node& b) { return key(a) < key(b); };
+}
+
 void Van::ProcessAddNodeAtScheduler(Message& msg, const std::shared_ptr<TcpConn>& conn) {
   if (ready_.load()) {
     // late registration = recovery: a restarted process re-joins and
@@ -382,11 +415,16 @@ void Van::ProcessAddNodeAtScheduler(Message& msg, const std::shared_ptr<TcpConn>
     if (static_cast<int>(pending_nodes_.size()) < expected) return;
     batch.swap(pending_nodes_);
   }
-  // deterministic order: by (host, port); recovery/aux pinning honored first
-  std::sort(batch.begin(), batch.end(), [](const auto& a, const auto& b) {
-    if (a.first.hostname != b.first.hostname) return a.first.hostname < b.first.hostname;
-    return a.first.port < b.first.port;
-  });
+  // deterministic order with the BytePS placement policies (ordered
+  // hosts / mixed mode — see NodeRankOrder); recovery/aux pinning first
+  {
+    std::vector<Node> nodes;
+    nodes.reserve(batch.size());
+    for (auto& p : batch) nodes.push_back(p.first);
+    auto less = NodeRankOrder(nodes);
+    std::sort(batch.begin(), batch.end(),
+              [&less](const auto& a, const auto& b) { return less(a.first, b.first); });
+  }
   std::set<int> taken;
   auto assign = [&](Node& n) {
     int rank = n.aux_id;

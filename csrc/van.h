@@ -15,6 +15,7 @@
 #include <thread>
 #include <unordered_map>
 #include <vector>
+#include <functional>
 
 #include "message.h"
 #include "tcp.h"
@@ -118,5 +119,15 @@ class Van {
   FILE* trace_file_ = nullptr;
   std::mutex trace_mu_;
 };
+
+// Rank-assignment ordering for a full ADD_NODE batch (reference
+// src/van.cc:126-177 placement policies):
+//   BYTEPS_ORDERED_HOSTS="h1,h2,..."  — nodes sort by the host's position
+//     in the list (unlisted hosts come after, by name);
+//   BYTEPS_ENABLE_MIXED_MODE=1        — servers on hosts with NO worker
+//     (non-colocated) sort before colocated servers, so they take the
+//     low server ranks;
+// then deterministic (hostname, port). Exposed for unit tests.
+std::function<bool(const Node&, const Node&)> NodeRankOrder(const std::vector<Node>& batch);
 
 }  // namespace xps

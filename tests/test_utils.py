@@ -69,3 +69,26 @@ def test_shm_gc_reaps_dead_owner_segments(tmp_path):
     assert os.path.exists(seg), "leaked segment expected"
     ps._core._gc_stale_shm()
     assert not os.path.exists(seg), "GC did not reap the dead-owner segment"
+
+
+def test_rank_ordering_policies():
+    """BytePS placement policies (reference van.cc:126-177): ordered
+    hosts honored; mixed mode gives non-colocated servers the low
+    server ranks."""
+    nodes = [("worker", "hostB", 1), ("server", "hostB", 2),
+             ("server", "hostC", 3), ("worker", "hostA", 4),
+             ("server", "hostA", 5)]
+    # default: (host, port)
+    out = ps._core._order_nodes(nodes)
+    assert [h for _, h, _ in out] == ["hostA", "hostA", "hostB", "hostB", "hostC"]
+    # ordered hosts win over name order
+    ps._core.init_env({"BYTEPS_ORDERED_HOSTS": "hostC,hostB"})
+    out = ps._core._order_nodes(nodes)
+    assert [h for _, h, _ in out] == ["hostC", "hostB", "hostB", "hostA", "hostA"]
+    # mixed mode: hostC has no worker -> its server sorts before
+    # colocated servers regardless of hostname order
+    ps._core.init_env({"BYTEPS_ORDERED_HOSTS": "", "BYTEPS_ENABLE_MIXED_MODE": "1"})
+    out = ps._core._order_nodes(nodes)
+    servers = [h for r, h, _ in out if r == "server"]
+    assert servers[0] == "hostC", out
+    ps._core.init_env({"BYTEPS_ENABLE_MIXED_MODE": "0"})
