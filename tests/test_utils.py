@@ -92,3 +92,18 @@ def test_rank_ordering_policies():
     servers = [h for r, h, _ in out if r == "server"]
     assert servers[0] == "hostC", out
     ps._core.init_env({"BYTEPS_ENABLE_MIXED_MODE": "0"})
+
+
+def test_partition_tensor():
+    from ps_lite_amd.utils.partition import partition_tensor
+    # 18 MB over 4 servers at 4 MB parts -> 5 parts, exact coverage
+    keys, lens = partition_tensor(18 << 20, num_servers=4)
+    assert len(keys) == 5 == len(lens)
+    assert int(lens.astype(np.int64).sum()) * 4 == 18 << 20
+    assert all(int(n) * 4 <= (4 << 20) + 4 for n in lens)
+    # keys sorted + spread over all 4 server ranges
+    assert np.all(np.diff(keys.astype(np.uint64)) > 0)
+    step = (1 << 64) // 4
+    servers = {int(k) // step for k in keys}
+    assert servers == {0, 1, 2, 3}
+    # a worker can push/pull with them directly (end-to-end)
