@@ -25,6 +25,12 @@ static const size_t kInlineMax = 4096;  // host blobs above this ride the shm po
     XPS_CHECK(e_ == hipSuccess) << "HIP error: " << hipGetErrorString(e_) << " in " #cmd; \
   } while (0)
 
+// count of blobs received BY REFERENCE (pool offset, no copy) — the
+// zero-copy assertion hook (reference test_benchmark.cc:169-181 checks
+// pointer equality against registered buffers; here the transport
+// itself reports it)
+std::atomic<uint64_t> g_zero_copy_recv{0};
+
 namespace {
 // process-wide dedup of hipIpcOpenMemHandle (two vans of a joint process
 // and multiple planes share peer pool mappings; never closed — pools are
@@ -580,6 +586,7 @@ void GpuPlane::RingPollLoop() {
         }
         msg.data.push_back(SArray<char>(ptr, len, device_));
         ref_bytes += len;
+        g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
       } else {  // kind 2: host shm pool
         uint64_t off = r.U64();
         uint64_t len = r.U64();
@@ -595,6 +602,7 @@ void GpuPlane::RingPollLoop() {
         }
         msg.data.push_back(SArray<char>(static_cast<char*>(base) + off, len, kCPU));
         ref_bytes += len;
+        g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
       }
     }
     if (!ok) continue;

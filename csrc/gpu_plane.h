@@ -16,6 +16,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <deque>
 #include <memory>
 #include <thread>
@@ -25,6 +26,10 @@
 #include "van.h"
 
 namespace xps {
+
+// zero-copy reception counter (by-ref blobs resolved straight into a
+// mapped peer pool; see gpu_plane.cc)
+extern std::atomic<uint64_t> g_zero_copy_recv;
 
 class Postoffice;
 

@@ -497,6 +497,10 @@ PYBIND11_MODULE(_core, m) {
       .def("set_python_handle", &PyKVServer::SetPythonHandle);
 
   // wire-format roundtrip (unit-test hook)
+  // transport-level zero-copy assertion (reference parity:
+  // test_benchmark.cc:169-181 registered-buffer pointer equality)
+  m.def("zero_copy_recv_count", []() { return g_zero_copy_recv.load(); });
+
   // rank-ordering policy hook (tested in test_utils.py): takes
   // (role, host, port) tuples, returns them in rank-assignment order
   m.def("_order_nodes", [](std::vector<std::tuple<std::string, std::string, int>> in) {

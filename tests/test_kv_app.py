@@ -217,3 +217,58 @@ def _random_shard_worker(ps_mod, rank):
 def test_random_sharding_four_servers():
     results = launch_local(1, 4, _random_shard_worker, timeout=240)
     assert results[0] is True
+
+
+def _zero_copy_worker(ps_mod, rank):
+    """Transport-level zero-copy assertion (reference
+    test_benchmark.cc:169-181 pointer-equality check): pushes from the
+    host pool must arrive at the server BY REFERENCE, not staged."""
+    w = ps_mod.KVWorker(0, 0)
+    before = ps_mod._core.zero_copy_recv_count()
+    n = 1 << 16  # 256 KB > inline budget -> must ride by-ref
+    buf = ps_mod.host_alloc(n * 4)
+    buf.copy_from(np.ones(n, dtype=np.float32))
+    keys = np.array([11], dtype=np.uint64)
+    lens = np.array([n], dtype=np.int32)
+    w.wait(w.zpush_ptr(keys, buf.ptr, n * 4, -1, lens, cmd=2))
+    out = w.pull(keys)
+    assert np.allclose(out, 1.0)
+    # the worker-side counter counts its own received by-ref responses;
+    # the PUSH was received by the server process, so probe via a second
+    # push from this process to itself? No: in separate-process mode we
+    # can only check our own receptions — the pull response must have
+    # been delivered in place (kOptInPlace) or by-ref
+    return int(ps_mod._core.zero_copy_recv_count() - before)
+
+
+def test_zero_copy_reception():
+    results = launch_local(1, 1, _zero_copy_worker, timeout=180)
+    # worker side alone sees >= 0; the real assertion runs in joint mode
+    # below where worker and server share a process
+    assert results[0] >= 0
+
+
+def _zero_copy_joint_worker(ps_mod, rank):
+    server = ps_mod.KVServer(0)
+    server.set_default_handle()
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    w = ps_mod.KVWorker(0, 0)
+    before = ps_mod._core.zero_copy_recv_count()
+    n = 1 << 16
+    buf = ps_mod.host_alloc(n * 4)
+    buf.copy_from(np.full(n, 2.0, dtype=np.float32))
+    keys = np.array([12], dtype=np.uint64)
+    lens = np.array([n], dtype=np.int32)
+    w.wait(w.zpush_ptr(keys, buf.ptr, n * 4, -1, lens, cmd=2))
+    got = w.pull(keys)
+    assert np.allclose(got, 2.0)
+    after = ps_mod._core.zero_copy_recv_count()
+    # joint process: the server's reception of our 256 KB push MUST have
+    # been by-reference into the mapped host pool (zero-copy)
+    assert after - before >= 1, (before, after)
+    return True, server
+
+
+def test_zero_copy_reception_joint():
+    results = launch_local(1, 1, _zero_copy_joint_worker, joint=True, timeout=180)
+    assert results[0] is True
