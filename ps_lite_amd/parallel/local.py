@@ -105,3 +105,58 @@ def launch_local(num_workers, num_servers, worker_fn, server_fn=None, env_extra=
     c = LocalCluster(num_workers, num_servers, env_extra=env_extra, joint=joint,
                      devices=devices)
     return c.run(worker_fn, server_fn=server_fn, timeout=timeout, worker_args=worker_args)
+
+
+def main():
+    """CLI launcher (reference parity: tracker/dmlc_local.py): run an
+    arbitrary app command as S servers + W workers + scheduler on this
+    host, with the keepalive restart loop (retry a nonzero-exit process
+    up to --retries times)."""
+    import argparse
+    import subprocess
+    import sys
+    import threading
+
+    p = argparse.ArgumentParser(description=main.__doc__)
+    p.add_argument("--workers", type=int, required=True)
+    p.add_argument("--servers", type=int, required=True)
+    p.add_argument("--root-port", type=int, default=9100)
+    p.add_argument("--retries", type=int, default=3,
+                   help="keepalive restarts per process (dmlc_local.py:15-23)")
+    p.add_argument("cmd", nargs=argparse.REMAINDER)
+    a = p.parse_args()
+    cmd = " ".join(c for c in a.cmd if c != "--")
+    base = {
+        "DMLC_NUM_WORKER": str(a.workers),
+        "DMLC_NUM_SERVER": str(a.servers),
+        "DMLC_PS_ROOT_URI": "127.0.0.1",
+        "DMLC_PS_ROOT_PORT": str(a.root_port),
+    }
+    rcs = []
+
+    def run(role, rank):
+        env = dict(os.environ, **base, DMLC_ROLE=role)
+        if rank >= 0:
+            env["DMLC_RANK"] = str(rank)
+        for _ in range(max(1, a.retries)):
+            rc = subprocess.call(cmd, shell=True, env=env)
+            if rc == 0:
+                rcs.append(0)
+                return
+        rcs.append(rc)
+
+    sched = subprocess.Popen(
+        [sys.executable, "-c",
+         "import ps_lite_amd as ps; ps.start(role='scheduler', device=-1); "
+         "ps.finalize(role='scheduler')"],
+        env=dict(os.environ, **base, DMLC_ROLE="scheduler"))
+    threads = [threading.Thread(target=run, args=("server", r)) for r in range(a.servers)]
+    threads += [threading.Thread(target=run, args=("worker", r)) for r in range(a.workers)]
+    [t.start() for t in threads]
+    [t.join() for t in threads]
+    sched.wait()
+    sys.exit(0 if all(rc == 0 for rc in rcs) else 1)
+
+
+if __name__ == "__main__":
+    main()

@@ -79,3 +79,13 @@ def test_mpi_launcher_end_to_end(tmp_path):
                        capture_output=True, text=True, timeout=180)
     assert p.returncode == 0, (p.stdout[-2000:], p.stderr[-2000:])
     assert "WORKER_OK" in p.stdout, (p.stdout[-2000:], p.stderr[-2000:])
+
+
+def test_local_cli_launcher_end_to_end(tmp_path):
+    cmd = [sys.executable, "-m", "ps_lite_amd.parallel.local",
+           "--workers", "1", "--servers", "1", "--root-port", "24981",
+           "--", sys.executable, "tests/_launcher_app.py"]
+    p = subprocess.run(cmd, cwd=REPO, env=_clean_env(str(tmp_path)),
+                       capture_output=True, text=True, timeout=180)
+    assert p.returncode == 0, (p.stdout[-2000:], p.stderr[-2000:])
+    assert "WORKER_OK" in p.stdout, (p.stdout[-2000:], p.stderr[-2000:])
