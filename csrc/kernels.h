@@ -20,8 +20,6 @@ void DenseAssign(void* dst, const void* src, size_t nbytes, hipStream_t s);
 void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s);
 // dst[i] += sum_j srcs[j][i], up to 8 sources in one pass (one read of
 // dst, one write — HBM-optimal multi-worker reduction)
-void DenseSumMultiF32(float* dst, const float* const* srcs_host, int nsrc, size_t n,
-                      hipStream_t s);
 // Sparse ops: local row index = (rows[r] >> key_shift) - row_base, so a
 // server can index its local table shard from globally-sharded keys
 // (key = global_row << key_shift spreads rows over the PS key space).

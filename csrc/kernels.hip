@@ -62,26 +62,6 @@ __global__ void sum_tail_f32(float* __restrict__ dst, const float* __restrict__ 
   if (i < end) dst[i] += src[i];
 }
 
-struct SrcList {
-  const float4* p[8];
-};
-
-__global__ void sum_multi_f32(float4* __restrict__ dst, SrcList srcs, int nsrc, size_t n4) {
-  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
-  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
-  for (; i < n4; i += stride) {
-    float4 d = dst[i];
-    for (int j = 0; j < nsrc; ++j) {
-      float4 s = srcs.p[j][i];
-      d.x += s.x;
-      d.y += s.y;
-      d.z += s.z;
-      d.w += s.w;
-    }
-    dst[i] = d;
-  }
-}
-
 // flat indexing: thread i handles element i of the CONCATENATED rows
 // (row = i / row_len4), so narrow rows (e.g. 64 floats = 16 float4)
 // still use every lane — a row-per-block mapping left 94 % of the block
@@ -250,20 +230,6 @@ void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s) {
   }
   if (n % 4) {
     hipLaunchKernelGGL(sum_tail_f32, dim3(1), dim3(kBlock), 0, s, dst, src, n4 * 4, n);
-  }
-}
-
-void DenseSumMultiF32(float* dst, const float* const* srcs_host, int nsrc, size_t n,
-                      hipStream_t s) {
-  SrcList list{};
-  for (int i = 0; i < nsrc && i < 8; ++i) list.p[i] = reinterpret_cast<const float4*>(srcs_host[i]);
-  size_t n4 = n / 4;
-  hipLaunchKernelGGL(sum_multi_f32, dim3(GridFor(n4)), dim3(kBlock), 0, s,
-                     reinterpret_cast<float4*>(dst), list, nsrc, n4);
-  for (int j = 0; j < nsrc; ++j) {
-    if (n % 4) {
-      hipLaunchKernelGGL(sum_tail_f32, dim3(1), dim3(kBlock), 0, s, dst, srcs_host[j], n4 * 4, n);
-    }
   }
 }
 

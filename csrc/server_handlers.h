@@ -88,6 +88,10 @@ class GpuDenseHandler {
   hipStream_t fallback_stream_ = nullptr;
 };
 
+// NOTE on key uniqueness: a single sparse push message must not repeat a
+// row (ps-lite's sorted-unique key contract); duplicate rows across
+// MESSAGES are fine (atomic scatters with >1 workers; handler-serialized
+// otherwise).
 class GpuSparseHandler {
  public:
   // Allocates (and zeroes) a rows x row_len fp32 table shard in the pool.
