@@ -34,3 +34,15 @@ def test_dp_allreduce_two_workers():
     for step in range(2):
         for g0, g1 in zip(results[0][step], results[1][step]):
             assert np.allclose(np.array(g0), np.array(g1), atol=1e-6)
+
+
+def test_train_dp_example():
+    """The examples/python/train_dp.py script must run and converge."""
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    p = subprocess.run([sys.executable, os.path.join(repo, "examples/python/train_dp.py")],
+                       capture_output=True, text=True, timeout=280, cwd=repo)
+    assert p.returncode == 0, (p.stdout[-1500:], p.stderr[-1500:])
+    assert "converged" in p.stdout
