@@ -14,6 +14,7 @@
 #include <algorithm>
 #include <atomic>
 #include <functional>
+#include <set>
 #include <unordered_map>
 #include <vector>
 
@@ -515,6 +516,117 @@ struct KVServerDefaultHandle {
     server->Response(req, res);
   }
   std::unordered_map<Key, std::vector<V>> store;
+};
+
+// CPU reduce-round handle: the BytePS round semantics of the GPU
+// DenseMode::kReduce handler (server_handlers.cc) with synchronous host
+// arithmetic — the first push of a round assigns, the rest accumulate,
+// pulls are HELD until every worker pushed, the round resets after every
+// worker pulled, and next-round traffic arriving early is deferred. Runs
+// the same protocol the GPU path uses, minus streams/events, so the
+// round logic is testable without a GPU (and `bench.py --cpu` can run
+// the rn50 reduce mode).
+template <typename V>
+struct KVServerReduceHandle {
+  explicit KVServerReduceHandle(int num_workers) : num_workers_(std::max(1, num_workers)) {}
+
+  void operator()(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
+    if (req.push) {
+      HandlePush(req, kvs, server);
+    } else if (req.pull) {
+      HandlePull(req, kvs, server);
+    } else {
+      server->Response(req);
+    }
+  }
+
+  struct Entry {
+    std::vector<V> buf;
+    int pushes = 0;
+    int pulls = 0;
+    std::vector<KVMeta> waiting_pulls;
+    std::vector<KVMeta> waiting_next_pulls;
+    std::vector<std::pair<KVMeta, KVPairs<V>>> waiting_pushes;
+    std::set<int> pulled;
+  };
+  std::unordered_map<Key, Entry> store;
+  int num_workers_;
+
+ private:
+  void HandlePush(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
+    XPS_CHECK_EQ(kvs.keys.size(), 1u) << "reduce mode is single-key-per-message";
+    XPS_CHECK(!kvs.vals.on_device()) << "KVServerReduceHandle is CPU-only";
+    Entry& e = store[kvs.keys[0]];
+    if (e.pushes >= num_workers_) {
+      // a fast worker started the next round before this round's pulls
+      // drained (the KVPairs copy keeps the payload alive)
+      e.waiting_pushes.emplace_back(req, kvs);
+      return;
+    }
+    size_t len = kvs.lens.empty() ? kvs.vals.size() : static_cast<size_t>(kvs.lens[0]);
+    if (e.buf.size() < len) e.buf.resize(len, V(0));
+    const V* __restrict__ v = kvs.vals.data();
+    V* __restrict__ b = e.buf.data();
+    if (e.pushes == 0) {
+      memcpy(b, v, len * sizeof(V));
+    } else {
+      for (size_t j = 0; j < len; ++j) b[j] += v[j];
+    }
+    e.pushes++;
+    server->Response(req);
+    if (e.pushes >= num_workers_) {
+      std::vector<KVMeta> waiting;
+      waiting.swap(e.waiting_pulls);
+      for (auto& w : waiting) RespondPull(w, kvs.keys[0], server);
+    }
+  }
+
+  void HandlePull(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
+    XPS_CHECK_EQ(kvs.keys.size(), 1u) << "reduce mode is single-key-per-message";
+    // a pull may precede the round's pushes (it just waits)
+    Entry& e = store[kvs.keys[0]];
+    if (e.pulled.count(req.sender)) {
+      e.waiting_next_pulls.push_back(req);  // next-round pull, too early
+      return;
+    }
+    if (e.pushes < num_workers_) {
+      e.waiting_pulls.push_back(req);  // released by the round's last push
+      return;
+    }
+    RespondPull(req, kvs.keys[0], server);
+  }
+
+  void RespondPull(const KVMeta& req, Key key, KVServer<V>* server) {
+    Entry& e = store[key];
+    KVPairs<V> res;
+    res.keys = SArray<Key>({key});
+    // synchronous handler: the response is serialized before we return,
+    // so a view of buf is safe (and zero-copy on the shm plane)
+    res.vals = SArray<V>(e.buf.data(), e.buf.size(), kCPU);
+    SArray<int> lens(1);
+    lens[0] = static_cast<int>(e.buf.size());
+    res.lens = lens;
+    server->Response(req, res);
+    e.pulled.insert(req.sender);
+    e.pulls++;
+    if (e.pulls >= num_workers_) {
+      e.pushes = 0;
+      e.pulls = 0;
+      e.pulled.clear();
+      std::vector<std::pair<KVMeta, KVPairs<V>>> dpush;
+      dpush.swap(e.waiting_pushes);
+      for (auto& d : dpush) HandlePush(d.first, d.second, server);
+      std::vector<KVMeta> dpull;
+      dpull.swap(e.waiting_next_pulls);
+      for (auto& d : dpull) {
+        if (e.pushes >= num_workers_) {
+          RespondPull(d, key, server);
+        } else {
+          e.waiting_pulls.push_back(d);
+        }
+      }
+    }
+  }
 };
 
 }  // namespace xps

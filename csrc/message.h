@@ -107,6 +107,12 @@ struct Meta {
   int64_t val_len = 0;         // vals bytes
   int option = 0;
   uint64_t msg_sig = 0;        // resender signature (0 = none)
+  // per-(sender -> recver) data-message sequence number (1-based; 0 =
+  // unsequenced). A flow's messages may split between the shm/xGMI
+  // plane and the TCP fallback by payload kind; the receiver re-orders
+  // on this so per-peer FIFO holds across transports (the reference's
+  // UCX sid-reordering guarantee, ucx_van.h:1217-1257).
+  uint64_t seq = 0;
   // per-data-blob types (parallel to Message::data)
   std::vector<int> data_type;
   // device ordinal the vals blob lives on at the SENDER (-1 host);

@@ -205,6 +205,14 @@ class PyKVServer {
     });
   }
 
+  // CPU BytePS reduce rounds (same protocol as the GPU reduce handler)
+  void SetReduceHandle(int num_workers) {
+    auto h = std::make_shared<KVServerReduceHandle<float>>(num_workers);
+    s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
+      (*h)(m, kvs, srv);
+    });
+  }
+
   // checkpoint/resume of the installed handler's server state
   void SaveCheckpoint(const std::string& path) {
     py::gil_scoped_release rel;
@@ -488,6 +496,7 @@ PYBIND11_MODULE(_core, m) {
   py::class_<PyKVServer>(m, "KVServer")
       .def(py::init<int, int>(), py::arg("app_id") = 0, py::arg("instance_idx") = 0)
       .def("set_default_handle", &PyKVServer::SetDefaultHandle)
+      .def("set_reduce_handle", &PyKVServer::SetReduceHandle, py::arg("num_workers"))
       .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("mode") = "assign")
       .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),
            py::arg("row_len"), py::arg("accumulate") = true, py::arg("key_shift") = 0)
@@ -591,6 +600,7 @@ PYBIND11_MODULE(_core, m) {
     m.addr = 1234567;
     m.val_len = 1 << 20;
     m.option = kOptPullAddr | kOptHostAddr;
+    m.seq = 987654321;
     m.data_type = {kUint64, kFloat32, kInt32};
     m.control.cmd = Control::ADD_NODE;
     Node n;
@@ -616,6 +626,7 @@ PYBIND11_MODULE(_core, m) {
               out.recver == m.recver && out.request == m.request && out.push == m.push &&
               out.head == m.head && out.body == m.body && out.key == m.key &&
               out.addr == m.addr && out.val_len == m.val_len && out.option == m.option &&
+              out.seq == m.seq &&
               out.data_type == m.data_type && out.control.cmd == m.control.cmd &&
               out.control.node.size() == 1;
     auto& on = out.control.node[0];

@@ -282,9 +282,9 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   if (mode_ == DenseMode::kReduce) {
     XPS_CHECK_EQ(n, 1u) << "reduce mode is single-key-per-message";
     std::lock_guard<std::mutex> lk(mu_);
-    auto it = store_.find(kvs.keys[0]);
-    XPS_CHECK(it != store_.end()) << "pull of unknown key " << kvs.keys[0];
-    Entry* e = &it->second;
+    // a pull may legitimately precede the round's pushes (it just
+    // waits); create the entry on demand
+    Entry* e = &store_[kvs.keys[0]];
     if (e->pulled_senders.count(req.sender)) {
       // this sender already pulled the current round: a NEXT-round pull
       e->waiting_next_pulls.push_back(req);

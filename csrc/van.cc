@@ -163,6 +163,10 @@ int64_t Van::Send(Message& msg) {
 
 int64_t Van::SendToNode(Message& msg, int id) {
   MaybeTrace(msg, /*recv=*/false);
+  if (id != my_node_.id && msg.meta.control.empty() && msg.meta.seq == 0) {
+    std::lock_guard<std::mutex> lk(order_mu_);
+    msg.meta.seq = ++send_seq_[id];
+  }
   if (id == my_node_.id) {
     // loopback without touching the wire
     Message copy = msg;
@@ -311,7 +315,95 @@ void Van::Deliver(Message msg) {
   }
 }
 
+// single execution lane for sequenced flows: handler runs on the
+// DELIVERING thread (plane poll thread or TCP recv thread), serialized
+// per customer by handle_mu_ — a queued lane for one transport and an
+// inline lane for the other would reorder right after the gate
+void Van::InlineData(Message& msg) {
+  int app_id = msg.meta.app_id;
+  int customer_id = po_->is_worker() ? msg.meta.customer_id : app_id;
+  Customer* c = po_->GetCustomer(app_id, customer_id, 15);
+  XPS_CHECK(c) << "no customer (app=" << app_id << ", customer=" << customer_id
+               << ") on node " << my_node_.id << " for " << msg.DebugString();
+  c->ProcessInline(msg);
+}
+
+void Van::GatedDeliver(Message& msg) {
+  uint64_t seq = msg.meta.seq;
+  Flow* f;
+  {
+    std::lock_guard<std::mutex> lk(order_mu_);
+    auto& slot = recv_flows_[msg.meta.sender];
+    if (!slot) slot.reset(new Flow());
+    f = slot.get();
+  }
+  // f->mu is held ACROSS handler execution: release order and execution
+  // order are then the same thing, even when one message arrived over
+  // the shm ring (inline poll thread) and its predecessor over the TCP
+  // fallback (recv thread)
+  std::unique_lock<std::mutex> lk(f->mu);
+  if (seq < f->expected) {
+    // late duplicate (resend whose original arrived): deliver — the
+    // resender's signature dedup upstream already filtered true dups
+    InlineData(msg);
+    return;
+  }
+  if (f->held.empty()) f->hold_since = std::chrono::steady_clock::now();
+  f->held.emplace(seq, std::move(msg));
+  while (!f->held.empty() && f->held.begin()->first == f->expected) {
+    Message m = std::move(f->held.begin()->second);
+    f->held.erase(f->held.begin());
+    f->expected++;
+    InlineData(m);
+    f->hold_since = std::chrono::steady_clock::now();
+  }
+  if (f->held.empty()) return;
+  // a later message overtook its predecessor on the other transport and
+  // the gap has not filled. Liveness guard: without a resender a lost
+  // message would stall the flow forever — flush in order after a
+  // timeout (ordering is then best-effort, matching a lossy fault).
+  int timeout_ms = Environment::Get()->GetInt("XPS_ORDER_TIMEOUT_MS", 10000);
+  if (std::chrono::steady_clock::now() - f->hold_since >
+      std::chrono::milliseconds(timeout_ms)) {
+    XPS_LOG(Warning) << "sequence gap from node "
+                     << f->held.begin()->second.meta.sender << " (expected " << f->expected
+                     << ") unfilled for " << timeout_ms << " ms; flushing out of order";
+    while (!f->held.empty()) {
+      Message m = std::move(f->held.begin()->second);
+      uint64_t s2 = f->held.begin()->first;
+      f->held.erase(f->held.begin());
+      f->expected = s2 + 1;
+      InlineData(m);
+    }
+  }
+}
+
+void Van::ResetFlow(int peer_id) {
+  Flow* f = nullptr;
+  {
+    std::lock_guard<std::mutex> lk(order_mu_);
+    send_seq_.erase(peer_id);
+    auto it = recv_flows_.find(peer_id);
+    if (it != recv_flows_.end()) f = it->second.get();
+  }
+  if (f) {
+    // reset in place: erasing would destroy a mutex another thread may
+    // be holding mid-delivery
+    std::lock_guard<std::mutex> lk(f->mu);
+    f->expected = 1;
+    f->held.clear();
+  }
+}
+
 void Van::DeliverData(Message& msg) {
+  if (msg.meta.seq != 0) {
+    GatedDeliver(msg);
+    return;
+  }
+  DeliverDataNow(msg);
+}
+
+void Van::DeliverDataNow(Message& msg) {
   int app_id = msg.meta.app_id;
   int customer_id = po_->is_worker() ? msg.meta.customer_id : app_id;
   Customer* c = po_->GetCustomer(app_id, customer_id, 15);
@@ -325,11 +417,15 @@ void Van::DeliverInline(Message& msg) {
     ProcessControl(msg, nullptr);
     return;
   }
+  if (msg.meta.seq != 0) {
+    GatedDeliver(msg);
+    return;
+  }
   int app_id = msg.meta.app_id;
   int customer_id = po_->is_worker() ? msg.meta.customer_id : app_id;
   Customer* c = po_->GetCustomer(app_id, customer_id, 15);
-  XPS_CHECK(c) << "no customer (app=" << app_id << ", customer=" << customer_id << ") on node "
-               << my_node_.id << " for " << msg.DebugString();
+  XPS_CHECK(c) << "no customer (app=" << app_id << ", customer=" << customer_id
+               << ") on node " << my_node_.id << " for " << msg.DebugString();
   c->ProcessInline(msg);
 }
 
@@ -547,12 +643,15 @@ void Van::ProcessNodeListAssigned(Message& msg) {
         // the id was re-assigned (recovery): drop the stale connection so
         // the next send redials the new address. (shm_uid==0 means a
         // pre-bootstrap stub record, NOT a re-assignment.)
-        std::lock_guard<std::mutex> lk2(conn_mu_);
-        auto cit = conns_.find(n.id);
-        if (cit != conns_.end()) {
-          cit->second->Close();
-          conns_.erase(cit);
+        {
+          std::lock_guard<std::mutex> lk2(conn_mu_);
+          auto cit = conns_.find(n.id);
+          if (cit != conns_.end()) {
+            cit->second->Close();
+            conns_.erase(cit);
+          }
         }
+        ResetFlow(n.id);  // the replacement node starts a fresh sequence
       }
       nodes_[n.id] = n;
       if (n.shm_uid == my_uid_ && n.role == my_node_.role) {

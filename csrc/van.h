@@ -13,6 +13,8 @@
 #include <memory>
 #include <random>
 #include <thread>
+#include <chrono>
+#include <map>
 #include <unordered_map>
 #include <vector>
 #include <functional>
@@ -84,6 +86,12 @@ class Van {
   void ProcessBarrierAtScheduler(Message& msg);
   void ProcessHeartbeat(Message& msg);
   void DeliverData(Message& msg);
+  void DeliverDataNow(Message& msg);
+  void InlineData(Message& msg);
+  // per-sender FIFO across transports: a sequenced data message is
+  // delivered (inline, in seq order, under the flow's lock) or held
+  void GatedDeliver(Message& msg);
+  void ResetFlow(int peer_id);  // recovery: peer restarted with fresh counters
   void HeartbeatLoop();
 
   Postoffice* po_;
@@ -93,6 +101,16 @@ class Van {
   TcpListener listener_;
 
   std::mutex conn_mu_;
+  // cross-transport ordering state (see Meta::seq)
+  std::mutex order_mu_;  // guards send_seq_ + the flow map (not delivery)
+  std::unordered_map<int, uint64_t> send_seq_;
+  struct Flow {
+    std::mutex mu;  // held ACROSS delivery so seq order = execution order
+    uint64_t expected = 1;
+    std::map<uint64_t, Message> held;
+    std::chrono::steady_clock::time_point hold_since;
+  };
+  std::unordered_map<int, std::unique_ptr<Flow>> recv_flows_;
   std::unordered_map<int, std::shared_ptr<TcpConn>> conns_;  // node id -> conn
   std::vector<std::thread> recv_threads_;
   std::atomic<bool> stopping_{false};

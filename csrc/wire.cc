@@ -2,7 +2,7 @@
 
 namespace xps {
 
-static const uint8_t kWireVersion = 1;
+static const uint8_t kWireVersion = 2;
 
 static void PackNode(const Node& n, ByteWriter* w) {
   w->I32(n.role);
@@ -61,6 +61,7 @@ void PackMeta(const Meta& m, std::string* out) {
   w.I64(m.val_len);
   w.I32(m.option);
   w.U64(m.msg_sig);
+  w.U64(m.seq);
   w.I32(m.src_dev);
   w.I32(m.dst_dev);
   w.U8(static_cast<uint8_t>(m.data_type.size()));
@@ -98,6 +99,7 @@ void UnpackMeta(const char* buf, size_t len, Meta* m) {
   m->val_len = r.I64();
   m->option = r.I32();
   m->msg_sig = r.U64();
+  m->seq = r.U64();
   m->src_dev = r.I32();
   m->dst_dev = r.I32();
   int nt = r.U8();

@@ -272,3 +272,55 @@ def _zero_copy_joint_worker(ps_mod, rank):
 def test_zero_copy_reception_joint():
     results = launch_local(1, 1, _zero_copy_joint_worker, joint=True, timeout=180)
     assert results[0] is True
+
+
+def _cpu_reduce_worker(ps_mod, rank):
+    """BytePS reduce rounds on CPU (KVServerReduceHandle): overlapped
+    push+pull, no inter-worker barrier — same protocol as the GPU
+    reduce handler, testable without a GPU."""
+    server = ps_mod.KVServer(0)
+    server.set_reduce_handle(num_workers=2)
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    w = ps_mod.KVWorker(0, 0)
+    n = 2048
+    keys = np.array([21], dtype=np.uint64)
+    lens = np.array([n], dtype=np.int32)
+    outs = []
+    for step in range(5):  # rounds exercise reset + deferred replay
+        vals = np.full(n, float(rank + 1 + step), dtype=np.float32)
+        ts1 = w.push(keys, vals, lens)
+        out = w.pull(keys)  # overlapped: pull issued while pushes in flight
+        w.wait(ts1)
+        outs.append(float(out[0]))
+        assert np.allclose(out, out[0])
+    return outs, server
+
+
+def test_cpu_reduce_rounds_two_workers():
+    results = launch_local(2, 2, _cpu_reduce_worker, joint=True, timeout=300)
+    # round k: workers push (1+k) and (2+k) -> both pull 3+2k
+    for rank, outs in results.items():
+        assert outs == [3.0 + 2 * k for k in range(5)], results
+
+
+def _ordering_worker(ps_mod, rank):
+    """Cross-transport FIFO: an 8 KB push rides the TCP fallback (host
+    heap, > inline budget) while the pull request rides the shm ring —
+    without sequence gating the pull overtakes the push and reads
+    zeros/stale. Meta.seq + the receive-side gate must serialize them."""
+    w = ps_mod.KVWorker(0, 0)
+    n = 2048  # 8 KB > kInlineMax
+    lens = np.array([n], dtype=np.int32)
+    for it in range(10):
+        keys = np.array([500 + it], dtype=np.uint64)
+        vals = np.full(n, float(it + 1), dtype=np.float32)
+        ts = w.push(keys, vals, lens)  # fresh key: sum(0 + vals) == vals
+        got = w.pull(keys)  # issued immediately: must NOT overtake the push
+        w.wait(ts)
+        assert np.allclose(got, vals), (it, got[:3].tolist())
+    return True
+
+
+def test_cross_transport_ordering():
+    results = launch_local(1, 1, _ordering_worker, timeout=240)
+    assert results[0] is True
