@@ -324,3 +324,33 @@ def _ordering_worker(ps_mod, rank):
 def test_cross_transport_ordering():
     results = launch_local(1, 1, _ordering_worker, timeout=240)
     assert results[0] is True
+
+
+def _mixed_drop_worker(ps_mod, rank):
+    """Mixed transports + fault injection: small messages ride the
+    lossless shm plane, 8 KB ones the TCP fallback where PS_DROP_MSG
+    drops some. A dropped TCP push leaves a sequence gap; the gate must
+    HOLD later ring messages until the resend fills it — order and
+    values both intact."""
+    w = ps_mod.KVWorker(0, 0)
+    big, small = 2048, 128  # 8 KB (TCP) / 512 B (inline -> ring)
+    total = 0.0
+    for it in range(12):
+        n = big if it % 2 == 0 else small
+        keys = np.array([900], dtype=np.uint64)
+        lens = np.array([n], dtype=np.int32)
+        w.wait(w.push(keys, np.full(n, 1.0, dtype=np.float32), lens))
+        total += 1.0
+    out = w.pull(np.array([900], dtype=np.uint64))
+    # the store entry grows to `big`; elements [0, small) saw every push,
+    # the tail only the big ones
+    assert out.shape[0] == big
+    assert np.allclose(out[:small], total), out[:4].tolist()
+    assert np.allclose(out[small:], 6.0), out[small:small + 4].tolist()
+    return True
+
+
+def test_ordering_with_drops_and_resend():
+    env = {"PS_RESEND": "1", "PS_RESEND_TIMEOUT": "200", "PS_DROP_MSG": "20"}
+    results = launch_local(1, 1, _mixed_drop_worker, env_extra=env, timeout=240)
+    assert results[0] is True
