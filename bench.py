@@ -120,7 +120,7 @@ def main():
     from ps_lite_amd.models import EmbeddingSpec, resnet50_grad_buckets
 
     if args.cpu:
-        assert args.mode == "dense", "--cpu supports the dense mode (config #1)"
+        assert args.mode in ("dense", "rn50"), "--cpu supports dense and rn50"
         device = -1
     else:
         count = ps.gpu_count()
@@ -171,7 +171,9 @@ def main():
     trace("cluster up")
 
     server = ps.KVServer(0)
-    if args.cpu:
+    if args.cpu and args.mode == "rn50":
+        server.set_reduce_handle(num_workers=n)  # CPU BytePS rounds
+    elif args.cpu:
         server.set_default_handle()
     elif args.mode == "sparse":
         spec = EmbeddingSpec(rows=args.emb_rows, width=args.emb_width)

@@ -31,3 +31,17 @@ def test_rn50_buckets_shape():
     buckets = resnet50_grad_buckets()
     assert sum(buckets) == total * 4
     assert max(buckets) <= 4 << 20
+
+
+def test_bench_cpu_rn50_contract():
+    """`bench.py --cpu --mode rn50` runs BytePS reduce rounds on the CPU
+    reduce handle and emits the JSON contract line."""
+    out = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--cpu", "--mode", "rn50",
+         "--smoke", "--no-rtt"],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    j = json.loads(lines[0])
+    assert j["config"]["mode"] == "rn50-cpu"
+    assert j["value"] > 0
