@@ -256,7 +256,9 @@ char* GpuPlane::ResolvePeer(Peer* p, uint64_t global_off, uint64_t len) {
   size_t idx = static_cast<size_t>(global_off / slab_bytes);
   uint64_t local = global_off % slab_bytes;
   std::lock_guard<std::mutex> lk(p->mu);
-  if (idx >= p->slab_bases.size() || local + len > slab_bytes) {
+  // compare against the remaining room, never local + len (which can
+  // wrap for a hostile/corrupt length)
+  if (idx >= p->slab_bases.size() || len > slab_bytes - local) {
     XPS_LOG(Warning) << "peer offset out of bounds: off=" << global_off << " len=" << len
                      << " slab=" << idx;
     return nullptr;
@@ -385,7 +387,8 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
       (msg.meta.option & kOptHostAddr)) {
     void* base = HostShmPool::MapPeer(p->node.host_pool_uid, p->node.host_pool_capacity);
     SArray<char> vals = msg.data[1];
-    if (base && msg.meta.addr + vals.size() <= p->node.host_pool_capacity) {
+    if (base && msg.meta.addr <= p->node.host_pool_capacity &&
+        vals.size() <= p->node.host_pool_capacity - msg.meta.addr) {
       memcpy(static_cast<char*>(base) + msg.meta.addr, vals.data(), vals.size());
       Message meta_msg;
       meta_msg.meta = msg.meta;
@@ -594,7 +597,8 @@ void GpuPlane::RingPollLoop() {
         void* base = sender ? HostShmPool::MapPeer(sender->node.host_pool_uid,
                                                    sender->node.host_pool_capacity)
                             : nullptr;
-        if (!base || off + len > sender->node.host_pool_capacity) {
+        if (!base || off > sender->node.host_pool_capacity ||
+            len > sender->node.host_pool_capacity - off) {
           XPS_LOG(Warning) << "dropping host-ref blob: sender host pool not mapped (from "
                            << msg.meta.sender << ")";
           ok = false;
