@@ -10,6 +10,7 @@
 #include <cstring>
 
 #include "hip_pool.h"
+#include "host_par.h"
 #include "host_pool.h"
 #include "kernels.h"
 #include "postoffice.h"
@@ -389,7 +390,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     SArray<char> vals = msg.data[1];
     if (base && msg.meta.addr <= p->node.host_pool_capacity &&
         vals.size() <= p->node.host_pool_capacity - msg.meta.addr) {
-      memcpy(static_cast<char*>(base) + msg.meta.addr, vals.data(), vals.size());
+      HostPar::CopyBytes(static_cast<char*>(base) + msg.meta.addr, vals.data(), vals.size());
       Message meta_msg;
       meta_msg.meta = msg.meta;
       meta_msg.meta.option |= kOptInPlace;

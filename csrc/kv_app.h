@@ -20,6 +20,7 @@
 
 #include "hip_pool.h"
 #include "hip_util.h"
+#include "host_par.h"
 #include "host_pool.h"
 #include "simple_app.h"
 
@@ -468,6 +469,12 @@ class KVServer : public SimpleApp {
 // Default CPU handle: store[key] op= vals (sum by default, memcpy for
 // cmd=kCmdAssign); pull echoes the store (ps-lite kv_app.h:431-452
 // behavior, plus the assign op the EmptyHandler benchmark semantics use).
+inline void SumInto(float* dst, const float* src, size_t n) { HostPar::SumF32(dst, src, n); }
+template <typename V>
+inline void SumInto(V* dst, const V* src, size_t n) {
+  for (size_t j = 0; j < n; ++j) dst[j] += src[j];
+}
+
 template <typename V>
 struct KVServerDefaultHandle {
   void operator()(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
@@ -481,11 +488,9 @@ struct KVServerDefaultHandle {
         auto& entry = store[kvs.keys[i]];
         if (entry.size() < len) entry.resize(len, V(0));
         if (req.cmd == kCmdAssign) {
-          memcpy(entry.data(), kvs.vals.data() + off, len * sizeof(V));
+          HostPar::CopyBytes(entry.data(), kvs.vals.data() + off, len * sizeof(V));
         } else {
-          V* __restrict__ e = entry.data();
-          const V* __restrict__ v = kvs.vals.data() + off;
-          for (size_t j = 0; j < len; ++j) e[j] += v[j];
+          SumInto(entry.data(), kvs.vals.data() + off, len);
         }
         off += len;
       }
@@ -568,9 +573,9 @@ struct KVServerReduceHandle {
     const V* __restrict__ v = kvs.vals.data();
     V* __restrict__ b = e.buf.data();
     if (e.pushes == 0) {
-      memcpy(b, v, len * sizeof(V));
+      HostPar::CopyBytes(b, v, len * sizeof(V));
     } else {
-      for (size_t j = 0; j < len; ++j) b[j] += v[j];
+      SumInto(b, v, len);
     }
     e.pushes++;
     server->Response(req);
