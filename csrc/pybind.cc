@@ -208,6 +208,7 @@ class PyKVServer {
   // CPU BytePS reduce rounds (same protocol as the GPU reduce handler)
   void SetReduceHandle(int num_workers) {
     auto h = std::make_shared<KVServerReduceHandle<float>>(num_workers);
+    reduce_ = h;
     s_.set_request_handle([h](const KVMeta& m, const KVPairs<float>& kvs, KVServer<float>* srv) {
       (*h)(m, kvs, srv);
     });
@@ -220,16 +221,24 @@ class PyKVServer {
       dense_->Save(path);
     } else if (sparse_) {
       sparse_->Save(path);
-    } else if (default_) {
+    } else if (default_ || reduce_) {
       FILE* f = fopen(path.c_str(), "wb");
       XPS_CHECK(f) << "cannot open checkpoint " << path;
-      uint64_t n = default_->store.size();
+      uint64_t n = default_ ? default_->store.size() : reduce_->store.size();
       fwrite(&n, 8, 1, f);
-      for (auto& kv : default_->store) {
-        uint64_t key = kv.first, len = kv.second.size() * sizeof(float);
+      auto write_one = [f](uint64_t key, const float* data, uint64_t len) {
         fwrite(&key, 8, 1, f);
         fwrite(&len, 8, 1, f);
-        fwrite(kv.second.data(), 1, len, f);
+        fwrite(data, 1, len, f);
+      };
+      if (default_) {
+        for (auto& kv : default_->store) {
+          write_one(kv.first, kv.second.data(), kv.second.size() * sizeof(float));
+        }
+      } else {
+        for (auto& kv : reduce_->store) {
+          write_one(kv.first, kv.second.buf.data(), kv.second.buf.size() * sizeof(float));
+        }
       }
       fclose(f);
     } else {
@@ -243,7 +252,7 @@ class PyKVServer {
       dense_->Load(path);
     } else if (sparse_) {
       sparse_->Load(path);
-    } else if (default_) {
+    } else if (default_ || reduce_) {
       FILE* f = fopen(path.c_str(), "rb");
       XPS_CHECK(f) << "cannot open checkpoint " << path;
       uint64_t n = 0;
@@ -252,7 +261,8 @@ class PyKVServer {
         uint64_t key, len;
         XPS_CHECK_EQ(fread(&key, 8, 1, f), 1u);
         XPS_CHECK_EQ(fread(&len, 8, 1, f), 1u);
-        auto& v = default_->store[key];
+        auto& v = default_ ? default_->store[key]
+                           : reduce_->store[key].buf;
         v.resize(len / sizeof(float));
         XPS_CHECK_EQ(fread(v.data(), 1, len, f), len);
       }
@@ -326,6 +336,7 @@ class PyKVServer {
   std::shared_ptr<GpuSparseHandler> sparse_;
   std::shared_ptr<GpuDenseHandler> dense_;
   std::shared_ptr<KVServerDefaultHandle<float>> default_;
+  std::shared_ptr<KVServerReduceHandle<float>> reduce_;
 };
 
 class PySimpleApp {

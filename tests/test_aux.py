@@ -81,3 +81,40 @@ def test_stage_timing_table(capfd):
     assert results[0] is True
     err = capfd.readouterr().err
     assert "stage timing" in err and "plane_send" in err, err[-2000:]
+
+
+def _reduce_ckpt_worker(ps, rank):
+    worker = ps.KVWorker(0, 0)
+    keys = np.array([33], dtype=np.uint64)
+    vals = np.full(16, 5.0, dtype=np.float32)
+    lens = np.array([16], dtype=np.int32)
+    worker.wait(worker.push(keys, vals, lens))
+    out1 = worker.pull(keys).tolist()  # completes the 1-worker round
+    ps.barrier("worker", ps.SERVER_GROUP | ps.WORKER_GROUP)  # server ckpts
+    ps.barrier("worker", ps.SERVER_GROUP | ps.WORKER_GROUP)  # reloaded as default
+    out2 = worker.pull(keys).tolist()  # default handle serves it directly
+    return [out1, out2]
+
+
+def _reduce_ckpt_server(ps, rank):
+    tmpdir = os.environ["XPS_TEST_TMPDIR"]
+    path = os.path.join(tmpdir, "reduce_ckpt.bin")
+    server = ps.KVServer(0)
+    server.set_reduce_handle(num_workers=1)
+    ps.barrier("server", ps.SERVER_GROUP | ps.WORKER_GROUP)
+    server.save_checkpoint(path)
+    # the on-disk format is shared: swap the SAME server to the default
+    # handle and reload (reduce rounds are transient; the stored KV is
+    # what counts)
+    server.set_default_handle()
+    server.load_checkpoint(path)
+    ps.barrier("server", ps.SERVER_GROUP | ps.WORKER_GROUP)
+    return None, server
+
+
+def test_reduce_checkpoint_into_default_handle(tmp_path):
+    env = {"XPS_TEST_TMPDIR": str(tmp_path)}
+    results = launch_local(1, 1, _reduce_ckpt_worker, server_fn=_reduce_ckpt_server,
+                           env_extra=env, timeout=180)
+    out1, out2 = results[0]
+    assert np.allclose(out1, 5.0) and np.allclose(out2, 5.0)
