@@ -665,6 +665,53 @@ PYBIND11_MODULE(_core, m) {
     PackMeta(m, &buf);
     return py::bytes(buf);
   });
+  // property-test hook: build a Meta from fields, round-trip through the
+  // wire, compare (hypothesis drives the field values)
+  m.def("_meta_roundtrip_fields",
+        [](int app_id, int customer_id, int timestamp, int sender, int recver, bool request,
+           bool push, bool pull, bool simple_app, int head, const std::string& body, uint64_t key,
+           uint64_t addr, int64_t val_len, int option, uint64_t msg_sig, uint64_t seq,
+           const std::string& hostname, int port, uint64_t shm_uid, int nhandles) {
+          Meta m;
+          m.app_id = app_id;
+          m.customer_id = customer_id;
+          m.timestamp = timestamp;
+          m.sender = sender;
+          m.recver = recver;
+          m.request = request;
+          m.push = push;
+          m.pull = pull;
+          m.simple_app = simple_app;
+          m.head = head;
+          m.body = body;
+          m.key = key;
+          m.addr = addr;
+          m.val_len = val_len;
+          m.option = option;
+          m.msg_sig = msg_sig;
+          m.seq = seq;
+          m.control.cmd = Control::ADD_NODE;
+          Node n;
+          n.hostname = hostname;
+          n.port = port;
+          n.shm_uid = shm_uid;
+          n.pool_handles.resize(nhandles);
+          m.control.node.push_back(n);
+          std::string buf;
+          PackMeta(m, &buf);
+          Meta o;
+          UnpackMeta(buf.data(), buf.size(), &o);
+          auto& on = o.control.node.at(0);
+          return o.app_id == m.app_id && o.customer_id == m.customer_id &&
+                 o.timestamp == m.timestamp && o.sender == m.sender && o.recver == m.recver &&
+                 o.request == m.request && o.push == m.push && o.pull == m.pull &&
+                 o.simple_app == m.simple_app && o.head == m.head && o.body == m.body &&
+                 o.key == m.key && o.addr == m.addr && o.val_len == m.val_len &&
+                 o.option == m.option && o.msg_sig == m.msg_sig && o.seq == m.seq &&
+                 on.hostname == n.hostname && on.port == n.port && on.shm_uid == n.shm_uid &&
+                 on.pool_handles.size() == static_cast<size_t>(nhandles);
+        });
+
   m.def("_unpack_meta_raw", [](py::bytes b) {
     std::string s = b;
     Meta out;

@@ -107,3 +107,29 @@ def test_partition_tensor():
     servers = {int(k) // step for k in keys}
     assert servers == {0, 1, 2, 3}
     # a worker can push/pull with them directly (end-to-end)
+
+
+def test_wire_property_roundtrip():
+    """Property test: arbitrary Meta field values survive the wire
+    byte-stream round trip exactly (hypothesis-driven)."""
+    from hypothesis import given, settings, strategies as st
+
+    i32 = st.integers(min_value=-(2**31), max_value=2**31 - 1)
+    u64 = st.integers(min_value=0, max_value=2**64 - 1)
+    i64 = st.integers(min_value=-(2**63), max_value=2**63 - 1)
+
+    @settings(max_examples=200, deadline=None)
+    @given(app=i32, cust=i32, ts=i32, snd=i32, rcv=i32,
+           req=st.booleans(), push=st.booleans(), pull=st.booleans(),
+           sapp=st.booleans(), head=i32,
+           body=st.text(max_size=64).map(lambda t: t.encode("utf-8", "ignore")[:64].decode("utf-8", "ignore")),
+           key=u64, addr=u64, vlen=i64, opt=i32, sig=u64, seq=u64,
+           host=st.text(alphabet="abc123.-", max_size=32), port=i32,
+           uid=u64, nh=st.integers(min_value=0, max_value=40))
+    def check(app, cust, ts, snd, rcv, req, push, pull, sapp, head, body,
+              key, addr, vlen, opt, sig, seq, host, port, uid, nh):
+        assert ps._core._meta_roundtrip_fields(
+            app, cust, ts, snd, rcv, req, push, pull, sapp, head, body,
+            key, addr, vlen, opt, sig, seq, host, port, uid, nh)
+
+    check()
