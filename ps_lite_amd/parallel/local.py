@@ -48,7 +48,10 @@ class LocalCluster:
         self.num_servers = num_servers
         self.joint = joint
         self.devices = devices or {}
-        port = root_port or random.randint(20000, 50000)
+        # stay BELOW the kernel's ephemeral range (default 32768+): a
+        # transient outgoing socket from another process can otherwise
+        # hold the scheduler's port and fail its bind
+        port = root_port or random.randint(20000, 32000)
         self.env = {
             "DMLC_NUM_WORKER": str(num_workers),
             "DMLC_NUM_SERVER": str(num_servers),
@@ -102,9 +105,18 @@ class LocalCluster:
 
 def launch_local(num_workers, num_servers, worker_fn, server_fn=None, env_extra=None,
                  joint=False, timeout=120, worker_args=(), devices=None):
-    c = LocalCluster(num_workers, num_servers, env_extra=env_extra, joint=joint,
-                     devices=devices)
-    return c.run(worker_fn, server_fn=server_fn, timeout=timeout, worker_args=worker_args)
+    last = None
+    for _ in range(3):  # a busy random port fails the scheduler bind: retry fresh
+        c = LocalCluster(num_workers, num_servers, env_extra=env_extra, joint=joint,
+                         devices=devices)
+        try:
+            return c.run(worker_fn, server_fn=server_fn, timeout=timeout,
+                         worker_args=worker_args)
+        except RuntimeError as e:
+            if "failed to bind port" not in str(e):
+                raise
+            last = e
+    raise last
 
 
 def main():
