@@ -67,3 +67,45 @@ def test_dense_reduce_two_workers():
     results = launch_local(2, 2, _dense_reduce_worker, timeout=300)
     for rank, bad in results.items():
         assert bad == [], (rank, bad)
+
+
+def _reduce_chaos_worker(ps, rank):
+    """4 workers x 4 keys x 8 reduce rounds with random per-worker
+    delays: exercises every deferral path of the round protocol (pushes
+    arriving before the previous round drained, next-round pulls,
+    replay after reset) under chaotic timing."""
+    import random
+    import time as _t
+
+    nw, nkeys, rounds, width = 4, 4, 8, 64
+    server = ps.KVServer(0)
+    server.set_reduce_handle(num_workers=nw)
+    ps.barrier("worker", ps.WORKER_GROUP)
+    w = ps.KVWorker(0, 0)
+    rng = random.Random(1000 + rank)
+    step = (1 << 64) // nw
+    keys = [np.array([s * step + 7], dtype=np.uint64) for s in range(nkeys)]
+    lens = np.array([width], dtype=np.int32)
+    bad = []
+    for r in range(rounds):
+        tss = []
+        outs = []
+        for k in range(nkeys):
+            if rng.random() < 0.5:
+                _t.sleep(rng.random() * 0.01)
+            val = float((rank + 1) * 1000 + r)
+            tss.append(w.push(keys[k], np.full(width, val, dtype=np.float32), lens))
+            outs.append(w.pull(keys[k]))  # overlapped; held until all pushed
+        for ts in tss:
+            w.wait(ts)
+        expect = sum((i + 1) * 1000 + r for i in range(nw))
+        for k, out in enumerate(outs):
+            if not np.allclose(out, float(expect)):
+                bad.append((r, k, float(out[0]), expect))
+    return bad, server
+
+
+def test_reduce_round_chaos_four_workers():
+    results = launch_local(4, 4, _reduce_chaos_worker, joint=True, timeout=420)
+    for rank, bad in results.items():
+        assert bad == [], (rank, bad)
