@@ -107,7 +107,7 @@ void Van::Start(int customer_id) {
 
   // wait until the node list is assigned / all nodes joined
   auto deadline = std::chrono::steady_clock::now() + std::chrono::seconds(
-      Environment::Get()->GetInt("XPS_BOOTSTRAP_TIMEOUT", 120));
+      Environment::Get()->GetInt("XPS_BOOTSTRAP_TIMEOUT", 300));
   while (!ready_.load()) {
     XPS_CHECK(std::chrono::steady_clock::now() < deadline)
         << "bootstrap timeout: " << my_node_.DebugString();
