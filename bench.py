@@ -52,6 +52,9 @@ def parse_args():
     p.add_argument("--op", choices=["assign", "sum"], default="assign")
     p.add_argument("--batch-keys", action="store_true",
                    help="dense: one multi-key message per server per round")
+    p.add_argument("--per-key", action="store_true",
+                   help="rn50: legacy one-message-per-bucket reduce rounds "
+                        "(default is one multi-key message per server per round)")
     p.add_argument("--no-overlap", action="store_true",
                    help="dense: separate push and pull phases (the reference "
                         "PUSH_PULL loop issues ZPush+ZPull per key together, "
@@ -226,46 +229,73 @@ def main():
     else:
         keys_np = np.array(keys, dtype=np.uint64)
         batch = args.batch_keys and args.mode == "dense" and device >= 0
-        push_bufs, pull_bufs = [], []
+        batch_reduce = args.mode == "rn50" and not args.per_key
         alloc = ps.pool_alloc if device >= 0 else ps.host_alloc
-        for sz in msg_sizes:
-            # host buffers come from the shm pool so the CPU config rides
-            # the same-host zero-copy plane (not TCP)
-            b = alloc(sz)
-            b.copy_from(rng.standard_normal(sz // 4).astype(np.float32))
-            push_bufs.append(b)
-            pull_bufs.append(alloc(sz))
-        push_ptrs = [b.ptr for b in push_bufs]
-        pull_ptrs = [b.ptr for b in pull_bufs]
-        uniform = len(set(msg_sizes)) == 1
-        if batch:
-            k = len(keys) // n
-            srv_keys = [np.sort(keys_np[keys_np // np.uint64(step_range) == s])
-                        for s in range(n)]
-            blens = np.full(k, msg_sizes[0] // 4, dtype=np.int32)
-            bpush = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
-            bpull = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
-            for b in bpush:
-                b.copy_from(rng.standard_normal(k * msg_sizes[0] // 4).astype(np.float32))
+        if batch_reduce:
+            # BytePS bucketed DenseReduce: ONE multi-key ZPush + ONE
+            # multi-key ZPull of every bucket per step (the slicer splits
+            # them into one message per server; the server runs one
+            # batched kernel chain per round instead of 169 per-key
+            # dispatches). Buckets are laid out contiguously in sorted
+            # key order (the slicer requires sorted keys).
+            order = np.argsort(keys_np, kind="stable")
+            skeys = keys_np[order]
+            ssizes = [msg_sizes[i] for i in order]
+            lens_el = np.array([s // 4 for s in ssizes], dtype=np.int32)
+            total_b = int(sum(ssizes))
+            bpush_r = alloc(total_b)
+            bpull_r = alloc(total_b)
+            bpush_r.copy_from(rng.standard_normal(total_b // 4).astype(np.float32))
 
-        def one_step():
+            def one_step():
+                tsp = worker.zpush_ptr(skeys, bpush_r.ptr, total_b, device, lens_el, cmd=cmd)
+                tsq = worker.zpull_ptr(skeys, bpull_r.ptr, total_b, device, lens_el, cmd=cmd)
+                worker.wait(tsp)
+                worker.wait(tsq)
+
+            bytes_per_worker_step = 2.0 * total_msg_bytes
+            push_bufs, pull_bufs = [], []
+            push_ptrs, pull_ptrs = [], []
+        else:
+            push_bufs, pull_bufs = [], []
+            for sz in msg_sizes:
+                # host buffers come from the shm pool so the CPU config
+                # rides the same-host zero-copy plane (not TCP)
+                b = alloc(sz)
+                b.copy_from(rng.standard_normal(sz // 4).astype(np.float32))
+                push_bufs.append(b)
+                pull_bufs.append(alloc(sz))
+            push_ptrs = [b.ptr for b in push_bufs]
+            pull_ptrs = [b.ptr for b in pull_bufs]
+            uniform = len(set(msg_sizes)) == 1
             if batch:
-                tss = [worker.zpush_ptr(srv_keys[s], bpush[s].ptr, k * msg_sizes[0], device,
-                                        blens, cmd=cmd) for s in range(n)]
-                for ts in tss:
-                    worker.wait(ts)
-                tss = [worker.zpull_ptr(srv_keys[s], bpull[s].ptr, k * msg_sizes[0], device,
-                                        blens, cmd=cmd) for s in range(n)]
-                for ts in tss:
-                    worker.wait(ts)
-            elif uniform and not overlap_pull:
-                worker.round(keys_np, push_ptrs, msg_sizes[0], device, cmd, False)
-                worker.round(keys_np, pull_ptrs, msg_sizes[0], device, cmd, True)
-            else:
-                worker.round_mixed(keys_np, push_ptrs, pull_ptrs, msg_sizes, device, cmd,
-                                   overlap_pull)
+                k = len(keys) // n
+                srv_keys = [np.sort(keys_np[keys_np // np.uint64(step_range) == s])
+                            for s in range(n)]
+                blens = np.full(k, msg_sizes[0] // 4, dtype=np.int32)
+                bpush = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
+                bpull = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
+                for b in bpush:
+                    b.copy_from(rng.standard_normal(k * msg_sizes[0] // 4).astype(np.float32))
 
-        bytes_per_worker_step = 2.0 * total_msg_bytes
+            def one_step():
+                if batch:
+                    tss = [worker.zpush_ptr(srv_keys[s], bpush[s].ptr, k * msg_sizes[0], device,
+                                            blens, cmd=cmd) for s in range(n)]
+                    for ts in tss:
+                        worker.wait(ts)
+                    tss = [worker.zpull_ptr(srv_keys[s], bpull[s].ptr, k * msg_sizes[0], device,
+                                            blens, cmd=cmd) for s in range(n)]
+                    for ts in tss:
+                        worker.wait(ts)
+                elif uniform and not overlap_pull:
+                    worker.round(keys_np, push_ptrs, msg_sizes[0], device, cmd, False)
+                    worker.round(keys_np, pull_ptrs, msg_sizes[0], device, cmd, True)
+                else:
+                    worker.round_mixed(keys_np, push_ptrs, pull_ptrs, msg_sizes, device, cmd,
+                                       overlap_pull)
+
+            bytes_per_worker_step = 2.0 * total_msg_bytes
 
     trace("buffers ready")
     for w in range(args.warmup):

@@ -139,8 +139,11 @@ void GpuPlane::Stop() {
       for (auto& p : kv.second) {
         (void)hipEventSynchronize(p.ev);
         Peer* peer = GetPeer(p.peer_id);
-        if (peer && EnsureRing(peer)) {
-          peer->ring.Push(p.payload.data(), static_cast<uint32_t>(p.payload.size()));
+        bool sent = peer && EnsureRing(peer) &&
+                    peer->ring.Push(p.payload.data(), static_cast<uint32_t>(p.payload.size()));
+        if (!sent) {
+          // van TCP conns are still open (plane stops first): best-effort
+          (void)po_->van()->SendOverTcp(p.resend, p.peer_id);
         }
         (void)hipEventDestroy(p.ev);
       }
@@ -210,6 +213,37 @@ bool GpuPlane::ImportPeerSlabs(Peer* p) {
   bool same_process =
       pool->initialized() && nslabs == pool->slab_count() &&
       memcmp(p->node.pool_handles[0].data(), pool->slab_handle(0), kIpcHandleBytes) == 0;
+  // Cross-DEVICE import: our kernels will load/store the mapping over
+  // xGMI, which needs peer access from our device to the exporter's.
+  // Enable it eagerly at bootstrap — a lazy failure would surface as a
+  // memory fault inside a steady-state kernel. peer dev_id is the peer
+  // PROCESS's device index; it translates directly when all ranks see
+  // the same HIP_VISIBLE_DEVICES ordering (the one-proc-per-GPU layout).
+  int peer_dev = p->node.dev_id;
+  if (!same_process && peer_dev >= 0 && peer_dev != device_ &&
+      Environment::Get()->GetInt("XPS_PEER_ACCESS_CHECK", 1)) {
+    int ndev = 0;
+    (void)hipGetDeviceCount(&ndev);
+    if (peer_dev < ndev) {
+      int can = 0;
+      (void)hipDeviceCanAccessPeer(&can, device_, peer_dev);
+      if (!can) {
+        XPS_LOG(Warning) << "device " << device_ << " cannot peer-access device " << peer_dev
+                         << " (peer node " << p->node.id
+                         << "): keeping this peer on the TCP path";
+        std::lock_guard<std::mutex> lk(p->mu);
+        p->slab_bases.clear();
+        return false;
+      }
+      XPS_HIP_CHECK(hipSetDevice(device_));
+      hipError_t pe = hipDeviceEnablePeerAccess(peer_dev, 0);
+      if (pe != hipSuccess && pe != hipErrorPeerAccessAlreadyEnabled) {
+        XPS_LOG(Warning) << "hipDeviceEnablePeerAccess(" << peer_dev
+                         << ") failed: " << hipGetErrorString(pe)
+                         << " (continuing; hipIpc lazy enable may still cover it)";
+      }
+    }
+  }
   for (size_t i = 0; i < nslabs; ++i) {
     void* base = nullptr;
     if (same_process) {
@@ -232,8 +266,21 @@ bool GpuPlane::ImportPeerSlabs(Peer* p) {
         XPS_VLOG(3) << "ipc-open slab " << i << " of peer " << p->node.id << " -> " << base;
         if (e != hipSuccess) {
           XPS_LOG(Warning) << "hipIpcOpenMemHandle(peer " << p->node.id << " slab " << i
-                           << ") failed: " << hipGetErrorString(e);
+                           << ") failed: " << hipGetErrorString(e)
+                           << " — peer stays on the TCP path";
           base = nullptr;
+        } else {
+          // probe the mapping NOW (one 8-byte D2H read): a dead or
+          // access-less mapping must fail loudly at bootstrap, not as a
+          // memory fault inside a steady-state kernel
+          char probe[8];
+          hipError_t e2 = hipMemcpy(probe, base, sizeof(probe), hipMemcpyDeviceToHost);
+          if (e2 != hipSuccess) {
+            XPS_LOG(Warning) << "peer " << p->node.id << " slab " << i
+                             << " mapped but unreadable (" << hipGetErrorString(e2)
+                             << ") — peer stays on the TCP path";
+            base = nullptr;
+          }
         }
         g_mapped[key] = base;
       }
@@ -447,11 +494,15 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
       XPS_CHECK(Serialize(meta_msg, {}, &payload));
     }
     int64_t bytes = static_cast<int64_t>(vals.size() + payload.size());
-    msg.data.clear();  // keepalive: vals (and the store view) live in meta_msg? no —
-    // the original msg owns vals; keep it alive until the event fires
+    msg.data.clear();
+    // the original msg owns vals; keep it alive until the event fires.
+    // meta_msg (kOptInPlace, host blobs only) is the TCP-resendable form:
+    // by the time a fallback happens the event has fired, i.e. the
+    // in-place write already landed in the peer's pool.
     Message keepalive;
     keepalive.data.push_back(vals);
-    DeferSendInternal(p, peer_node.id, std::move(keepalive), std::move(payload), bytes);
+    DeferSendInternal(p, peer_node.id, std::move(meta_msg), std::move(keepalive),
+                      std::move(payload), bytes);
     return bytes;
   }
 
@@ -475,9 +526,10 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   }
   int64_t bytes = static_cast<int64_t>(payload.size()) + ref_bytes;
   if (response && device_ >= 0) {
-    // order behind any handler kernels on this peer's stream
-    Message keepalive = msg;
-    DeferSendInternal(p, peer_node.id, std::move(keepalive), std::move(payload), bytes);
+    // order behind any handler kernels on this peer's stream. The resend
+    // copy doubles as the keepalive (shallow: SArrays shared).
+    Message resend = msg;
+    DeferSendInternal(p, peer_node.id, std::move(resend), Message(), std::move(payload), bytes);
   } else {
     // requests — and every send of a host-only plane — go out now (host
     // responses were produced synchronously; nothing to wait for)
@@ -487,8 +539,8 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   return bytes;
 }
 
-void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::string payload,
-                                 int64_t bytes) {
+void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message resend, Message keepalive,
+                                 std::string payload, int64_t bytes) {
   XPS_STAGE(defer_queue);
   XPS_CHECK_GE(device_, 0) << "deferred sends are a GPU-plane feature";
   hipStream_t stream = StreamForPeer(peer_id);
@@ -497,8 +549,8 @@ void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::s
   XPS_HIP_CHECK(hipEventRecord(ev, stream));
   {
     std::lock_guard<std::mutex> lk(pend_mu_);
-    pending_[peer_id].push_back(
-        Pending{ev, peer_id, std::move(payload), std::move(keepalive), bytes});
+    pending_[peer_id].push_back(Pending{ev, peer_id, std::move(payload), std::move(resend),
+                                        std::move(keepalive), bytes});
   }
   pending_count_.fetch_add(1);
   XPS_VLOG(3) << "deferred send queued -> " << peer_id;
@@ -513,6 +565,9 @@ void GpuPlane::CompletionLoop() {
   int idle = 0;
   while (!stop_.load()) {
     bool did = false;
+    // failed ring pushes are replayed over TCP AFTER releasing pend_mu_
+    // (dialing can block); seq gating at the receiver restores order
+    std::vector<std::pair<int, Message>> tcp_fallback;
     if (pending_count_.load() > 0) {
       std::lock_guard<std::mutex> lk(pend_mu_);
       for (auto& kv : pending_) {
@@ -524,17 +579,28 @@ void GpuPlane::CompletionLoop() {
           XPS_STAGE(defer_release);
           XPS_CHECK(e == hipSuccess) << "hipEventQuery: " << hipGetErrorString(e);
           Peer* peer = GetPeer(front.peer_id);
-          if (peer && EnsureRing(peer)) {
-            peer->ring.Push(front.payload.data(), static_cast<uint32_t>(front.payload.size()));
+          bool sent = peer && EnsureRing(peer) &&
+                      peer->ring.Push(front.payload.data(),
+                                      static_cast<uint32_t>(front.payload.size()));
+          if (sent) {
             XPS_VLOG(3) << "deferred send done -> " << front.peer_id;
           } else {
-            XPS_LOG(Warning) << "deferred send DROPPED -> " << front.peer_id;
+            // never drop: the requester is blocked in Wait on this
+            // response — deliver it over the TCP path instead
+            XPS_LOG(Warning) << "deferred plane send failed -> " << front.peer_id
+                             << "; falling back to TCP";
+            tcp_fallback.emplace_back(front.peer_id, std::move(front.resend));
           }
           PutEvent(front.ev);
           dq.pop_front();
           pending_count_.fetch_sub(1);
           did = true;
         }
+      }
+    }
+    for (auto& f : tcp_fallback) {
+      if (po_->van()->SendOverTcp(f.second, f.first) < 0) {
+        XPS_LOG(Warning) << "TCP fallback to " << f.first << " failed too (peer dead?)";
       }
     }
     if (did) {

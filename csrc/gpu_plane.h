@@ -74,6 +74,11 @@ class GpuPlane : public DataPlane {
     hipEvent_t ev;
     int peer_id;
     std::string payload;
+    // TCP-resendable form of this message: if the ring push fails after
+    // the event fires, the van sends this instead (never drop a response
+    // — the requester is blocked in Wait). For in-place writes this is
+    // the meta-only kOptInPlace message (the data already landed).
+    Message resend;
     Message keepalive;  // holds pool temporaries until the event fires
     int64_t bytes;
   };
@@ -90,8 +95,8 @@ class GpuPlane : public DataPlane {
   bool Serialize(const Message& msg, const std::vector<char>& by_ref, std::string* out);
   void RingPollLoop();
   void CompletionLoop();
-  void DeferSendInternal(Peer* p, int peer_id, Message keepalive, std::string payload,
-                         int64_t bytes);
+  void DeferSendInternal(Peer* p, int peer_id, Message resend, Message keepalive,
+                         std::string payload, int64_t bytes);
 
   Postoffice* po_;
   int device_;

@@ -56,8 +56,33 @@ void HbmPool::Init(int device, size_t capacity_bytes) {
     slabs_.push_back(std::move(s));
   }
   capacity_ = global;
+  exported_slabs_ = slabs_.size();
   XPS_VLOG(1) << "HbmPool: " << (capacity_ >> 20) << " MiB in " << slabs_.size()
               << " slabs on device " << device_;
+}
+
+bool HbmPool::GrowLocked() {
+  Slab s;
+  s.capacity = slab_bytes_;
+  if (hipSetDevice(device_) != hipSuccess) return false;
+  hipError_t e = hipMalloc(&s.base, s.capacity);
+  if (e != hipSuccess) {
+    XPS_LOG(Warning) << "HbmPool growth failed: hipMalloc(" << s.capacity
+                     << ") -> " << hipGetErrorString(e);
+    return false;
+  }
+  // growth slabs live OUTSIDE the advertised global-offset space: peers
+  // never learned their ipc handles, so OffsetOf must not match them
+  // (by-ref sends from here fall back to TCP staging)
+  s.global_begin = ~0ull;
+  s.exported = false;
+  s.free_[0] = s.capacity;
+  slabs_.push_back(std::move(s));
+  XPS_LOG(Warning) << "HbmPool grown by " << (slab_bytes_ >> 20)
+                   << " MiB (local-only slab " << slabs_.size() - 1
+                   << "; cross-process zero-copy uses the bootstrap slabs — raise "
+                      "XPS_POOL_GB to keep everything on the fast path)";
+  return true;
 }
 
 void* HbmPool::Alloc(size_t nbytes) {
@@ -67,24 +92,30 @@ void* HbmPool::Alloc(size_t nbytes) {
       << "single allocation exceeds the slab size (" << slab_bytes_
       << " B); raise XPS_SLAB_BYTES (< 2 GiB) or split the buffer";
   std::lock_guard<std::mutex> lk(mu_);
-  for (auto& slab : slabs_) {
-    for (auto it = slab.free_.begin(); it != slab.free_.end(); ++it) {
-      if (it->second >= nbytes) {
-        size_t off = it->first;
-        size_t rest = it->second - nbytes;
-        slab.free_.erase(it);
-        if (rest) slab.free_[off + nbytes] = rest;
-        slab.used_[off] = nbytes;
-        return static_cast<char*>(slab.base) + off;
+  for (int attempt = 0; attempt < 2; ++attempt) {
+    for (auto& slab : slabs_) {
+      for (auto it = slab.free_.begin(); it != slab.free_.end(); ++it) {
+        if (it->second >= nbytes) {
+          size_t off = it->first;
+          size_t rest = it->second - nbytes;
+          slab.free_.erase(it);
+          if (rest) slab.free_[off + nbytes] = rest;
+          slab.used_[off] = nbytes;
+          return static_cast<char*>(slab.base) + off;
+        }
       }
     }
+    // exhausted: grow by one local-only slab and retry once (288 GB of
+    // HBM3E means the device almost always has room — only the
+    // bootstrap-advertised zero-copy window is fixed)
+    if (attempt == 0 && !GrowLocked()) break;
   }
   size_t in_use = 0;
   for (auto& slab : slabs_) {
     for (auto& kv : slab.used_) in_use += kv.second;
   }
   XPS_LOG(Fatal) << "HbmPool exhausted: want " << nbytes << " bytes, capacity " << capacity_
-                 << " (in use " << in_use << "); raise XPS_POOL_GB";
+                 << " (in use " << in_use << ") and growth failed; raise XPS_POOL_GB";
   return nullptr;
 }
 
@@ -122,6 +153,8 @@ bool HbmPool::OffsetOf(const void* p, uint64_t* global_off) const {
   for (auto& slab : slabs_) {
     const char* b = static_cast<const char*>(slab.base);
     if (c >= b && c < b + slab.capacity) {
+      // growth slabs are unknown to peers: no wire-referenceable offset
+      if (!slab.exported) return false;
       *global_off = slab.global_begin + static_cast<uint64_t>(c - b);
       return true;
     }

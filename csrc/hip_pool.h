@@ -37,7 +37,8 @@ class HbmPool {
   int device() const { return device_; }
   size_t capacity() const { return capacity_; }
 
-  size_t slab_count() const { return slabs_.size(); }
+  // bootstrap-exported slabs only (growth slabs are local, see below)
+  size_t slab_count() const { return exported_slabs_; }
   size_t slab_bytes() const { return slab_bytes_; }
   uint64_t slab_capacity(size_t i) const { return slabs_[i].capacity; }
   const char* slab_handle(size_t i) const { return slabs_[i].ipc_handle; }
@@ -63,14 +64,24 @@ class HbmPool {
     uint64_t capacity = 0;
     uint64_t global_begin = 0;  // global offset of byte 0
     char ipc_handle[64] = {0};
+    // slabs allocated AFTER bootstrap (on-demand growth) are not in any
+    // peer's slab table, so they must never be referenced by global
+    // offset on the wire — OffsetOf skips them and traffic from them
+    // takes the TCP staging path (slow but correct)
+    bool exported = true;
     std::map<size_t, size_t> free_;  // local offset -> size
     std::map<size_t, size_t> used_;
   };
+
+  // allocate one more (non-exported) slab; called on exhaustion.
+  // Returns false if hipMalloc fails. Caller holds mu_.
+  bool GrowLocked();
 
   HbmPool() = default;
   int device_ = -1;
   size_t capacity_ = 0;
   size_t slab_bytes_ = kDefaultSlabBytes;
+  size_t exported_slabs_ = 0;  // slabs advertised at bootstrap
   mutable std::mutex mu_;
   std::vector<Slab> slabs_;
 };

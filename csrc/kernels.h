@@ -23,17 +23,22 @@ void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s);
 // Sparse ops: local row index = (rows[r] >> key_shift) - row_base, so a
 // server can index its local table shard from globally-sharded keys
 // (key = global_row << key_shift spreads rows over the PS key space).
+// table_rows bounds every decoded row index: out-of-range rows (corrupt
+// or misrouted keys) are skipped by scatters and read as zeros by
+// gathers, never touching memory outside the table. The ~0 default
+// disables the check (trusted keys).
 // out[r][:] = table[row(r)][:] for r in [0, nrows)
 void SparseGatherF32(const float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                     float* out, hipStream_t s, int key_shift = 0, uint64_t row_base = 0);
+                     float* out, hipStream_t s, int key_shift = 0, uint64_t row_base = 0,
+                     uint64_t table_rows = ~0ull);
 // table[row(r)][:] += src[r][:]; atomic=true tolerates duplicate rows
 void SparseScatterAddF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
                          const float* src, bool atomic, hipStream_t s, int key_shift = 0,
-                         uint64_t row_base = 0);
+                         uint64_t row_base = 0, uint64_t table_rows = ~0ull);
 // table[row(r)][:] = src[r][:]
 void SparseScatterAssignF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
                             const float* src, hipStream_t s, int key_shift = 0,
-                            uint64_t row_base = 0);
+                            uint64_t row_base = 0, uint64_t table_rows = ~0ull);
 
 // Batched segmented copy/accumulate: one launch serving up to
 // kMaxBatch (dst, src, nbytes) segments — the device-side slicing/merge

@@ -65,30 +65,39 @@ __global__ void sum_tail_f32(float* __restrict__ dst, const float* __restrict__ 
 // flat indexing: thread i handles element i of the CONCATENATED rows
 // (row = i / row_len4), so narrow rows (e.g. 64 floats = 16 float4)
 // still use every lane — a row-per-block mapping left 94 % of the block
-// idle at width 64
+// idle at width 64.
+// Every kernel bounds-checks the decoded local row against tab_rows: a
+// misrouted or corrupt key must not scribble over the shared pool (or a
+// hipIpc-mapped peer). The unsigned compare also catches keys below
+// row_base (the subtraction wraps to a huge value). Gather returns
+// zeros for such rows (defined output); scatters skip them.
 __global__ void gather_rows_f32(const float4* __restrict__ table, const uint64_t* __restrict__ rows,
                                 size_t nrows, size_t row_len4, float4* __restrict__ out,
-                                int shift, uint64_t base) {
+                                int shift, uint64_t base, uint64_t tab_rows) {
   size_t total = nrows * row_len4;
   size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (; i < total; i += stride) {
     size_t r = i / row_len4;
     size_t c = i - r * row_len4;
-    out[i] = table[((rows[r] >> shift) - base) * row_len4 + c];
+    uint64_t row = (rows[r] >> shift) - base;
+    out[i] = row < tab_rows ? table[row * row_len4 + c] : float4{0.f, 0.f, 0.f, 0.f};
   }
 }
 
 __global__ void scatter_add_rows_f32(float4* __restrict__ table, const uint64_t* __restrict__ rows,
                                      size_t nrows, size_t row_len4,
-                                     const float4* __restrict__ src, int shift, uint64_t base) {
+                                     const float4* __restrict__ src, int shift, uint64_t base,
+                                     uint64_t tab_rows) {
   size_t total = nrows * row_len4;
   size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (; i < total; i += stride) {
     size_t r = i / row_len4;
     size_t c = i - r * row_len4;
-    float4* dst = table + ((rows[r] >> shift) - base) * row_len4 + c;
+    uint64_t row = (rows[r] >> shift) - base;
+    if (row >= tab_rows) continue;
+    float4* dst = table + row * row_len4 + c;
     float4 d = *dst;
     float4 v = src[i];
     d.x += v.x;
@@ -102,34 +111,40 @@ __global__ void scatter_add_rows_f32(float4* __restrict__ table, const uint64_t*
 __global__ void gather_rows_scalar_f32(const float* __restrict__ table,
                                        const uint64_t* __restrict__ rows, size_t nrows,
                                        size_t row_len, float* __restrict__ out, int shift,
-                                       uint64_t base) {
+                                       uint64_t base, uint64_t tab_rows) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    const float* src = table + ((rows[r] >> shift) - base) * row_len;
+    uint64_t row = (rows[r] >> shift) - base;
+    const float* src = table + row * row_len;
     float* dst = out + r * row_len;
-    for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) dst[c] = src[c];
+    for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) {
+      dst[c] = row < tab_rows ? src[c] : 0.f;
+    }
   }
 }
 
 __global__ void scatter_assign_rows_f32(float4* __restrict__ table,
                                         const uint64_t* __restrict__ rows, size_t nrows,
                                         size_t row_len4, const float4* __restrict__ src,
-                                        int shift, uint64_t base) {
+                                        int shift, uint64_t base, uint64_t tab_rows) {
   size_t total = nrows * row_len4;
   size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (; i < total; i += stride) {
     size_t r = i / row_len4;
     size_t c = i - r * row_len4;
-    table[((rows[r] >> shift) - base) * row_len4 + c] = src[i];
+    uint64_t row = (rows[r] >> shift) - base;
+    if (row < tab_rows) table[row * row_len4 + c] = src[i];
   }
 }
 
 __global__ void scatter_assign_rows_scalar_f32(float* __restrict__ table,
                                                const uint64_t* __restrict__ rows, size_t nrows,
                                                size_t row_len, const float* __restrict__ src,
-                                               int shift, uint64_t base) {
+                                               int shift, uint64_t base, uint64_t tab_rows) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float* dst = table + ((rows[r] >> shift) - base) * row_len;
+    uint64_t row = (rows[r] >> shift) - base;
+    if (row >= tab_rows) continue;
+    float* dst = table + row * row_len;
     const float* s = src + r * row_len;
     for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) dst[c] = s[c];
   }
@@ -138,9 +153,11 @@ __global__ void scatter_assign_rows_scalar_f32(float* __restrict__ table,
 __global__ void scatter_add_rows_atomic_f32(float* __restrict__ table,
                                             const uint64_t* __restrict__ rows, size_t nrows,
                                             size_t row_len, const float* __restrict__ src,
-                                            int shift, uint64_t base) {
+                                            int shift, uint64_t base, uint64_t tab_rows) {
   for (size_t r = blockIdx.x; r < nrows; r += gridDim.x) {
-    float* dst = table + ((rows[r] >> shift) - base) * row_len;
+    uint64_t row = (rows[r] >> shift) - base;
+    if (row >= tab_rows) continue;
+    float* dst = table + row * row_len;
     const float* s = src + r * row_len;
     for (size_t c = threadIdx.x; c < row_len; c += blockDim.x) {
       atomicAdd(&dst[c], s[c]);
@@ -234,41 +251,45 @@ void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s) {
 }
 
 void SparseGatherF32(const float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                     float* out, hipStream_t s, int key_shift, uint64_t row_base) {
+                     float* out, hipStream_t s, int key_shift, uint64_t row_base,
+                     uint64_t table_rows) {
   if (row_len % 4 == 0) {
     hipLaunchKernelGGL(gather_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)), dim3(kBlock),
                        0, s, reinterpret_cast<const float4*>(table), rows_dev, nrows, row_len / 4,
-                       reinterpret_cast<float4*>(out), key_shift, row_base);
+                       reinterpret_cast<float4*>(out), key_shift, row_base, table_rows);
   } else {
     hipLaunchKernelGGL(gather_rows_scalar_f32, dim3(GridFor(nrows * row_len, 16384)), dim3(kBlock),
-                       0, s, table, rows_dev, nrows, row_len, out, key_shift, row_base);
+                       0, s, table, rows_dev, nrows, row_len, out, key_shift, row_base, table_rows);
   }
 }
 
 void SparseScatterAssignF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
-                            const float* src, hipStream_t s, int key_shift, uint64_t row_base) {
+                            const float* src, hipStream_t s, int key_shift, uint64_t row_base,
+                            uint64_t table_rows) {
   if (row_len % 4 == 0) {
     hipLaunchKernelGGL(scatter_assign_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)),
                        dim3(kBlock), 0, s, reinterpret_cast<float4*>(table), rows_dev, nrows,
-                       row_len / 4, reinterpret_cast<const float4*>(src), key_shift, row_base);
+                       row_len / 4, reinterpret_cast<const float4*>(src), key_shift, row_base,
+                       table_rows);
   } else {
     hipLaunchKernelGGL(scatter_assign_rows_scalar_f32, dim3(GridFor(nrows * row_len, 16384)),
                        dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src, key_shift,
-                       row_base);
+                       row_base, table_rows);
   }
 }
 
 void SparseScatterAddF32(float* table, const uint64_t* rows_dev, size_t nrows, size_t row_len,
                          const float* src, bool atomic, hipStream_t s, int key_shift,
-                         uint64_t row_base) {
+                         uint64_t row_base, uint64_t table_rows) {
   if (atomic || row_len % 4 != 0) {
     hipLaunchKernelGGL(scatter_add_rows_atomic_f32, dim3(GridFor(nrows * row_len, 16384)),
                        dim3(kBlock), 0, s, table, rows_dev, nrows, row_len, src, key_shift,
-                       row_base);
+                       row_base, table_rows);
   } else {
     hipLaunchKernelGGL(scatter_add_rows_f32, dim3(GridFor(nrows * (row_len / 4), 16384)),
                        dim3(kBlock), 0, s, reinterpret_cast<float4*>(table), rows_dev, nrows,
-                       row_len / 4, reinterpret_cast<const float4*>(src), key_shift, row_base);
+                       row_len / 4, reinterpret_cast<const float4*>(src), key_shift, row_base,
+                       table_rows);
   }
 }
 

@@ -13,7 +13,10 @@
 
 #include <algorithm>
 #include <atomic>
+#include <chrono>
+#include <condition_variable>
 #include <functional>
+#include <mutex>
 #include <set>
 #include <unordered_map>
 #include <vector>
@@ -394,8 +397,12 @@ class KVServer : public SimpleApp {
   }
 
   void set_request_handle(ReqHandle h) {
-    request_handle2_ = std::move(h);
-    handle_set_.store(true, std::memory_order_release);
+    {
+      std::lock_guard<std::mutex> lk(handle_mu_);
+      request_handle2_ = std::move(h);
+      handle_set_.store(true, std::memory_order_release);
+    }
+    handle_cv_.notify_all();
   }
 
   void Response(const KVMeta& req, const KVPairs<V>& res = KVPairs<V>()) {
@@ -452,17 +459,20 @@ class KVServer : public SimpleApp {
       kvs.keys = SArray<Key>({msg.meta.key});
     }
     // a worker's first request can arrive between construction and
-    // set_request_handle — wait for the app to install it
+    // set_request_handle — block (deterministically, on a cv) until the
+    // app installs it
     if (!handle_set_.load(std::memory_order_acquire)) {
-      for (int i = 0; i < 30000 && !handle_set_.load(std::memory_order_acquire); ++i) {
-        usleep(1000);
-      }
+      std::unique_lock<std::mutex> lk(handle_mu_);
+      bool ok = handle_cv_.wait_for(lk, std::chrono::seconds(60),
+                                    [this] { return handle_set_.load(); });
+      XPS_CHECK(ok) << "KVServer has no request handle after 60 s";
     }
-    XPS_CHECK(handle_set_.load()) << "KVServer has no request handle";
     request_handle2_(meta, kvs, this);
   }
 
   ReqHandle request_handle2_;
+  std::mutex handle_mu_;
+  std::condition_variable handle_cv_;
   std::atomic<bool> handle_set_{false};
 };
 
@@ -545,8 +555,14 @@ struct KVServerReduceHandle {
     }
   }
 
-  struct Entry {
-    std::vector<V> buf;
+  // Round state is per KEY-SET ("group"), exactly like the GPU reduce
+  // handler: one multi-key message per worker per server per round
+  // (BytePS bucketed DenseReduce); single-key messages are groups of 1.
+  // All workers must use the same key->message grouping.
+  struct Group {
+    std::vector<Key> keys;
+    std::vector<size_t> lens;  // elements per key
+    std::vector<V> buf;        // concatenated reduce buffer
     int pushes = 0;
     int pulls = 0;
     std::vector<KVMeta> waiting_pulls;
@@ -554,80 +570,103 @@ struct KVServerReduceHandle {
     std::vector<std::pair<KVMeta, KVPairs<V>>> waiting_pushes;
     std::set<int> pulled;
   };
-  std::unordered_map<Key, Entry> store;
+  std::unordered_map<uint64_t, Group> store;
   int num_workers_;
+
+  Group& GroupFor(const SArray<Key>& keys) {
+    uint64_t h = 1469598103934665603ull;
+    const unsigned char* p = reinterpret_cast<const unsigned char*>(keys.data());
+    for (size_t i = 0; i < keys.nbytes(); ++i) {
+      h ^= p[i];
+      h *= 1099511628211ull;
+    }
+    Group& g = store[h];
+    if (g.keys.empty()) {
+      g.keys.assign(keys.begin(), keys.end());
+    } else {
+      XPS_CHECK(g.keys.size() == keys.size() &&
+                std::equal(g.keys.begin(), g.keys.end(), keys.begin()))
+          << "reduce mode needs a consistent key->message grouping across workers";
+    }
+    return g;
+  }
 
  private:
   void HandlePush(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
-    XPS_CHECK_EQ(kvs.keys.size(), 1u) << "reduce mode is single-key-per-message";
     XPS_CHECK(!kvs.vals.on_device()) << "KVServerReduceHandle is CPU-only";
-    Entry& e = store[kvs.keys[0]];
-    if (e.pushes >= num_workers_) {
+    size_t n = kvs.keys.size();
+    Group& g = GroupFor(kvs.keys);
+    if (g.pushes >= num_workers_) {
       // a fast worker started the next round before this round's pulls
       // drained (the KVPairs copy keeps the payload alive)
-      e.waiting_pushes.emplace_back(req, kvs);
+      g.waiting_pushes.emplace_back(req, kvs);
       return;
     }
-    size_t len = kvs.lens.empty() ? kvs.vals.size() : static_cast<size_t>(kvs.lens[0]);
-    if (e.buf.size() < len) e.buf.resize(len, V(0));
-    const V* __restrict__ v = kvs.vals.data();
-    V* __restrict__ b = e.buf.data();
-    if (e.pushes == 0) {
-      HostPar::CopyBytes(b, v, len * sizeof(V));
-    } else {
-      SumInto(b, v, len);
+    if (g.lens.size() != n) {
+      g.lens.resize(n);
+      for (size_t i = 0; i < n; ++i) {
+        g.lens[i] = kvs.lens.empty() ? kvs.vals.size() / n : static_cast<size_t>(kvs.lens[i]);
+      }
     }
-    e.pushes++;
+    size_t total = 0;
+    for (size_t l : g.lens) total += l;
+    if (g.buf.size() < total) g.buf.resize(total, V(0));
+    const V* __restrict__ v = kvs.vals.data();
+    V* __restrict__ b = g.buf.data();
+    if (g.pushes == 0) {
+      HostPar::CopyBytes(b, v, total * sizeof(V));
+    } else {
+      SumInto(b, v, total);
+    }
+    g.pushes++;
     server->Response(req);
-    if (e.pushes >= num_workers_) {
+    if (g.pushes >= num_workers_) {
       std::vector<KVMeta> waiting;
-      waiting.swap(e.waiting_pulls);
-      for (auto& w : waiting) RespondPull(w, kvs.keys[0], server);
+      waiting.swap(g.waiting_pulls);
+      for (auto& w : waiting) RespondPull(w, g, server);
     }
   }
 
   void HandlePull(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
-    XPS_CHECK_EQ(kvs.keys.size(), 1u) << "reduce mode is single-key-per-message";
     // a pull may precede the round's pushes (it just waits)
-    Entry& e = store[kvs.keys[0]];
-    if (e.pulled.count(req.sender)) {
-      e.waiting_next_pulls.push_back(req);  // next-round pull, too early
+    Group& g = GroupFor(kvs.keys);
+    if (g.pulled.count(req.sender)) {
+      g.waiting_next_pulls.push_back(req);  // next-round pull, too early
       return;
     }
-    if (e.pushes < num_workers_) {
-      e.waiting_pulls.push_back(req);  // released by the round's last push
+    if (g.pushes < num_workers_) {
+      g.waiting_pulls.push_back(req);  // released by the round's last push
       return;
     }
-    RespondPull(req, kvs.keys[0], server);
+    RespondPull(req, g, server);
   }
 
-  void RespondPull(const KVMeta& req, Key key, KVServer<V>* server) {
-    Entry& e = store[key];
+  void RespondPull(const KVMeta& req, Group& g, KVServer<V>* server) {
     KVPairs<V> res;
-    res.keys = SArray<Key>({key});
+    res.keys = SArray<Key>(g.keys);
     // synchronous handler: the response is serialized before we return,
     // so a view of buf is safe (and zero-copy on the shm plane)
-    res.vals = SArray<V>(e.buf.data(), e.buf.size(), kCPU);
-    SArray<int> lens(1);
-    lens[0] = static_cast<int>(e.buf.size());
+    res.vals = SArray<V>(g.buf.data(), g.buf.size(), kCPU);
+    SArray<int> lens(g.keys.size());
+    for (size_t i = 0; i < g.keys.size(); ++i) lens[i] = static_cast<int>(g.lens[i]);
     res.lens = lens;
     server->Response(req, res);
-    e.pulled.insert(req.sender);
-    e.pulls++;
-    if (e.pulls >= num_workers_) {
-      e.pushes = 0;
-      e.pulls = 0;
-      e.pulled.clear();
+    g.pulled.insert(req.sender);
+    g.pulls++;
+    if (g.pulls >= num_workers_) {
+      g.pushes = 0;
+      g.pulls = 0;
+      g.pulled.clear();
       std::vector<std::pair<KVMeta, KVPairs<V>>> dpush;
-      dpush.swap(e.waiting_pushes);
+      dpush.swap(g.waiting_pushes);
       for (auto& d : dpush) HandlePush(d.first, d.second, server);
       std::vector<KVMeta> dpull;
-      dpull.swap(e.waiting_next_pulls);
+      dpull.swap(g.waiting_next_pulls);
       for (auto& d : dpull) {
-        if (e.pushes >= num_workers_) {
-          RespondPull(d, key, server);
+        if (g.pushes >= num_workers_) {
+          RespondPull(d, g, server);
         } else {
-          e.waiting_pulls.push_back(d);
+          g.waiting_pulls.push_back(d);
         }
       }
     }

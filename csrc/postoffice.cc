@@ -112,9 +112,13 @@ Customer* Postoffice::GetCustomer(int app_id, int customer_id, int timeout_sec) 
 
 void Postoffice::Barrier(int customer_id, int group) {
   XPS_CHECK(van_->IsReady());
+  int token;
   {
     std::lock_guard<std::mutex> lk(barrier_mu_);
     barrier_done_ = false;
+    token = ++barrier_seq_;
+    barrier_token_ = token;
+    barrier_group_ = group;
   }
   Message req;
   req.meta.control.cmd = Control::BARRIER;
@@ -123,6 +127,7 @@ void Postoffice::Barrier(int customer_id, int group) {
   req.meta.recver = kScheduler;
   req.meta.app_id = 0;
   req.meta.customer_id = customer_id;
+  req.meta.timestamp = token;  // echoed by the scheduler; matched in Manage
   // NOTE: the lock must NOT be held across Send — on the scheduler the
   // loopback delivery can complete the barrier inline on this thread and
   // re-enter Manage (which takes barrier_mu_)
@@ -134,6 +139,16 @@ void Postoffice::Barrier(int customer_id, int group) {
 void Postoffice::Manage(const Message& msg) {
   if (msg.meta.control.cmd == Control::BARRIER && !msg.meta.request) {
     std::lock_guard<std::mutex> lk(barrier_mu_);
+    // a stale or duplicated response (e.g. a resend whose original was
+    // slow) must not release a LATER Barrier() call: the scheduler
+    // echoes the group and this node's per-call token; both must match
+    // the outstanding call
+    if (msg.meta.control.barrier_group != barrier_group_ ||
+        msg.meta.timestamp != barrier_token_) {
+      XPS_VLOG(2) << "ignoring stale barrier response (group "
+                  << msg.meta.control.barrier_group << " token " << msg.meta.timestamp << ")";
+      return;
+    }
     barrier_done_ = true;
     barrier_cv_.notify_all();
   }

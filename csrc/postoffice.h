@@ -103,6 +103,9 @@ class Postoffice {
   std::mutex barrier_mu_;
   std::condition_variable barrier_cv_;
   bool barrier_done_ = false;
+  int barrier_seq_ = 0;    // per-call token generator
+  int barrier_token_ = 0;  // token of the outstanding Barrier() call
+  int barrier_group_ = 0;  // group of the outstanding Barrier() call
 
   std::mutex heartbeat_mu_;
   std::unordered_map<int, time_t> heartbeats_;

@@ -76,51 +76,103 @@ void GpuDenseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
   }
 }
 
+GpuDenseHandler::Group* GpuDenseHandler::GroupFor(const SArray<Key>& keys) {
+  // FNV-1a over the key bytes identifies the key-set
+  uint64_t h = 1469598103934665603ull;
+  const unsigned char* p = reinterpret_cast<const unsigned char*>(keys.data());
+  for (size_t i = 0; i < keys.nbytes(); ++i) {
+    h ^= p[i];
+    h *= 1099511628211ull;
+  }
+  Group* g;
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    g = &groups_[h];
+  }
+  if (g->keys.empty()) {
+    g->keys.assign(keys.begin(), keys.end());
+    g->ents.reserve(keys.size());
+    std::lock_guard<std::mutex> lk(mu_);
+    for (Key k : g->keys) g->ents.push_back(&store_[k]);
+  } else {
+    XPS_CHECK(g->keys.size() == keys.size() &&
+              std::equal(g->keys.begin(), g->keys.end(), keys.begin()))
+        << "reduce mode needs a consistent key->message grouping across workers "
+           "(key-set hash collision or inconsistent bucketing)";
+  }
+  return g;
+}
+
 void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& kvs,
                                        KVServer<float>* server) {
   XPS_STAGE(reduce_push);
-  XPS_CHECK_EQ(kvs.keys.size(), 1u) << "reduce mode is single-key-per-message";
-  size_t len = kvs.lens.empty() ? kvs.vals.nbytes()
-                                : static_cast<size_t>(kvs.lens[0]) * sizeof(float);
-  Entry* e;
-  {
-    std::lock_guard<std::mutex> lk(mu_);
-    e = &store_[kvs.keys[0]];
-  }
-  if (e->pushes >= num_workers_) {
+  size_t n = kvs.keys.size();
+  Group* g = GroupFor(kvs.keys);
+  if (g->pushes >= num_workers_) {
     // a fast worker started the next round before this round's pulls
     // drained: defer (the KVPairs copy keeps the remote buffer alive)
-    e->waiting_pushes.emplace_back(req, kvs);
+    g->waiting_pushes.emplace_back(req, kvs);
     return;
   }
-  if (e->buf.size() < len) {
-    e->buf = HbmPool::Get()->AllocArray(len);
+  XPS_CHECK(kvs.vals.on_device()) << "reduce mode needs device vals (pool buffers)";
+  if (g->lens.size() != n) {
+    g->lens.resize(n);
+    for (size_t i = 0; i < n; ++i) {
+      g->lens[i] = kvs.lens.empty() ? kvs.vals.nbytes() / n
+                                    : static_cast<size_t>(kvs.lens[i]) * sizeof(float);
+    }
   }
   hipStream_t stream = Stream(req.sender);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
-  XPS_CHECK(kvs.vals.on_device()) << "reduce mode needs device vals (pool buffers)";
-  OrderAfter(e, stream);  // chain behind the previous kernel on this key
-  if (e->pushes == 0) {
+  if (g->last_ev) XPS_HIP_CHECK(hipStreamWaitEvent(stream, g->last_ev.get(), 0));
+  bool first = g->pushes == 0;
+  if (first) {
     // the previous round's pull copies must finish before we overwrite
-    for (auto& ev : e->pull_guard) {
+    for (auto& ev : g->pull_guard) {
       XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev.get(), 0));
     }
-    e->pull_guard.clear();
-    kern::DenseAssign(e->buf.data(), kvs.vals.data(), len, stream);
-  } else {
-    kern::DenseSumF32(reinterpret_cast<float*>(e->buf.data()),
-                      reinterpret_cast<const float*>(kvs.vals.data()), len / sizeof(float),
-                      stream);
+    g->pull_guard.clear();
   }
-  e->pushes++;
+  // one batched kernel chain for the whole key-set (3 launches for 169
+  // rn50 buckets at kMaxBatch=64), one event per push
+  std::vector<kern::CopyDesc> descs;
+  descs.reserve(n);
+  size_t off = 0;
+  bool aligned = true;
+  for (size_t i = 0; i < n; ++i) {
+    size_t len = g->lens[i];
+    Entry* e = g->ents[i];
+    if (e->buf.size() < len) e->buf = HbmPool::Get()->AllocArray(len);
+    descs.push_back({e->buf.data(), reinterpret_cast<const char*>(kvs.vals.data()) + off, len});
+    aligned = aligned && (len % 16 == 0);
+    off += len;
+  }
+  if (aligned) {
+    if (first) {
+      kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
+    } else {
+      kern::BatchedSumF32(descs.data(), static_cast<int>(n), stream);
+    }
+  } else {
+    for (auto& d : descs) {
+      if (first) {
+        kern::DenseAssign(d.dst, d.src, d.nbytes, stream);
+      } else {
+        kern::DenseSumF32(reinterpret_cast<float*>(d.dst),
+                          reinterpret_cast<const float*>(d.src), d.nbytes / sizeof(float),
+                          stream);
+      }
+    }
+  }
+  g->pushes++;
   EventRef ev = MakeEventRef(po_, stream);
-  e->round_events.push_back(ev);
-  e->last_ev = ev;
+  g->round_events.push_back(ev);
+  g->last_ev = ev;
   server->Response(req);
-  if (e->pushes >= num_workers_) {
+  if (g->pushes >= num_workers_) {
     std::vector<KVMeta> waiting;
-    waiting.swap(e->waiting_pulls);
-    for (auto& w : waiting) RespondPull(w, e, server);
+    waiting.swap(g->waiting_pulls);
+    for (auto& w : waiting) RespondPull(w, g, server);
   }
 }
 
@@ -233,41 +285,98 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   server->Response(req);
 }
 
-void GpuDenseHandler::RespondPull(const KVMeta& req, Entry* e, KVServer<float>* server) {
+void GpuDenseHandler::RespondPull(const KVMeta& req, Group* g, KVServer<float>* server) {
   XPS_STAGE(reduce_respond_pull);
   hipStream_t stream = Stream(req.sender);
-  for (auto& ev : e->round_events) {
+  for (auto& ev : g->round_events) {
     XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev.get(), 0));
   }
-  KVPairs<float> res;
-  res.keys = SArray<Key>({req.key});
-  res.vals = SArray<float>::View(e->buf);
-  SArray<int> lens(1);
-  lens[0] = static_cast<int>(e->buf.size() / sizeof(float));
-  res.lens = lens;
-  if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
-  server->Response(req, res);  // the plane enqueues the copy on `stream` here
+  size_t n = g->keys.size();
+  size_t total = 0;
+  for (size_t len : g->lens) total += len;
+  SArray<Key> keys(g->keys);
+  SArray<int> lens(n);
+  for (size_t i = 0; i < n; ++i) lens[i] = static_cast<int>(g->lens[i] / sizeof(float));
+  bool responded = false;
+  // fast path: write the whole group straight into the requester's
+  // advertised pool destination (one batched kernel, meta-only response)
+  if (req.option & kOptPullAddr) {
+    if (auto* plane = ThePlane(po_)) {
+      std::vector<kern::CopyDesc> descs;
+      descs.reserve(n);
+      uint64_t off = 0;
+      bool ok = true;
+      for (size_t i = 0; i < n && ok; ++i) {
+        size_t len = g->lens[i];
+        char* dst = plane->PeerDst(req.sender, req.addr + off, len);
+        ok = dst != nullptr && len % 16 == 0;
+        if (ok) {
+          descs.push_back({dst, g->ents[i]->buf.data(), len});
+          off += len;
+        }
+      }
+      if (ok) {
+        kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
+        KVMeta r2 = req;
+        r2.option |= kOptInPlace;
+        r2.val_len = static_cast<int64_t>(total);
+        KVPairs<float> res2;
+        res2.keys = keys;
+        res2.lens = lens;
+        server->Response(r2, res2);  // plane defers the meta on `stream`
+        responded = true;
+      }
+    }
+  }
+  if (!responded) {
+    // staging path (TCP fallback / no advertised destination)
+    KVPairs<float> res;
+    res.keys = keys;
+    res.lens = lens;
+    if (n == 1) {
+      res.vals = SArray<float>::View(g->ents[0]->buf);
+    } else {
+      SArray<char> tmp = HbmPool::Get()->AllocArray(total);
+      std::vector<kern::CopyDesc> descs;
+      descs.reserve(n);
+      size_t off = 0;
+      bool aligned = true;
+      for (size_t i = 0; i < n; ++i) {
+        descs.push_back({tmp.data() + off, g->ents[i]->buf.data(), g->lens[i]});
+        aligned = aligned && (g->lens[i] % 16 == 0) && (off % 16 == 0);
+        off += g->lens[i];
+      }
+      if (aligned) {
+        kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
+      } else {
+        for (auto& d : descs) kern::DenseAssign(d.dst, d.src, d.nbytes, stream);
+      }
+      res.vals = SArray<float>::View(tmp);  // plane keeps it alive until sent
+    }
+    if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
+    server->Response(req, res);  // plane enqueues the in-place read on `stream`
+  }
   EventRef pe = MakeEventRef(po_, stream);
-  e->pull_guard.push_back(pe);
-  e->last_ev = pe;
-  e->pulled_senders.insert(req.sender);
-  e->pulls++;
-  if (e->pulls >= num_workers_) {
+  g->pull_guard.push_back(pe);
+  g->last_ev = pe;
+  g->pulled_senders.insert(req.sender);
+  g->pulls++;
+  if (g->pulls >= num_workers_) {
     // round over: reset, then replay deferred next-round pushes + pulls
-    e->pushes = 0;
-    e->pulls = 0;
-    e->pulled_senders.clear();
-    e->round_events.clear();  // refs drop back to the event pool
+    g->pushes = 0;
+    g->pulls = 0;
+    g->pulled_senders.clear();
+    g->round_events.clear();  // refs drop back to the event pool
     std::vector<std::pair<KVMeta, KVPairs<float>>> dpush;
-    dpush.swap(e->waiting_pushes);
+    dpush.swap(g->waiting_pushes);
     for (auto& d : dpush) HandleReducePush(d.first, d.second, server);
     std::vector<KVMeta> dpull;
-    dpull.swap(e->waiting_next_pulls);
+    dpull.swap(g->waiting_next_pulls);
     for (auto& d : dpull) {
-      if (e->pushes >= num_workers_) {
-        RespondPull(d, e, server);
+      if (g->pushes >= num_workers_) {
+        RespondPull(d, g, server);
       } else {
-        e->waiting_pulls.push_back(d);
+        g->waiting_pulls.push_back(d);
       }
     }
   }
@@ -280,21 +389,22 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   XPS_CHECK_GT(n, 0u);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
   if (mode_ == DenseMode::kReduce) {
-    XPS_CHECK_EQ(n, 1u) << "reduce mode is single-key-per-message";
-    std::lock_guard<std::mutex> lk(mu_);
     // a pull may legitimately precede the round's pushes (it just
-    // waits); create the entry on demand
-    Entry* e = &store_[kvs.keys[0]];
-    if (e->pulled_senders.count(req.sender)) {
+    // waits); GroupFor creates the round group on demand. NOTE: no lock
+    // is held across RespondPull — its round-reset path replays deferred
+    // pushes through HandleReducePush (which locks mu_ for store_
+    // access); Group state itself is serialized by the handler lane.
+    Group* g = GroupFor(kvs.keys);
+    if (g->pulled_senders.count(req.sender)) {
       // this sender already pulled the current round: a NEXT-round pull
-      e->waiting_next_pulls.push_back(req);
+      g->waiting_next_pulls.push_back(req);
       return;
     }
-    if (e->pushes < num_workers_) {
-      e->waiting_pulls.push_back(req);  // released by the round's last push
+    if (g->pushes < num_workers_) {
+      g->waiting_pulls.push_back(req);  // released by the round's last push
       return;
     }
-    RespondPull(req, e, server);
+    RespondPull(req, g, server);
     return;
   }
   hipStream_t stream = Stream(req.sender);
@@ -496,10 +606,10 @@ void GpuSparseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
     const uint64_t* rows = DeviceKeys(kvs.keys, req.sender, stream);
     if (accumulate_ || req.cmd == kCmdSum) {
       kern::SparseScatterAddF32(table, rows, n, row_len_, kvs.vals.data(), atomic_,
-                                stream, key_shift_, row_base_);
+                                stream, key_shift_, row_base_, rows_);
     } else {
       kern::SparseScatterAssignF32(table, rows, n, row_len_, kvs.vals.data(), stream, key_shift_,
-                                   row_base_);
+                                   row_base_, rows_);
     }
     auto* plane = po_->van() ? po_->van()->plane() : nullptr;
     if (!plane) XPS_HIP_CHECK(hipStreamSynchronize(stream));
@@ -508,7 +618,7 @@ void GpuSparseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
     const uint64_t* rows = DeviceKeys(kvs.keys, req.sender, stream);
     SArray<char> out = pool->AllocArray(n * row_len_ * sizeof(float));
     kern::SparseGatherF32(table, rows, n, row_len_, reinterpret_cast<float*>(out.data()), stream,
-                          key_shift_, row_base_);
+                          key_shift_, row_base_, rows_);
     KVPairs<float> res;
     // keys stay meta-only: device keys must not be dereferenced host-side
     res.vals = SArray<float>::View(out);

@@ -52,7 +52,26 @@ class GpuDenseHandler {
  private:
   struct Entry {
     SArray<char> buf;
-    // reduce-mode round state (only the customer thread touches it)
+    // last kernel that touched buf, on WHICHEVER peer stream: with >1
+    // workers, concurrent senders use different streams, so every
+    // kernel touching this entry waits on last_ev and publishes a new
+    // one — otherwise two senders' sum kernels race (lost updates) and
+    // a pull can read a half-written store
+    EventRef last_ev;
+  };
+
+  // Reduce-mode round state, per KEY-SET: a worker's buckets for this
+  // server ride ONE multi-key message per round (BytePS bucketed
+  // DenseReduce — collapses 2x169 per-key messages/step to 2 per
+  // server), so round accounting is per key-set with one batched
+  // sum/assign kernel chain and one event per push. Single-key messages
+  // are just groups of size 1. CONTRACT: all workers must use the same
+  // key->message grouping (the BytePS layout guarantees it); a key must
+  // not appear in two different groups.
+  struct Group {
+    std::vector<Key> keys;
+    std::vector<Entry*> ents;   // store_ entries (node-stable pointers)
+    std::vector<size_t> lens;   // byte length per key (set by first push)
     int pushes = 0;
     int pulls = 0;
     std::vector<KVMeta> waiting_pulls;
@@ -64,18 +83,14 @@ class GpuDenseHandler {
     std::set<int> pulled_senders;
     std::vector<EventRef> round_events;  // one per pusher stream
     std::vector<EventRef> pull_guard;    // pull copies the next round must wait on
-    // last kernel that touched buf, on WHICHEVER peer stream: with >1
-    // workers, concurrent senders use different streams, so every
-    // kernel touching this entry waits on last_ev and publishes a new
-    // one — otherwise two senders' sum kernels race (lost updates) and
-    // a pull can read a half-written store
-    EventRef last_ev;
+    EventRef last_ev;                    // last kernel touching the group
   };
 
   void HandlePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   void HandleReducePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   void HandlePull(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
-  void RespondPull(const KVMeta& req, Entry* e, KVServer<float>* server);
+  void RespondPull(const KVMeta& req, Group* g, KVServer<float>* server);
+  Group* GroupFor(const SArray<Key>& keys);  // find-or-create round group
   hipStream_t Stream(int sender);
   void OrderAfter(Entry* e, hipStream_t s);  // wait the entry's last_ev
 
@@ -85,6 +100,7 @@ class GpuDenseHandler {
   bool chain_ = false;  // >1 workers: cross-stream same-key ordering needed
   std::mutex mu_;
   std::unordered_map<Key, Entry> store_;
+  std::unordered_map<uint64_t, Group> groups_;  // keyset-hash -> round group
   hipStream_t fallback_stream_ = nullptr;
 };
 

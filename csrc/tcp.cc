@@ -111,6 +111,17 @@ int64_t TcpConn::RecvFrame(std::string* meta, std::vector<SArray<char>>* data) {
   if (ndata > 16 || mlen > (64u << 20)) return -1;
   std::vector<uint64_t> lens(ndata);
   if (ndata && !RecvAll(lens.data(), 8 * ndata)) return -1;
+  // sanity-cap each blob: a corrupt 8-byte length must drop the
+  // connection cleanly, not trigger a near-2^64 allocation
+  static const uint64_t kMaxBlobBytes =
+      static_cast<uint64_t>(Environment::Get()->GetInt("XPS_MAX_BLOB_GB", 32)) << 30;
+  for (uint32_t i = 0; i < ndata; ++i) {
+    if (lens[i] > kMaxBlobBytes) {
+      XPS_LOG(Warning) << "frame blob " << i << " claims " << lens[i]
+                       << " bytes (cap " << kMaxBlobBytes << "); dropping connection";
+      return -1;
+    }
+  }
   meta->resize(mlen);
   if (mlen && !RecvAll(&(*meta)[0], mlen)) return -1;
   int64_t total = 12 + 8 * ndata + mlen;

@@ -123,11 +123,17 @@ void Van::Stop() {
   if (stopping_.exchange(true)) return;
   if (plane_) plane_->Stop();
   listener_.Stop();
+  // swap the thread list out under conn_mu_: GetOrDial/OnNewConnection
+  // check stopping_ under the same lock before spawning, so no thread
+  // can be added after the swap (joining while a concurrent dialer
+  // emplaces into the same vector would be a data race)
+  std::vector<std::thread> threads;
   {
     std::lock_guard<std::mutex> lk(conn_mu_);
     for (auto& kv : conns_) kv.second->Close();
+    threads.swap(recv_threads_);
   }
-  for (auto& t : recv_threads_) {
+  for (auto& t : threads) {
     if (t.joinable()) t.join();
   }
   if (heartbeat_thread_.joinable()) heartbeat_thread_.join();
@@ -232,6 +238,10 @@ std::shared_ptr<TcpConn> Van::GetOrDial(int id) {
   auto conn = std::make_shared<TcpConn>(fd);
   {
     std::lock_guard<std::mutex> lk(conn_mu_);
+    if (stopping_.load()) {  // Stop already joined the recv threads
+      conn->Close();
+      return nullptr;
+    }
     auto it = conns_.find(id);
     if (it != conns_.end()) {
       conn->Close();
@@ -683,11 +693,11 @@ void Van::ProcessNodeListAssigned(Message& msg) {
 void Van::ProcessBarrierAtScheduler(Message& msg) {
   XPS_CHECK(po_->is_scheduler());
   int group = msg.meta.control.barrier_group;
-  std::vector<int> release;
+  std::vector<std::pair<int, int>> release;  // (node id, its call token)
   {
     std::lock_guard<std::mutex> lk(sched_mu_);
     auto& waiters = barrier_waiters_[group];
-    waiters.push_back(msg.meta.sender);
+    waiters.emplace_back(msg.meta.sender, msg.meta.timestamp);
     size_t expected = po_->GetNodeIDs(group).size();
     XPS_VLOG(2) << "barrier group " << group << ": " << waiters.size() << "/" << expected;
     if (waiters.size() < expected) return;
@@ -696,16 +706,17 @@ void Van::ProcessBarrierAtScheduler(Message& msg) {
   // release self LAST: waking our own Finalize first would let Van::Stop
   // close the connections before the other waiters get their responses
   std::stable_partition(release.begin(), release.end(),
-                        [this](int id) { return id != my_node_.id; });
+                        [this](const auto& w) { return w.first != my_node_.id; });
   Message res;
   res.meta.control.cmd = Control::BARRIER;
   res.meta.request = false;
   res.meta.control.barrier_group = group;
   res.meta.sender = my_node_.id;
-  for (int id : release) {
-    res.meta.recver = id;
+  for (auto& w : release) {
+    res.meta.recver = w.first;
+    res.meta.timestamp = w.second;  // echo the waiter's token (Manage matches it)
     Message copy = res;
-    SendToNode(copy, id);
+    SendToNode(copy, w.first);
   }
 }
 

@@ -70,12 +70,16 @@ class Van {
 
   Postoffice* postoffice() const { return po_; }
 
+  // Slow-path send used by the data plane when a deferred fast-path
+  // delivery fails (ring gone / full): the response must reach the
+  // waiting requester over TCP rather than be dropped (liveness).
+  int64_t SendOverTcp(Message& msg, int id);
+
   std::atomic<int64_t> send_bytes_{0};
   std::atomic<int64_t> recv_bytes_{0};
 
  private:
   int64_t SendToNode(Message& msg, int id);
-  int64_t SendOverTcp(Message& msg, int id);
   std::shared_ptr<TcpConn> GetOrDial(int id);
   void OnNewConnection(int fd);
   void RecvLoop(std::shared_ptr<TcpConn> conn);
@@ -121,7 +125,8 @@ class Van {
   // scheduler-only bootstrap/barrier state
   std::mutex sched_mu_;
   std::vector<std::pair<Node, std::shared_ptr<TcpConn>>> pending_nodes_;
-  std::unordered_map<int, std::vector<int>> barrier_waiters_;  // group -> requester ids
+  // group -> (requester id, its per-call token) pairs
+  std::unordered_map<int, std::vector<std::pair<int, int>>> barrier_waiters_;
   std::unordered_map<int, std::unordered_map<uint64_t, time_t>> dead_ignore_;  // unused yet
 
   std::shared_ptr<DataPlane> plane_;

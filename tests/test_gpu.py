@@ -198,6 +198,34 @@ def test_sparse_handler_gpu():
         _down_joint()
 
 
+def test_sparse_out_of_range_keys_gpu():
+    """Regression (round-1 advisor, medium): a misrouted/corrupt key must
+    not scribble outside the table shard. Scatters of out-of-range rows
+    are skipped; gathers of them read zeros; in-range rows still work."""
+    _boot_joint_inproc()
+    try:
+        rows, width = 1 << 10, 64
+        server = ps.KVServer(0)
+        server.set_gpu_sparse_handle(rows, width, accumulate=True)
+        worker = ps.KVWorker(0, 0)
+        # one valid row, one far out of range (would index GBs past the
+        # table without the bounds check)
+        idx = np.array([7, rows + 12345], dtype=np.uint64)
+        vals = np.stack([np.full(width, 3.0, dtype=np.float32),
+                         np.full(width, 666.0, dtype=np.float32)])
+        vbuf = ps.pool_alloc(vals.nbytes)
+        vbuf.copy_from(vals.reshape(-1))
+        dst = ps.pool_alloc(vals.nbytes)
+        lens = np.full(2, width, dtype=np.int32)
+        worker.wait(worker.zpush_ptr(idx, vbuf.ptr, vals.nbytes, 0, lens, cmd=2))
+        worker.wait(worker.zpull_ptr(idx, dst.ptr, vals.nbytes, 0, lens))
+        out = dst.to_numpy_f32().reshape(2, width)
+        assert np.allclose(out[0], 3.0)   # valid row round-trips
+        assert np.allclose(out[1], 0.0)   # out-of-range row reads zeros
+    finally:
+        _down_joint()
+
+
 def test_pool_tensor_fast_path():
     """torch tensors backed by pool memory ride the zero-copy plane."""
     import torch
@@ -298,6 +326,75 @@ def test_reduce_mode_two_joint_on_one_gpu():
     # round k: workers push (1+k) and (2+k) -> both pull 3+2k
     for rank, outs in results.items():
         assert outs == [3.0 + 2 * k for k in range(4)], results
+
+
+def _reduce_multikey_worker_fn(ps_mod, rank):
+    """Bucketed reduce rounds on GPU: one multi-key message per round
+    per server (group round accounting + batched kernels), mixed bucket
+    lengths — the rn50 config #4 protocol."""
+    server = ps_mod.KVServer(0)
+    server.set_gpu_dense_handle(mode="reduce")
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker = ps_mod.KVWorker(0, 0)
+    lens = np.array([1024, 4096, 64, 2048], dtype=np.int32)
+    keys = np.array([41, 42, 43, 44], dtype=np.uint64)
+    total = int(lens.sum())
+    src = ps_mod.pool_alloc(total * 4)
+    dst = ps_mod.pool_alloc(total * 4)
+    outs = []
+    for step in range(4):
+        src.copy_from(np.full(total, float(rank + 1 + step), dtype=np.float32))
+        ts1 = worker.zpush_ptr(keys, src.ptr, total * 4, 0, lens)
+        ts2 = worker.zpull_ptr(keys, dst.ptr, total * 4, 0, lens)  # overlapped
+        worker.wait(ts1)
+        worker.wait(ts2)
+        out = dst.to_numpy_f32()
+        assert np.allclose(out, out[0]), out[:4].tolist()
+        outs.append(float(out[0]))
+    return outs, server
+
+
+def test_reduce_multikey_two_joint_on_one_gpu():
+    results = launch_local(2, 2, _reduce_multikey_worker_fn, joint=True, devices={0: 0, 1: 0},
+                           env_extra={"XPS_POOL_GB": 4}, timeout=300)
+    for rank, outs in results.items():
+        assert outs == [3.0 + 2 * k for k in range(4)], results
+
+
+def _reduce_deferred_push_worker_fn(ps_mod, rank):
+    """Regression (round-1 advisor, high): a deferred next-round push
+    replayed from inside the pull path double-locked the GPU handler's
+    non-recursive mutex — the server hung. Push round 2 BEFORE pulling
+    round 1 so the replay path runs."""
+    server = ps_mod.KVServer(0)
+    server.set_gpu_dense_handle(mode="reduce")
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker = ps_mod.KVWorker(0, 0)
+    n = 4096
+    keys = np.array([88], dtype=np.uint64)
+    lens = np.array([n], dtype=np.int32)
+    src1 = ps_mod.pool_alloc(n * 4)
+    src2 = ps_mod.pool_alloc(n * 4)
+    dst = ps_mod.pool_alloc(n * 4)
+    src1.copy_from(np.full(n, 5.0, dtype=np.float32))
+    src2.copy_from(np.full(n, 9.0, dtype=np.float32))
+    ts1 = worker.zpush_ptr(keys, src1.ptr, n * 4, 0, lens)
+    ts2 = worker.zpush_ptr(keys, src2.ptr, n * 4, 0, lens)  # deferred on server
+    tp1 = worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens)   # triggers the replay
+    worker.wait(ts1)
+    worker.wait(tp1)
+    out1 = float(dst.to_numpy_f32()[0])
+    tp2 = worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens)
+    worker.wait(ts2)
+    worker.wait(tp2)
+    out2 = float(dst.to_numpy_f32()[0])
+    return [out1, out2], server
+
+
+def test_reduce_deferred_push_replay_gpu():
+    results = launch_local(1, 1, _reduce_deferred_push_worker_fn, joint=True, devices={0: 0},
+                           env_extra={"XPS_POOL_GB": 4}, timeout=240)
+    assert results[0] == [5.0, 9.0], results
 
 
 def test_multiprocess_hipipc_two_joint_on_one_gpu():

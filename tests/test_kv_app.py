@@ -303,6 +303,63 @@ def test_cpu_reduce_rounds_two_workers():
         assert outs == [3.0 + 2 * k for k in range(5)], results
 
 
+def _cpu_reduce_multikey_worker(ps_mod, rank):
+    """Bucketed reduce rounds: ONE multi-key message per round (the
+    BytePS DenseReduce batching — group round accounting, not per-key).
+    Mixed bucket lengths exercise the per-key offsets."""
+    server = ps_mod.KVServer(0)
+    server.set_reduce_handle(num_workers=2)
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    w = ps_mod.KVWorker(0, 0)
+    lens = np.array([64, 256, 128], dtype=np.int32)
+    keys = np.array([31, 32, 33], dtype=np.uint64)
+    total = int(lens.sum())
+    outs = []
+    for step in range(4):
+        vals = np.full(total, float(rank + 1 + step), dtype=np.float32)
+        ts1 = w.push(keys, vals, lens)
+        out = w.pull(keys)  # overlapped: pull issued while pushes in flight
+        w.wait(ts1)
+        assert out.shape[0] == total
+        assert np.allclose(out, out[0]), out[:4].tolist()
+        outs.append(float(out[0]))
+    return outs, server
+
+
+def test_cpu_reduce_rounds_multikey():
+    results = launch_local(2, 2, _cpu_reduce_multikey_worker, joint=True, timeout=300)
+    for rank, outs in results.items():
+        assert outs == [3.0 + 2 * k for k in range(4)], results
+
+
+def _cpu_reduce_deferred_push_worker(ps_mod, rank):
+    """Deferred next-round replay: the worker pushes round k+1 BEFORE
+    pulling round k, so when the round's last pull arrives the deferred
+    push replays inside the pull path (the code path that double-locked
+    the GPU handler's mutex before the round-2 fix)."""
+    server = ps_mod.KVServer(0)
+    server.set_reduce_handle(num_workers=1)
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    w = ps_mod.KVWorker(0, 0)
+    keys = np.array([77], dtype=np.uint64)
+    n = 512
+    lens = np.array([n], dtype=np.int32)
+    ts1 = w.push(keys, np.full(n, 5.0, dtype=np.float32), lens)
+    # round-2 push sent BEFORE the round-1 pull: per-sender FIFO delivers
+    # it first, so it lands in waiting_pushes until the pull drains
+    ts2 = w.push(keys, np.full(n, 9.0, dtype=np.float32), lens)
+    out1 = w.pull(keys)   # releases round 1, replays the deferred push
+    out2 = w.pull(keys)   # round 2 value
+    w.wait(ts1)
+    w.wait(ts2)
+    return [float(out1[0]), float(out2[0])], server
+
+
+def test_cpu_reduce_deferred_push_replay():
+    results = launch_local(1, 1, _cpu_reduce_deferred_push_worker, joint=True, timeout=240)
+    assert results[0] == [5.0, 9.0], results
+
+
 def _ordering_worker(ps_mod, rank):
     """Cross-transport FIFO: an 8 KB push rides the TCP fallback (host
     heap, > inline budget) while the pull request rides the shm ring —
