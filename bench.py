@@ -60,6 +60,9 @@ def parse_args():
                         "PUSH_PULL loop issues ZPush+ZPull per key together, "
                         "so overlap is the default)")
     p.add_argument("--hot-keys", type=int, default=8192, help="sparse keys per step")
+    p.add_argument("--no-fused", action="store_true",
+                   help="sparse: separate push and pull trips (default is the "
+                        "fused ZPushPull single-trip round)")
     p.add_argument("--emb-rows", type=int, default=1 << 20)
     p.add_argument("--emb-width", type=int, default=64)
     p.add_argument("--rtt-iters", type=int, default=100)
@@ -214,16 +217,25 @@ def main():
         lens = np.array([], dtype=np.int32)
         state = {"i": 0}
 
+        fused = not args.no_fused
+
         def one_step():
             i = state["i"] % nsets
             state["i"] += 1
             k, kb = key_sets[i], key_dev_bufs[i]
-            ts = worker.zpush_ptr(k, push_buf.ptr, vbytes, device, lens, cmd=2,
-                                  keys_dev_ptr=kb.ptr if kb else 0)
-            worker.wait(ts)
-            ts = worker.zpull_ptr(k, pull_buf.ptr, vbytes, device, lens, cmd=0,
-                                  keys_dev_ptr=kb.ptr if kb else 0)
-            worker.wait(ts)
+            if fused:
+                # one trip: scatter the update and gather the post-update
+                # rows in a single request per server
+                ts = worker.zpushpull_ptr(k, push_buf.ptr, pull_buf.ptr, vbytes, device,
+                                          lens, cmd=2, keys_dev_ptr=kb.ptr if kb else 0)
+                worker.wait(ts)
+            else:
+                ts = worker.zpush_ptr(k, push_buf.ptr, vbytes, device, lens, cmd=2,
+                                      keys_dev_ptr=kb.ptr if kb else 0)
+                worker.wait(ts)
+                ts = worker.zpull_ptr(k, pull_buf.ptr, vbytes, device, lens, cmd=0,
+                                      keys_dev_ptr=kb.ptr if kb else 0)
+                worker.wait(ts)
 
         bytes_per_worker_step = 2.0 * vbytes
     else:
@@ -378,6 +390,16 @@ def main():
                 "p50_rtt_us": {k: round(v, 1) for k, v in rtts.items()},
             },
         }
+        if not args.cpu:
+            # per-peer plane traffic (rank 0's view): each peer pair is
+            # its own xGMI link, so at N>1 this is per-link utilization
+            try:
+                pb = ps.plane_peer_bytes("worker")
+                out["config"]["per_peer_gb"] = {
+                    int(k): [round(v[0] / 1e9, 2), round(v[1] / 1e9, 2)]
+                    for k, v in pb.items()}
+            except Exception:
+                pass
         print(json.dumps(out, ensure_ascii=False))
 
     cluster.finish()

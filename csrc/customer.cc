@@ -1,5 +1,7 @@
 #include "customer.h"
 
+#include <chrono>
+
 #include "postoffice.h"
 
 namespace xps {
@@ -21,46 +23,57 @@ Customer::~Customer() {
 int Customer::NewRequest(int recver) {
   std::lock_guard<std::mutex> lk(mu_);
   int expected = recver >= 8 ? 1 : static_cast<int>(po_->GetNodeIDs(recver).size());
-  tracker_.emplace_back(expected, 0);
+  tracker_.emplace_back();
+  tracker_.back().expected = expected;
   return static_cast<int>(tracker_.size()) - 1;
 }
 
+Customer::Slot* Customer::GetSlot(int ts) {
+  std::lock_guard<std::mutex> lk(mu_);
+  XPS_CHECK_LT(static_cast<size_t>(ts), tracker_.size());
+  return &tracker_[ts];  // deque: stable address across growth
+}
+
 void Customer::WaitRequest(int ts) {
-  // brief spin first: responses on the data plane land in ~10 µs and a
-  // cv sleep/wake costs ~5-10 µs per side
-  for (int i = 0; i < 4000; ++i) {
-    std::lock_guard<std::mutex> lk(mu_);
-    if (tracker_[ts].first == tracker_[ts].second) return;
+  Slot* s = GetSlot(ts);
+  // Lock-free spin first: data-plane responses land in ~10 µs while a
+  // cv sleep/wake costs ~5-10 µs per side. The old spin re-took mu_
+  // every iteration and fought the delivery thread's tracker update for
+  // the same cache line — atomics keep the delivery path untouched.
+  auto deadline = std::chrono::steady_clock::now() + std::chrono::microseconds(150);
+  while (s->received.load(std::memory_order_acquire) < s->expected) {
+    if (std::chrono::steady_clock::now() >= deadline) {
+      std::unique_lock<std::mutex> lk(mu_);
+      cv_.wait(lk, [s] { return s->received.load(std::memory_order_acquire) >= s->expected; });
+      return;
+    }
   }
-  std::unique_lock<std::mutex> lk(mu_);
-  cv_.wait(lk, [this, ts] { return tracker_[ts].first == tracker_[ts].second; });
 }
 
 bool Customer::IsFinished(int ts) {
-  std::lock_guard<std::mutex> lk(mu_);
-  return tracker_[ts].first == tracker_[ts].second;
+  Slot* s = GetSlot(ts);
+  return s->received.load(std::memory_order_acquire) >= s->expected;
 }
 
 int Customer::NumResponse(int ts) {
-  std::lock_guard<std::mutex> lk(mu_);
-  return tracker_[ts].second;
+  return GetSlot(ts)->received.load(std::memory_order_acquire);
 }
 
 void Customer::AddResponse(int ts, int num) {
-  std::lock_guard<std::mutex> lk(mu_);
-  tracker_[ts].second += num;
-  if (tracker_[ts].second >= tracker_[ts].first) cv_.notify_all();
+  Slot* s = GetSlot(ts);
+  int got = s->received.fetch_add(num, std::memory_order_acq_rel) + num;
+  if (got >= s->expected) {
+    // empty critical section pairs with the waiter's predicate check:
+    // without it a waiter could test-and-sleep between our update and
+    // notify (lost wakeup)
+    std::lock_guard<std::mutex> lk(mu_);
+    cv_.notify_all();
+  }
 }
 
 void Customer::RunHandle(Message& msg) {
   handle_(msg);
-  if (!msg.meta.request) {
-    std::lock_guard<std::mutex> lk(mu_);
-    tracker_[msg.meta.timestamp].second++;
-    if (tracker_[msg.meta.timestamp].second >= tracker_[msg.meta.timestamp].first) {
-      cv_.notify_all();
-    }
-  }
+  if (!msg.meta.request) AddResponse(msg.meta.timestamp, 1);
 }
 
 void Customer::ProcessInline(Message& msg) {

@@ -5,10 +5,13 @@
 // draining a queue into the app's handle).
 #pragma once
 
+#include <atomic>
+#include <condition_variable>
+#include <deque>
 #include <functional>
+#include <mutex>
 #include <thread>
 #include <utility>
-#include <vector>
 
 #include "message.h"
 #include "queue.h"
@@ -47,8 +50,17 @@ class Customer {
   void ProcessInline(Message& msg);
 
  private:
+  // per-request slot: `received` is written lock-free by the delivery
+  // thread and spun on by WaitRequest (mu_ only guards deque growth and
+  // the cv sleep path)
+  struct Slot {
+    int expected = 0;
+    std::atomic<int> received{0};
+  };
+
   void Receiving();
   void RunHandle(Message& msg);
+  Slot* GetSlot(int ts);
 
   int app_id_;
   int customer_id_;
@@ -59,7 +71,7 @@ class Customer {
   std::mutex handle_mu_;
   std::mutex mu_;
   std::condition_variable cv_;
-  std::vector<std::pair<int, int>> tracker_;  // (expected, received) per ts
+  std::deque<Slot> tracker_;  // stable addresses; indexed by timestamp
 };
 
 }  // namespace xps

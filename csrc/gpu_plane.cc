@@ -88,6 +88,9 @@ IpcOpenLock g_ipc_lock;
 GpuPlane::GpuPlane(Postoffice* po, int device) : po_(po), device_(device) {
   my_host_hash_ = HostHash();
   inline_deliver_ = Environment::Get()->GetInt("XPS_INLINE_HANDLER", 1) != 0;
+  lanes_ = Environment::Get()->GetInt("XPS_STREAMS_PER_PEER", 2);
+  if (lanes_ < 1) lanes_ = 1;
+  if (lanes_ > 2) lanes_ = 2;
 }
 
 GpuPlane::~GpuPlane() { Stop(); }
@@ -159,7 +162,7 @@ void GpuPlane::Stop() {
 
 void GpuPlane::OnPeer(const Node& peer) {
   if (peer.host_hash != my_host_hash_) return;
-  std::lock_guard<std::mutex> lk(peers_mu_);
+  std::unique_lock<std::shared_timed_mutex> lk(peers_mu_);
   auto& p = peers_[peer.id];
   if (!p) p.reset(new Peer());
   p->node = peer;
@@ -169,7 +172,7 @@ void GpuPlane::ImportPeers() {
   // collect ids first (don't hold peers_mu_ across the import)
   std::vector<int> ids;
   {
-    std::lock_guard<std::mutex> lk(peers_mu_);
+    std::shared_lock<std::shared_timed_mutex> lk(peers_mu_);
     for (auto& kv : peers_) ids.push_back(kv.first);
   }
   for (int id : ids) {
@@ -183,7 +186,12 @@ void GpuPlane::ImportPeers() {
 }
 
 GpuPlane::Peer* GpuPlane::GetPeer(int id) {
-  std::lock_guard<std::mutex> lk(peers_mu_);
+  {
+    std::shared_lock<std::shared_timed_mutex> lk(peers_mu_);
+    auto it = peers_.find(id);
+    if (it != peers_.end()) return it->second.get();
+  }
+  std::unique_lock<std::shared_timed_mutex> lk(peers_mu_);
   auto& p = peers_[id];
   if (!p) {
     p.reset(new Peer());
@@ -193,12 +201,15 @@ GpuPlane::Peer* GpuPlane::GetPeer(int id) {
 }
 
 bool GpuPlane::EnsureRing(Peer* p) {
+  if (p->ring_ok.load(std::memory_order_acquire)) return true;
   std::lock_guard<std::mutex> lk(p->mu);
   if (p->ring.ok()) return true;
   if (p->ring_tried) return false;
   p->ring_tried = true;
   if (p->node.shm_uid == 0) return false;
-  return p->ring.Open(p->node.shm_uid);
+  bool ok = p->ring.Open(p->node.shm_uid);
+  if (ok) p->ring_ok.store(true, std::memory_order_release);
+  return ok;
 }
 
 bool GpuPlane::ImportPeerSlabs(Peer* p) {
@@ -314,14 +325,34 @@ char* GpuPlane::ResolvePeer(Peer* p, uint64_t global_off, uint64_t len) {
   return static_cast<char*>(p->slab_bases[idx]) + local;
 }
 
-hipStream_t GpuPlane::StreamForPeer(int node_id) {
+hipStream_t GpuPlane::StreamLane(int node_id, int lane) {
+  if (lane >= lanes_) lane = lanes_ - 1;
   Peer* p = GetPeer(node_id);
+  hipStream_t s = p->streams[lane].load(std::memory_order_acquire);
+  if (s) return s;
   std::lock_guard<std::mutex> lk(p->mu);
-  if (!p->stream) {
+  s = p->streams[lane].load(std::memory_order_relaxed);
+  if (!s) {
     XPS_HIP_CHECK(hipSetDevice(device_));
-    XPS_HIP_CHECK(hipStreamCreateWithFlags(&p->stream, hipStreamNonBlocking));
+    XPS_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+    p->streams[lane].store(s, std::memory_order_release);
   }
-  return p->stream;
+  return s;
+}
+
+hipStream_t GpuPlane::StreamForPeer(int node_id) { return StreamLane(node_id, 0); }
+
+hipStream_t GpuPlane::PullStreamForPeer(int node_id) { return StreamLane(node_id, 1); }
+
+std::vector<std::tuple<int, int64_t, int64_t>> GpuPlane::PeerBytes() {
+  std::vector<std::tuple<int, int64_t, int64_t>> out;
+  std::shared_lock<std::shared_timed_mutex> lk(peers_mu_);
+  for (auto& kv : peers_) {
+    int64_t tx = kv.second->tx_bytes.load(std::memory_order_relaxed);
+    int64_t rx = kv.second->rx_bytes.load(std::memory_order_relaxed);
+    if (tx || rx) out.emplace_back(kv.first, tx, rx);
+  }
+  return out;
 }
 
 hipEvent_t GpuPlane::GetEvent() {
@@ -422,10 +453,13 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     if (p->node.shm_uid == 0) p->node = peer_node;
   }
   bool response = !msg.meta.request;
+  auto sync_peer_lanes = [this, &peer_node] {
+    for (int l = 0; l < lanes_; ++l) (void)hipStreamSynchronize(StreamLane(peer_node.id, l));
+  };
   if (!EnsureRing(p)) {
     // TCP fallback for a response must not outrun handler kernels still
-    // running on this peer's stream (the worker may reuse buffers on ack)
-    if (response && device_ >= 0) (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
+    // running on this peer's streams (the worker may reuse buffers on ack)
+    if (response && device_ >= 0) sync_peer_lanes();
     return -1;
   }
 
@@ -456,6 +490,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
       }
       int64_t bytes = static_cast<int64_t>(vals.size() + payload.size());
       if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
+      p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
       return bytes;
     }
     return -1;  // cannot map the destination: TCP fallback
@@ -467,10 +502,13 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     SArray<char> vals = msg.data[1];
     char* dst = ResolvePeer(p, msg.meta.addr, vals.size());
     if (!dst) {
-      (void)hipStreamSynchronize(StreamForPeer(peer_node.id));
-      return -1;  // TCP fallback (stream drained first)
+      sync_peer_lanes();
+      return -1;  // TCP fallback (streams drained first)
     }
-    hipStream_t stream = StreamForPeer(peer_node.id);
+    // pull lane (only when the handler prepared its ordering there): the
+    // response copy overlaps the next push kernel on lane 0
+    hipStream_t stream = (msg.meta.option & kOptPullLane) ? PullStreamForPeer(peer_node.id)
+                                                          : StreamForPeer(peer_node.id);
     XPS_HIP_CHECK(hipSetDevice(device_));
     // copy KERNEL instead of hipMemcpyAsync: measured 5.2 vs 4.85 TB/s
     // same-device, and it reads/writes hipIpc-mapped peer memory alike
@@ -501,8 +539,9 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     // in-place write already landed in the peer's pool.
     Message keepalive;
     keepalive.data.push_back(vals);
-    DeferSendInternal(p, peer_node.id, std::move(meta_msg), std::move(keepalive),
+    DeferSendInternal(p, peer_node.id, stream, std::move(meta_msg), std::move(keepalive),
                       std::move(payload), bytes);
+    p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
     return bytes;
   }
 
@@ -526,24 +565,28 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   }
   int64_t bytes = static_cast<int64_t>(payload.size()) + ref_bytes;
   if (response && device_ >= 0) {
-    // order behind any handler kernels on this peer's stream. The resend
-    // copy doubles as the keepalive (shallow: SArrays shared).
+    // order behind any handler kernels on this peer's lane (the handler
+    // flags pull-lane work via kOptPullLane; everything else is lane 0).
+    // The resend copy doubles as the keepalive (shallow: SArrays shared).
+    hipStream_t stream = (msg.meta.option & kOptPullLane) ? PullStreamForPeer(peer_node.id)
+                                                          : StreamForPeer(peer_node.id);
     Message resend = msg;
-    DeferSendInternal(p, peer_node.id, std::move(resend), Message(), std::move(payload), bytes);
+    DeferSendInternal(p, peer_node.id, stream, std::move(resend), Message(), std::move(payload),
+                      bytes);
   } else {
     // requests — and every send of a host-only plane — go out now (host
     // responses were produced synchronously; nothing to wait for)
     if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
     XPS_VLOG(3) << "plane send done -> " << peer_node.id;
   }
+  p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
   return bytes;
 }
 
-void GpuPlane::DeferSendInternal(Peer* p, int peer_id, Message resend, Message keepalive,
-                                 std::string payload, int64_t bytes) {
+void GpuPlane::DeferSendInternal(Peer* p, int peer_id, hipStream_t stream, Message resend,
+                                 Message keepalive, std::string payload, int64_t bytes) {
   XPS_STAGE(defer_queue);
   XPS_CHECK_GE(device_, 0) << "deferred sends are a GPU-plane feature";
-  hipStream_t stream = StreamForPeer(peer_id);
   hipEvent_t ev = GetEvent();
   XPS_HIP_CHECK(hipSetDevice(device_));
   XPS_HIP_CHECK(hipEventRecord(ev, stream));
@@ -563,21 +606,43 @@ void GpuPlane::CompletionLoop() {
   prctl(PR_SET_TIMERSLACK, 1000);
   const int kSpin = Environment::Get()->GetInt("XPS_POLL_SPIN", 200000);
   int idle = 0;
+  // NOTE on locking: pend_mu_ is held only for deque push/pop/front
+  // snapshots, never across hipEventQuery or ring pushes — holding it
+  // through the poll loop put ~6 µs of lock contention into EVERY
+  // deferred response (measured via XPS_TIMING, round 2). This thread is
+  // the only popper, so a front observed ready stays front until we pop.
+  std::vector<int> ids;
   while (!stop_.load()) {
     bool did = false;
-    // failed ring pushes are replayed over TCP AFTER releasing pend_mu_
-    // (dialing can block); seq gating at the receiver restores order
-    std::vector<std::pair<int, Message>> tcp_fallback;
-    if (pending_count_.load() > 0) {
-      std::lock_guard<std::mutex> lk(pend_mu_);
-      for (auto& kv : pending_) {
-        auto& dq = kv.second;
-        while (!dq.empty()) {
-          Pending& front = dq.front();
-          hipError_t e = hipEventQuery(front.ev);
+    if (pending_count_.load(std::memory_order_acquire) > 0) {
+      ids.clear();
+      {
+        std::lock_guard<std::mutex> lk(pend_mu_);
+        for (auto& kv : pending_) {
+          if (!kv.second.empty()) ids.push_back(kv.first);
+        }
+      }
+      for (int id : ids) {
+        while (!stop_.load()) {
+          hipEvent_t front_ev = nullptr;
+          {
+            std::lock_guard<std::mutex> lk(pend_mu_);
+            auto& dq = pending_[id];
+            if (dq.empty()) break;
+            front_ev = dq.front().ev;
+          }
+          hipError_t e = hipEventQuery(front_ev);
           if (e == hipErrorNotReady) break;
           XPS_STAGE(defer_release);
           XPS_CHECK(e == hipSuccess) << "hipEventQuery: " << hipGetErrorString(e);
+          Pending front;
+          {
+            std::lock_guard<std::mutex> lk(pend_mu_);
+            auto& dq = pending_[id];
+            front = std::move(dq.front());
+            dq.pop_front();
+          }
+          pending_count_.fetch_sub(1);
           Peer* peer = GetPeer(front.peer_id);
           bool sent = peer && EnsureRing(peer) &&
                       peer->ring.Push(front.payload.data(),
@@ -589,18 +654,14 @@ void GpuPlane::CompletionLoop() {
             // response — deliver it over the TCP path instead
             XPS_LOG(Warning) << "deferred plane send failed -> " << front.peer_id
                              << "; falling back to TCP";
-            tcp_fallback.emplace_back(front.peer_id, std::move(front.resend));
+            if (po_->van()->SendOverTcp(front.resend, front.peer_id) < 0) {
+              XPS_LOG(Warning) << "TCP fallback to " << front.peer_id
+                               << " failed too (peer dead?)";
+            }
           }
           PutEvent(front.ev);
-          dq.pop_front();
-          pending_count_.fetch_sub(1);
           did = true;
         }
-      }
-    }
-    for (auto& f : tcp_fallback) {
-      if (po_->van()->SendOverTcp(f.second, f.first) < 0) {
-        XPS_LOG(Warning) << "TCP fallback to " << f.first << " failed too (peer dead?)";
       }
     }
     if (did) {
@@ -679,6 +740,9 @@ void GpuPlane::RingPollLoop() {
     if (!ok) continue;
     XPS_VLOG(3) << "ring recv: " << msg.DebugString();
     po_->van()->recv_bytes_ += n + ref_bytes;
+    if (msg.meta.sender != kEmptyNodeID) {
+      GetPeer(msg.meta.sender)->rx_bytes.fetch_add(n + ref_bytes, std::memory_order_relaxed);
+    }
     if (inline_deliver_) {
       po_->van()->DeliverInline(msg);
     } else {

@@ -19,8 +19,11 @@
 #include <atomic>
 #include <deque>
 #include <memory>
+#include <shared_mutex>
 #include <thread>
+#include <tuple>
 #include <unordered_map>
+#include <vector>
 
 #include "shm_ring.h"
 #include "van.h"
@@ -46,9 +49,20 @@ class GpuPlane : public DataPlane {
   void Stop() override;
 
   // ---- used by the GPU server handlers -------------------------------
-  // the per-peer HIP stream every op destined for / received from `peer`
-  // runs on (gives the per-(key,peer) ordering guarantee)
-  hipStream_t StreamForPeer(int node_id);
+  // Per-peer HIP stream LANES (XPS_STREAMS_PER_PEER, default 2):
+  // lane 0 carries push/accumulate kernels, the pull lane carries pull-
+  // response copies, so a key's response copy overlaps the next key's
+  // push kernel (same xGMI link, different queues). Same-key cross-lane
+  // ordering is the handlers' job via their last_ev event chains — the
+  // dense handler enables chaining whenever lanes > 1. With 1 lane both
+  // calls return the same stream (the round-1 behavior).
+  hipStream_t StreamForPeer(int node_id);      // lane 0: push/compute
+  hipStream_t PullStreamForPeer(int node_id);  // pull-response lane
+  int lanes() const { return lanes_; }
+  // per-peer traffic counters: (node id, tx bytes, rx bytes) — the
+  // per-link utilization report of SURVEY §5.8 (each peer pair rides
+  // its own xGMI link)
+  std::vector<std::tuple<int, int64_t, int64_t>> PeerBytes();
   int device() const { return device_; }
   // pooled events (shared with the server handlers)
   hipEvent_t GetEvent();
@@ -64,9 +78,15 @@ class GpuPlane : public DataPlane {
     Node node;
     ShmRing ring;  // producer handle on the peer's inbound ring
     bool ring_tried = false;
+    // fast-path flag: the ring is opened once and never closed while the
+    // plane runs, so senders read this without taking mu
+    std::atomic<bool> ring_ok{false};
     std::vector<void*> slab_bases;  // peer pool slabs mapped into our space
     bool pool_tried = false;
-    hipStream_t stream = nullptr;
+    // lane 0 = push, 1 = pull; created once, then read lock-free
+    std::atomic<hipStream_t> streams[2] = {{nullptr}, {nullptr}};
+    std::atomic<int64_t> tx_bytes{0};
+    std::atomic<int64_t> rx_bytes{0};
     std::mutex mu;
   };
 
@@ -95,17 +115,20 @@ class GpuPlane : public DataPlane {
   bool Serialize(const Message& msg, const std::vector<char>& by_ref, std::string* out);
   void RingPollLoop();
   void CompletionLoop();
-  void DeferSendInternal(Peer* p, int peer_id, Message resend, Message keepalive,
-                         std::string payload, int64_t bytes);
+  hipStream_t StreamLane(int node_id, int lane);
+  void DeferSendInternal(Peer* p, int peer_id, hipStream_t stream, Message resend,
+                         Message keepalive, std::string payload, int64_t bytes);
 
   Postoffice* po_;
   int device_;
+  int lanes_ = 2;  // XPS_STREAMS_PER_PEER
   uint64_t my_host_hash_;
   ShmRing in_ring_;
   bool started_ = false;
   bool inline_deliver_ = true;  // XPS_INLINE_HANDLER (default on)
 
-  std::mutex peers_mu_;
+  // read-mostly after bootstrap: shared lock on the hot send/recv paths
+  std::shared_timed_mutex peers_mu_;
   std::unordered_map<int, std::unique_ptr<Peer>> peers_;
 
   std::thread poll_thread_;

@@ -102,6 +102,31 @@ class KVWorker : public SimpleApp {
     return ts;
   }
 
+  // Fused push+pull: ONE request per server that carries the pushed
+  // vals AND advertises the pull destination; the server applies the
+  // push and answers with the post-push values in the same trip. Halves
+  // the round trips of the push-then-pull pattern (sparse embedding
+  // lookup+update, BASELINE config #5). push vals and the pull dst
+  // share the same keys/lens geometry.
+  int ZPushPull(const SArray<Key>& keys, const SArray<V>& vals, SArray<V>* outs,
+                const SArray<int>& lens = {}, int cmd = 0, const Callback& cb = nullptr,
+                const SArray<Key>& keys_dev = {}) {
+    XPS_CHECK(outs && outs->size()) << "ZPushPull needs a pre-sized outs buffer";
+    int ts = obj_->NewRequest(kServerGroup);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      pull_dst_[ts] = {outs, nullptr};
+    }
+    AddCallback(ts, cb);
+    KVPairs<V> kvs;
+    kvs.keys = keys;
+    kvs.vals = vals;
+    kvs.lens = lens;
+    kvs.keys_dev = keys_dev;
+    Send(ts, true, true, cmd, kvs, outs);
+    return ts;
+  }
+
   // zero-copy pull into a pre-allocated vals buffer (device or host)
   int ZPull(const SArray<Key>& keys, SArray<V>* vals, SArray<int>* lens = nullptr, int cmd = 0,
             const Callback& cb = nullptr, const SArray<Key>& keys_dev = {}) {
@@ -187,7 +212,8 @@ class KVWorker : public SimpleApp {
     callbacks_[ts] = cb;
   }
 
-  void Send(int ts, bool push, bool pull, int cmd, KVPairs<V>& kvs) {
+  // dst: fused push+pull destination buffer (same geometry as kvs.vals)
+  void Send(int ts, bool push, bool pull, int cmd, KVPairs<V>& kvs, SArray<V>* dst = nullptr) {
     SlicedKVs sliced;
     slicer_(kvs, po_->GetServerKeyRanges(), &sliced);
     int skipped = 0;
@@ -230,6 +256,23 @@ class KVWorker : public SimpleApp {
         msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
         msg.AddData(s.vals);
         if (!s.lens.empty()) msg.AddData(s.lens);
+        if (pull && dst) {
+          // fused round: advertise this slice's pull destination — the
+          // dst buffer shares the vals geometry, so the slice's element
+          // offset within kvs.vals locates its dst segment
+          size_t elem_off = static_cast<size_t>(s.vals.data() - kvs.vals.data());
+          const V* dptr = dst->data() + elem_off;
+          uint64_t off = 0;
+          if (dst->on_device()) {
+            if (HbmPool::Get()->OffsetOf(dptr, &off)) {
+              msg.meta.addr = off;
+              msg.meta.option |= kOptPullAddr;
+            }
+          } else if (HostShmPool::Get()->OffsetOf(dptr, &off)) {
+            msg.meta.addr = off;
+            msg.meta.option |= kOptPullAddr | kOptHostAddr;
+          }
+        }
       } else {
         // pull request: keys (+lens geometry) only; advertise destination
         msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
@@ -504,7 +547,10 @@ struct KVServerDefaultHandle {
         }
         off += len;
       }
-    } else if (req.pull) {
+    }
+    // plain pull, or the pull half of a fused ZPushPull (push applied
+    // above, so the response carries the post-push values)
+    if (req.pull) {
       res.keys = kvs.keys;
       if (n == 1) {
         // zero-copy view of the store entry (stable: map nodes don't move)

@@ -78,6 +78,10 @@ static const int kOptInPlace = 1;   // pull response already written into dst bu
 static const int kOptValsByRef = 2; // vals travel as {pool offset, len}, not bytes
 static const int kOptPullAddr = 4;  // meta.addr holds a valid pull-destination offset
 static const int kOptHostAddr = 8;  // meta.addr is a HOST-shm-pool offset (else HBM pool)
+// the response's handler work ran on the peer's PULL stream lane: the
+// plane must enqueue the in-place write / completion event on that lane
+// (handlers that stay on lane 0, e.g. sparse, leave this unset)
+static const int kOptPullLane = 16;
 
 enum DataType : int { kChar = 0, kInt32, kInt64, kUint64, kFloat32, kFloat64, kUint8 };
 inline size_t DataTypeSize(int t) {

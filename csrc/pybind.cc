@@ -99,6 +99,26 @@ class PyKVWorker {
     return w_.ZPush(k, v, l, cmd, nullptr, kd);
   }
 
+  // fused push+pull (ZPushPull): push vals_ptr, receive the post-push
+  // values into out_ptr — one round trip (sparse lookup+update)
+  int ZPushPullPtr(py::array_t<uint64_t> keys, uintptr_t vals_ptr, uintptr_t out_ptr,
+                   size_t vals_bytes, int device, py::array_t<int> lens, int cmd,
+                   uintptr_t keys_dev_ptr) {
+    SArray<Key> k;
+    k.CopyFrom(keys.data(), keys.size());
+    SArray<float> v(reinterpret_cast<float*>(vals_ptr), vals_bytes / sizeof(float), device);
+    auto* out = new SArray<float>(reinterpret_cast<float*>(out_ptr), vals_bytes / sizeof(float),
+                                  device);
+    SArray<int> l;
+    if (lens.size()) l.CopyFrom(lens.data(), lens.size());
+    SArray<Key> kd;
+    if (keys_dev_ptr) {
+      kd = SArray<Key>(reinterpret_cast<Key*>(keys_dev_ptr), keys.size(), device);
+    }
+    py::gil_scoped_release rel;
+    return w_.ZPushPull(k, v, out, l, cmd, [out]() { delete out; }, kd);
+  }
+
   int ZPullPtr(py::array_t<uint64_t> keys, uintptr_t vals_ptr, size_t vals_bytes, int device,
                py::array_t<int> lens, int cmd, uintptr_t keys_dev_ptr) {
     SArray<Key> k;
@@ -494,6 +514,10 @@ PYBIND11_MODULE(_core, m) {
       .def("zpush_ptr", &PyKVWorker::ZPushPtr, py::arg("keys"), py::arg("vals_ptr"),
            py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>(),
            py::arg("cmd") = 0, py::arg("keys_dev_ptr") = 0)
+      .def("zpushpull_ptr", &PyKVWorker::ZPushPullPtr, py::arg("keys"), py::arg("vals_ptr"),
+           py::arg("out_ptr"), py::arg("nbytes"), py::arg("device") = -1,
+           py::arg("lens") = py::array_t<int>(), py::arg("cmd") = 0,
+           py::arg("keys_dev_ptr") = 0)
       .def("zpull_ptr", &PyKVWorker::ZPullPtr, py::arg("keys"), py::arg("vals_ptr"),
            py::arg("vals_bytes"), py::arg("device"), py::arg("lens") = py::array_t<int>(),
            py::arg("cmd") = 0, py::arg("keys_dev_ptr") = 0)
@@ -520,6 +544,21 @@ PYBIND11_MODULE(_core, m) {
   // transport-level zero-copy assertion (reference parity:
   // test_benchmark.cc:169-181 registered-buffer pointer equality)
   m.def("zero_copy_recv_count", []() { return g_zero_copy_recv.load(); });
+
+  // per-peer plane traffic: {node_id: (tx_bytes, rx_bytes)} — each peer
+  // pair rides its own xGMI link, so this is the per-link utilization
+  // report of SURVEY §5.8 (bench.py emits it for the scaling runs)
+  m.def("plane_peer_bytes", [](const std::string& role, int idx) {
+    py::dict out;
+    auto* po = GetPO(role, idx);
+    auto* plane = po->van() ? dynamic_cast<GpuPlane*>(po->van()->plane()) : nullptr;
+    if (plane) {
+      for (auto& t : plane->PeerBytes()) {
+        out[py::int_(std::get<0>(t))] = py::make_tuple(std::get<1>(t), std::get<2>(t));
+      }
+    }
+    return out;
+  }, py::arg("role") = "worker", py::arg("idx") = 0);
 
   // rank-ordering policy hook (tested in test_utils.py): takes
   // (role, host, port) tuples, returns them in rank-assignment order

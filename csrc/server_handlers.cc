@@ -25,6 +25,17 @@ static hipStream_t PeerStream(Postoffice* po, int sender, hipStream_t* fallback)
   return *fallback;
 }
 
+// pull-response lane: copies here overlap the next push kernel on the
+// peer's lane 0; same-key ordering rides the handler's event chain
+static hipStream_t PeerPullStream(Postoffice* po, int sender, hipStream_t* fallback) {
+  auto* plane = ThePlane(po);
+  if (plane) return plane->PullStreamForPeer(sender);
+  if (!*fallback) {
+    XPS_HIP_CHECK(hipStreamCreateWithFlags(fallback, hipStreamNonBlocking));
+  }
+  return *fallback;
+}
+
 static hipEvent_t EventAlloc(Postoffice* po) {
   if (auto* plane = ThePlane(po)) return plane->GetEvent();
   hipEvent_t ev;
@@ -51,10 +62,11 @@ static EventRef MakeEventRef(Postoffice* po, hipStream_t s) {
 GpuDenseHandler::GpuDenseHandler(Postoffice* po, DenseMode mode) : po_(po), mode_(mode) {
   XPS_CHECK(HbmPool::Get()->initialized()) << "GpuDenseHandler needs the HBM pool";
   num_workers_ = std::max(1, po_->num_workers());
-  // one worker process = one peer stream = stream-ordered kernels; with
-  // more, same-key kernels land on different peer streams and must be
-  // chained through the entry's last_ev
-  chain_ = num_workers_ > 1;
+  // same-key kernels can land on different streams in two cases: >1
+  // worker (one stream set per peer) and >1 lanes per peer (push on
+  // lane 0, pull copies on the pull lane) — both need the last_ev chain
+  auto* plane = ThePlane(po_);
+  chain_ = num_workers_ > 1 || (plane && plane->lanes() > 1);
 }
 
 GpuDenseHandler::~GpuDenseHandler() = default;
@@ -64,6 +76,10 @@ void GpuDenseHandler::OrderAfter(Entry* e, hipStream_t s) {
 }
 
 hipStream_t GpuDenseHandler::Stream(int sender) { return PeerStream(po_, sender, &fallback_stream_); }
+
+hipStream_t GpuDenseHandler::PullStream(int sender) {
+  return PeerPullStream(po_, sender, &fallback_stream_);
+}
 
 void GpuDenseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
                                  KVServer<float>* server) {
@@ -287,7 +303,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
 
 void GpuDenseHandler::RespondPull(const KVMeta& req, Group* g, KVServer<float>* server) {
   XPS_STAGE(reduce_respond_pull);
-  hipStream_t stream = Stream(req.sender);
+  hipStream_t stream = PullStream(req.sender);
   for (auto& ev : g->round_events) {
     XPS_HIP_CHECK(hipStreamWaitEvent(stream, ev.get(), 0));
   }
@@ -318,7 +334,7 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Group* g, KVServer<float>* 
       if (ok) {
         kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
         KVMeta r2 = req;
-        r2.option |= kOptInPlace;
+        r2.option |= kOptInPlace | kOptPullLane;
         r2.val_len = static_cast<int64_t>(total);
         KVPairs<float> res2;
         res2.keys = keys;
@@ -354,7 +370,9 @@ void GpuDenseHandler::RespondPull(const KVMeta& req, Group* g, KVServer<float>* 
       res.vals = SArray<float>::View(tmp);  // plane keeps it alive until sent
     }
     if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
-    server->Response(req, res);  // plane enqueues the in-place read on `stream`
+    KVMeta r = req;
+    r.option |= kOptPullLane;  // ordering prepared on the pull lane
+    server->Response(r, res);  // plane enqueues the in-place read on `stream`
   }
   EventRef pe = MakeEventRef(po_, stream);
   g->pull_guard.push_back(pe);
@@ -407,7 +425,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
     RespondPull(req, g, server);
     return;
   }
-  hipStream_t stream = Stream(req.sender);
+  hipStream_t stream = PullStream(req.sender);
   // multi-key pull with an in-place destination: batched copy of every
   // store entry straight into the requester's mapped pool (no staging
   // buffer, meta-only response)
@@ -438,7 +456,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
         for (Entry* e : ents) OrderAfter(e, stream);
         kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
         KVMeta r2 = req;
-        r2.option |= kOptInPlace;
+        r2.option |= kOptInPlace | kOptPullLane;
         r2.val_len = static_cast<int64_t>(off);
         KVPairs<float> res2;
         res2.keys = kvs.keys;
@@ -506,7 +524,9 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   // TCP fallback (no in-place destination): the staging D2H copy below in
   // the van is stream-unaware — drain our stream first
   if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
-  server->Response(req, res);  // plane enqueues the in-place read on `stream`
+  KVMeta r = req;
+  r.option |= kOptPullLane;  // ordering prepared on the pull lane
+  server->Response(r, res);  // plane enqueues the in-place read on `stream`
   if (chain_) {
     EventRef ev = MakeEventRef(po_, stream);  // pushes must wait these reads
     for (Entry* e : touched) e->last_ev = ev;
@@ -610,6 +630,21 @@ void GpuSparseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
     } else {
       kern::SparseScatterAssignF32(table, rows, n, row_len_, kvs.vals.data(), stream, key_shift_,
                                    row_base_, rows_);
+    }
+    if (req.pull) {
+      // fused round (ZPushPull): gather the post-scatter rows on the
+      // SAME stream (ordered behind the scatter) and answer in one trip
+      SArray<char> out = pool->AllocArray(n * row_len_ * sizeof(float));
+      kern::SparseGatherF32(table, rows, n, row_len_, reinterpret_cast<float*>(out.data()),
+                            stream, key_shift_, row_base_, rows_);
+      KVPairs<float> res;
+      res.vals = SArray<float>::View(out);
+      SArray<int> lens(1);
+      lens[0] = static_cast<int>(n * row_len_);
+      res.lens = lens;
+      if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
+      server->Response(req, res);
+      return;
     }
     auto* plane = po_->van() ? po_->van()->plane() : nullptr;
     if (!plane) XPS_HIP_CHECK(hipStreamSynchronize(stream));

@@ -91,7 +91,8 @@ class GpuDenseHandler {
   void HandlePull(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   void RespondPull(const KVMeta& req, Group* g, KVServer<float>* server);
   Group* GroupFor(const SArray<Key>& keys);  // find-or-create round group
-  hipStream_t Stream(int sender);
+  hipStream_t Stream(int sender);      // lane 0: push/accumulate kernels
+  hipStream_t PullStream(int sender);  // pull lane: response copies
   void OrderAfter(Entry* e, hipStream_t s);  // wait the entry's last_ev
 
   Postoffice* po_;

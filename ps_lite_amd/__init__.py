@@ -45,6 +45,7 @@ from ._core import (  # noqa: F401
     node_id,
     num_servers,
     num_workers,
+    plane_peer_bytes,
     pool_alloc,
     pool_in_use,
     pool_init,

@@ -198,6 +198,31 @@ def test_sparse_handler_gpu():
         _down_joint()
 
 
+def test_sparse_fused_round_gpu():
+    """Fused ZPushPull on the sparse handler: scatter-add + gather in ONE
+    trip; the response must carry the post-update rows."""
+    _boot_joint_inproc()
+    try:
+        rows, width, nsel = 1 << 12, 64, 512
+        server = ps.KVServer(0)
+        server.set_gpu_sparse_handle(rows, width, accumulate=True)
+        worker = ps.KVWorker(0, 0)
+        rng = np.random.default_rng(7)
+        idx = np.sort(rng.choice(rows, size=nsel, replace=False)).astype(np.uint64)
+        grads = rng.standard_normal((nsel, width)).astype(np.float32)
+        vbuf = ps.pool_alloc(grads.nbytes)
+        vbuf.copy_from(grads.reshape(-1))
+        dst = ps.pool_alloc(grads.nbytes)
+        lens = np.full(nsel, width, dtype=np.int32)
+        for it in range(3):  # accumulate: round k returns (k+1)*grads
+            worker.wait(worker.zpushpull_ptr(idx, vbuf.ptr, dst.ptr, grads.nbytes, 0,
+                                             lens, cmd=2))
+            out = dst.to_numpy_f32().reshape(nsel, width)
+            assert np.allclose(out, (it + 1) * grads, atol=1e-5), it
+    finally:
+        _down_joint()
+
+
 def test_sparse_out_of_range_keys_gpu():
     """Regression (round-1 advisor, medium): a misrouted/corrupt key must
     not scribble outside the table shard. Scatters of out-of-range rows

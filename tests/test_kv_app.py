@@ -360,6 +360,34 @@ def test_cpu_reduce_deferred_push_replay():
     assert results[0] == [5.0, 9.0], results
 
 
+def _fused_pushpull_worker(ps_mod, rank):
+    """ZPushPull: one request carries the push AND returns the post-push
+    values (halves the sparse round's trips). CPU default handle: the
+    response must equal the accumulated store."""
+    server = ps_mod.KVServer(0)
+    server.set_default_handle()
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    w = ps_mod.KVWorker(0, 0)
+    n = 1024
+    keys = np.array([61, 62], dtype=np.uint64)
+    lens = np.array([n, n], dtype=np.int32)
+    push = ps_mod.host_alloc(2 * n * 4)
+    out = ps_mod.host_alloc(2 * n * 4)
+    vals = np.concatenate([np.full(n, 2.0, dtype=np.float32),
+                           np.full(n, 5.0, dtype=np.float32)])
+    push.copy_from(vals)
+    for it in range(3):  # sum handle: round k returns (k+1)*vals
+        w.wait(w.zpushpull_ptr(keys, push.ptr, out.ptr, 2 * n * 4, -1, lens, cmd=2))
+        got = out.to_numpy_f32()
+        assert np.allclose(got, (it + 1) * vals), (it, got[:3].tolist())
+    return True
+
+
+def test_fused_pushpull_cpu():
+    results = launch_local(1, 1, _fused_pushpull_worker, joint=True, timeout=240)
+    assert results[0] is True
+
+
 def _ordering_worker(ps_mod, rank):
     """Cross-transport FIFO: an 8 KB push rides the TCP fallback (host
     heap, > inline budget) while the pull request rides the shm ring —
