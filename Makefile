@@ -57,4 +57,13 @@ cppbench: $(CORE_OBJS) build/bench_kv.o
 	$(HIPCC) $(CPPBENCH_LD) build/bench_kv.o $(CORE_OBJS) \
 	    -o build/bench_kv -lpthread
 
-.PHONY: all clean cppbench
+# self-contained lint gate (reference parity: tests/lint.py + make lint)
+lint:
+	python3 scripts/lint.py
+
+# the local CI gate: build + lint + the CPU test suite (the GPU suite
+# needs an MI355X: `python -m pytest tests -m gpu` there)
+check: all lint
+	python3 -m pytest tests -q -m "not gpu"
+
+.PHONY: all clean cppbench lint check

@@ -363,7 +363,8 @@ def main():
 
     if rank == 0:
         out = {
-            "metric": "push+pull GB/s per worker + p50 round-trip µs, 1 MB & 64 MB, at 1/2/4/8 GPUs",
+            "metric": ("push+pull GB/s per worker + p50 round-trip µs, "
+                       "1 MB & 64 MB, at 1/2/4/8 GPUs"),
             "value": round(total_gbs, 3),
             "unit": "GB/s",
             "n_gpus": n,
@@ -379,7 +380,8 @@ def main():
                 "model": model_name,
                 "global_batch": len(msg_sizes) or args.hot_keys,
                 "seq_len": max(msg_sizes) if msg_sizes else args.emb_width,
-                "parallelism": ("cpu-tcp" if args.cpu else f"byteps-joint x{n} (worker+server per GPU)"),
+                "parallelism": ("cpu-tcp" if args.cpu
+                                else f"byteps-joint x{n} (worker+server per GPU)"),
                 "mode": args.mode + ("-cpu" if args.cpu else ""),
                 "msg_bytes": max(msg_sizes) if msg_sizes else args.hot_keys * args.emb_width * 4,
                 "keys_per_server": args.keys_per_server,

@@ -71,8 +71,20 @@ void Customer::AddResponse(int ts, int num) {
   }
 }
 
+// Depth of customer-handler frames on this thread. The same-process
+// direct-delivery path consults it: a response generated INSIDE a
+// handler must be queued rather than delivered inline, or two threads
+// could take two customers' handle_mu_ in opposite orders (server
+// handler -> respond inline -> worker mutex, vs worker callback ->
+// request inline -> server mutex) and deadlock.
+thread_local int g_in_handler = 0;
+
+bool InCustomerHandler() { return g_in_handler > 0; }
+
 void Customer::RunHandle(Message& msg) {
+  g_in_handler++;
   handle_(msg);
+  g_in_handler--;
   if (!msg.meta.request) AddResponse(msg.meta.timestamp, 1);
 }
 

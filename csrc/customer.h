@@ -20,6 +20,10 @@ namespace xps {
 
 class Postoffice;
 
+// true while the calling thread is inside any Customer handler frame
+// (see the direct-delivery deadlock note in customer.cc)
+bool InCustomerHandler();
+
 class Customer {
  public:
   using RecvHandle = std::function<void(const Message&)>;

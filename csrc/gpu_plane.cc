@@ -141,6 +141,11 @@ void GpuPlane::Stop() {
     for (auto& kv : pending_) {
       for (auto& p : kv.second) {
         (void)hipEventSynchronize(p.ev);
+        if (p.local_po) {
+          DeliverLocal(p.local_po, p.resend, p.bytes);
+          (void)hipEventDestroy(p.ev);
+          continue;
+        }
         Peer* peer = GetPeer(p.peer_id);
         bool sent = peer && EnsureRing(peer) &&
                     peer->ring.Push(p.payload.data(), static_cast<uint32_t>(p.payload.size()));
@@ -375,9 +380,31 @@ void GpuPlane::PutEvent(hipEvent_t ev) {
   event_pool_.push_back(ev);
 }
 
+Postoffice* GpuPlane::LocalPeer(Peer* p) {
+  Postoffice* po = p->local_po.load(std::memory_order_acquire);
+  if (po) return po;
+  if (p->node.id == kEmptyNodeID) return nullptr;
+  po = Postoffice::FindByNodeId(p->node.id);
+  if (po) p->local_po.store(po, std::memory_order_release);
+  return po;
+}
+
+// does this (device-vals) pull response REQUIRE the staging path? The
+// worker-side merge expects host bytes unless an in-place HBM
+// destination was advertised.
+static bool NeedsStaging(const Message& msg) {
+  return !msg.meta.request && msg.meta.pull && msg.data.size() > 1 &&
+         msg.data[1].on_device() && msg.data[1].size() > 0 &&
+         (!(msg.meta.option & kOptPullAddr) || (msg.meta.option & kOptHostAddr));
+}
+
 bool GpuPlane::CanSend(const Message& msg, const Node& peer) {
   if (!started_ || stop_.load()) return false;
   if (peer.host_hash != my_host_hash_ || peer.shm_uid == 0) return false;
+  // same-process peer (joint mode): everything rides the direct path —
+  // no ring-size limits, no pool-membership requirement (one address
+  // space) — except responses that need the TCP host-staging contract
+  if (LocalPeer(GetPeer(peer.id))) return !NeedsStaging(msg);
   // If the bootstrap import of this peer's pool failed, assume the
   // reverse import failed too (same mechanism, same host) and keep
   // everything on the TCP path — slow but never dropped.
@@ -444,6 +471,106 @@ bool GpuPlane::Serialize(const Message& msg, const std::vector<char>& by_ref, st
   return out->size() <= ShmRing::MaxPayload();
 }
 
+void GpuPlane::DeliverLocal(Postoffice* lpo, Message& msg, int64_t bytes) {
+  Van* van = lpo->van();
+  if (!van) {
+    XPS_LOG(Warning) << "local peer " << msg.meta.recver << " already finalized; dropping";
+    return;
+  }
+  van->recv_bytes_ += bytes;
+  // blobs handed over by reference in one address space = zero-copy
+  // reception (the transport-level analog of the reference's
+  // registered-buffer pointer-equality assertion)
+  for (auto& d : msg.data) {
+    if (d.size() && (d.on_device() || d.size() > kInlineMax)) {
+      g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
+    }
+  }
+  if (inline_deliver_ && !InCustomerHandler()) {
+    van->DeliverInline(msg);
+  } else {
+    // a response generated INSIDE a handler frame must be queued: inline
+    // delivery here would nest two customers' handle_mu_ in the opposite
+    // order of the worker-callback -> request chain (see customer.cc)
+    van->Deliver(std::move(msg));
+  }
+}
+
+// Same-process fast path (joint mode: worker + co-located server share
+// the process). This is the reference's IPCTransport/shared_node_mapping_
+// locality taken to its limit: no serialization, no ring hop, no poll
+// thread — a direct call, with GPU completion still gated by the lane
+// event for responses (the requester's buffers must not be released
+// before the kernels that read/write them finish).
+int64_t GpuPlane::SendLocal(Message& msg, Peer* p, Postoffice* lpo) {
+  XPS_STAGE(plane_send_local);
+  bool response = !msg.meta.request;
+  int64_t bytes = 64;
+  for (auto& d : msg.data) bytes += static_cast<int64_t>(d.size());
+
+  // device-vals pull response with an advertised HBM destination: run
+  // the in-place write now, deliver the meta once the copy completes
+  if (response && msg.meta.pull && msg.data.size() > 1 && msg.data[1].on_device() &&
+      (msg.meta.option & kOptPullAddr) && !(msg.meta.option & kOptHostAddr)) {
+    SArray<char> vals = msg.data[1];
+    char* dst = ResolvePeer(p, msg.meta.addr, vals.size());
+    if (!dst) {
+      for (int l = 0; l < lanes_; ++l) (void)hipStreamSynchronize(StreamLane(p->node.id, l));
+      return -1;  // TCP fallback (streams drained first)
+    }
+    hipStream_t stream = (msg.meta.option & kOptPullLane) ? PullStreamForPeer(p->node.id)
+                                                          : StreamForPeer(p->node.id);
+    XPS_HIP_CHECK(hipSetDevice(device_));
+    kern::DenseAssign(dst, vals.data(), vals.size(), stream);
+    g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
+    Message meta_msg;
+    meta_msg.meta = msg.meta;
+    meta_msg.meta.option |= kOptInPlace;
+    meta_msg.meta.val_len = static_cast<int64_t>(vals.size());
+    meta_msg.meta.data_type.clear();
+    for (size_t i = 0; i < msg.data.size(); ++i) {
+      if (i == 1 || msg.data[i].on_device()) continue;
+      meta_msg.data.push_back(msg.data[i]);
+      meta_msg.meta.data_type.push_back(msg.meta.data_type[i]);
+    }
+    msg.data.clear();
+    Message keepalive;
+    keepalive.data.push_back(vals);
+    hipEvent_t ev = GetEvent();
+    XPS_HIP_CHECK(hipEventRecord(ev, stream));
+    {
+      std::lock_guard<std::mutex> lk(pend_mu_);
+      pending_[p->node.id].push_back(Pending{ev, p->node.id, std::string(),
+                                             std::move(meta_msg), std::move(keepalive), bytes,
+                                             lpo});
+    }
+    pending_count_.fetch_add(1);
+    p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
+    return bytes;
+  }
+
+  if (response && device_ >= 0) {
+    // GPU handler output: deliver once this peer's lane drained
+    hipStream_t stream = (msg.meta.option & kOptPullLane) ? PullStreamForPeer(p->node.id)
+                                                          : StreamForPeer(p->node.id);
+    hipEvent_t ev = GetEvent();
+    XPS_HIP_CHECK(hipSetDevice(device_));
+    XPS_HIP_CHECK(hipEventRecord(ev, stream));
+    Message resend = msg;
+    {
+      std::lock_guard<std::mutex> lk(pend_mu_);
+      pending_[p->node.id].push_back(Pending{ev, p->node.id, std::string(), std::move(resend),
+                                             Message(), bytes, lpo});
+    }
+    pending_count_.fetch_add(1);
+  } else {
+    Message copy = msg;
+    DeliverLocal(lpo, copy, bytes);
+  }
+  p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
+  return bytes;
+}
+
 int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   XPS_STAGE(plane_send);
   XPS_VLOG(3) << "plane send -> " << peer_node.id << ": " << msg.DebugString();
@@ -452,6 +579,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     std::lock_guard<std::mutex> lk(p->mu);
     if (p->node.shm_uid == 0) p->node = peer_node;
   }
+  if (Postoffice* lpo = LocalPeer(p)) return SendLocal(msg, p, lpo);
   bool response = !msg.meta.request;
   auto sync_peer_lanes = [this, &peer_node] {
     for (int l = 0; l < lanes_; ++l) (void)hipStreamSynchronize(StreamLane(peer_node.id, l));
@@ -643,6 +771,12 @@ void GpuPlane::CompletionLoop() {
             dq.pop_front();
           }
           pending_count_.fetch_sub(1);
+          if (front.local_po) {  // same-process: direct delivery
+            DeliverLocal(front.local_po, front.resend, front.bytes);
+            PutEvent(front.ev);
+            did = true;
+            continue;
+          }
           Peer* peer = GetPeer(front.peer_id);
           bool sent = peer && EnsureRing(peer) &&
                       peer->ring.Push(front.payload.data(),

@@ -76,6 +76,9 @@ class GpuPlane : public DataPlane {
  private:
   struct Peer {
     Node node;
+    // set when this peer's Postoffice lives in OUR process (joint
+    // mode): sends bypass serialize/ring/poll entirely (direct call)
+    std::atomic<Postoffice*> local_po{nullptr};
     ShmRing ring;  // producer handle on the peer's inbound ring
     bool ring_tried = false;
     // fast-path flag: the ring is opened once and never closed while the
@@ -98,12 +101,20 @@ class GpuPlane : public DataPlane {
     // the event fires, the van sends this instead (never drop a response
     // — the requester is blocked in Wait). For in-place writes this is
     // the meta-only kOptInPlace message (the data already landed).
+    // For same-process peers it is the message delivered directly.
     Message resend;
     Message keepalive;  // holds pool temporaries until the event fires
     int64_t bytes;
+    Postoffice* local_po = nullptr;  // same-process: deliver, don't ring-push
   };
 
   Peer* GetPeer(int id);
+  // the peer's Postoffice when it lives in THIS process, else nullptr
+  Postoffice* LocalPeer(Peer* p);
+  // same-process send: no serialize, no ring — direct delivery (deferred
+  // on the lane event when GPU kernels must complete first)
+  int64_t SendLocal(Message& msg, Peer* p, Postoffice* lpo);
+  void DeliverLocal(Postoffice* lpo, Message& msg, int64_t bytes);
   bool EnsureRing(Peer* p);
   // import every slab of the peer's pool (idempotent)
   bool ImportPeerSlabs(Peer* p);

@@ -88,9 +88,23 @@ bool HbmPool::GrowLocked() {
 void* HbmPool::Alloc(size_t nbytes) {
   XPS_CHECK(!slabs_.empty()) << "HbmPool not initialized";
   nbytes = (nbytes + kAlign - 1) & ~(kAlign - 1);
-  XPS_CHECK_LE(nbytes, slab_bytes_)
-      << "single allocation exceeds the slab size (" << slab_bytes_
-      << " B); raise XPS_SLAB_BYTES (< 2 GiB) or split the buffer";
+  if (nbytes > slab_bytes_) {
+    // oversized (> the <2 GiB hipIpc slab ceiling): dedicated local-only
+    // slab — cannot be zero-copy-shared, but torch-allocator users may
+    // hold multi-GiB tensors that never cross the wire
+    std::lock_guard<std::mutex> lk(mu_);
+    Slab s;
+    s.capacity = nbytes;
+    XPS_HIP_CHECK(hipSetDevice(device_));
+    hipError_t e = hipMalloc(&s.base, s.capacity);
+    XPS_CHECK(e == hipSuccess) << "hipMalloc(" << nbytes << ") failed: "
+                               << hipGetErrorString(e);
+    s.global_begin = ~0ull;
+    s.exported = false;
+    s.used_[0] = nbytes;
+    slabs_.push_back(std::move(s));
+    return slabs_.back().base;
+  }
   std::lock_guard<std::mutex> lk(mu_);
   for (int attempt = 0; attempt < 2; ++attempt) {
     for (auto& slab : slabs_) {
@@ -177,3 +191,38 @@ size_t HbmPool::bytes_in_use() const {
 }
 
 }  // namespace xps
+
+// ---- torch pluggable-allocator hooks --------------------------------
+// Installed via torch.cuda.memory.CUDAPluggableAllocator(_core.so,
+// "xps_torch_alloc", "xps_torch_free"): every torch CUDA tensor then
+// lives in the HbmPool, so plain torch tensors ride the zero-copy
+// hipIpc plane (the reference's PinMemory/RegisterRecvBuffer use case —
+// ucx_van.h:603-623, kv_app.h:488 — without per-buffer registration).
+extern "C" {
+
+void* xps_torch_alloc(ssize_t size, int device, hipStream_t stream) {
+  (void)stream;
+  if (size <= 0) return nullptr;
+  auto* pool = xps::HbmPool::Get();
+  if (!pool->initialized()) pool->Init(device);
+  XPS_CHECK_EQ(pool->device(), device)
+      << "HbmPool torch allocator: one pool per process (one process per GPU)";
+  return pool->Alloc(static_cast<size_t>(size));
+}
+
+void xps_torch_free(void* ptr, ssize_t size, int device, hipStream_t stream) {
+  (void)size;
+  (void)device;
+  if (!ptr) return;
+  // stream-ordered free: the tensor's stream may still have kernels
+  // reading this memory — release the region only once they finished
+  if (stream) {
+    hipError_t e = hipLaunchHostFunc(
+        stream, [](void* p) { xps::HbmPool::Get()->Free(p); }, ptr);
+    if (e == hipSuccess) return;
+  }
+  (void)hipDeviceSynchronize();
+  xps::HbmPool::Get()->Free(ptr);
+}
+
+}  // extern "C"

@@ -166,7 +166,8 @@ std::vector<int> Postoffice::GetDeadNodes(int timeout_sec) {
   std::lock_guard<std::mutex> lk(heartbeat_mu_);
   for (int id : GetNodeIDs(kWorkerGroup | kServerGroup)) {
     auto it = heartbeats_.find(id);
-    if ((it == heartbeats_.end() || it->second + timeout_sec < now) && start_time_ + timeout_sec < now) {
+    if ((it == heartbeats_.end() || it->second + timeout_sec < now) &&
+        start_time_ + timeout_sec < now) {
       dead.push_back(id);
     }
   }
@@ -201,6 +202,21 @@ Postoffice* Postoffice::GetWorker(int idx) {
     reg_workers.emplace_back(new Postoffice(Node::WORKER, reg_workers.size()));
   }
   return reg_workers[idx].get();
+}
+
+Postoffice* Postoffice::FindByNodeId(int id) {
+  std::lock_guard<std::mutex> lk(reg_mu);
+  auto match = [id](Postoffice* po) {
+    return po && po->started_ && po->node_id_ == id && po->van_ && po->van_->IsReady();
+  };
+  if (match(reg_scheduler.get())) return reg_scheduler.get();
+  for (auto& p : reg_servers) {
+    if (match(p.get())) return p.get();
+  }
+  for (auto& p : reg_workers) {
+    if (match(p.get())) return p.get();
+  }
+  return nullptr;
 }
 
 void Postoffice::ClearRegistry() {

@@ -8,6 +8,7 @@
 // except JOINT); MI355X deployments run one JOINT instance pair per GPU.
 #pragma once
 
+#include <atomic>
 #include <condition_variable>
 #include <memory>
 #include <mutex>
@@ -82,6 +83,10 @@ class Postoffice {
   static Postoffice* GetScheduler();
   static Postoffice* GetServer(int idx = 0);
   static Postoffice* GetWorker(int idx = 0);
+  // the STARTED instance in THIS process whose assigned node id is
+  // `id`, or nullptr — the same-process direct-delivery fast path asks
+  // this for every data send (joint mode: worker+server in one process)
+  static Postoffice* FindByNodeId(int id);
   static void ClearRegistry();  // after finalize of everything
 
  private:
@@ -89,7 +94,8 @@ class Postoffice {
 
   int role_;
   int instance_idx_;
-  int node_id_ = kEmptyNodeID;
+  // atomic: FindByNodeId reads it from other threads during bootstrap
+  std::atomic<int> node_id_{kEmptyNodeID};
   int preferred_rank_ = -1;
   int num_workers_ = 0;
   int num_servers_ = 0;

@@ -75,7 +75,9 @@ void GpuDenseHandler::OrderAfter(Entry* e, hipStream_t s) {
   if (e->last_ev) XPS_HIP_CHECK(hipStreamWaitEvent(s, e->last_ev.get(), 0));
 }
 
-hipStream_t GpuDenseHandler::Stream(int sender) { return PeerStream(po_, sender, &fallback_stream_); }
+hipStream_t GpuDenseHandler::Stream(int sender) {
+  return PeerStream(po_, sender, &fallback_stream_);
+}
 
 hipStream_t GpuDenseHandler::PullStream(int sender) {
   return PeerPullStream(po_, sender, &fallback_stream_);

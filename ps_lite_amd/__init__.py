@@ -89,6 +89,23 @@ def pool_tensor(shape, dtype="float32"):
     return t
 
 
+def use_torch_pool_allocator():
+    """Route ALL torch CUDA allocations through the HbmPool (torch
+    pluggable allocator), so ordinary ``torch.empty(..., device='cuda')``
+    tensors live in the pool and ride the zero-copy hipIpc plane — no
+    per-buffer registration, no ``pool_tensor`` wrapper (the reference's
+    PinMemory / RegisterRecvBuffer use case, ucx_van.h:603-623).
+
+    Must be called BEFORE the first CUDA allocation in the process
+    (torch refuses to swap allocators afterwards).
+    """
+    import torch
+
+    alloc = torch.cuda.memory.CUDAPluggableAllocator(
+        _core.__file__, "xps_torch_alloc", "xps_torch_free")
+    torch.cuda.memory.change_current_allocator(alloc)
+
+
 def setup_env(num_workers, num_servers, root_uri="127.0.0.1", root_port=9100, **extra):
     """Set the DMLC_* environment both for this process and for children."""
     env = {

@@ -305,6 +305,42 @@ def test_dp_gradsync_gpu():
 # ------- cross-process hipIpc on a single GPU (config #2 layout x2) -------
 
 
+def _torch_allocator_worker_fn(ps_mod, rank):
+    """A PLAIN torch.empty(device='cuda') tensor (no pool_tensor wrapper)
+    must ride the zero-copy plane once the pluggable allocator routes
+    torch through the HbmPool. Asserted via zero_copy_recv_count (the
+    transport-level analog of the reference's registered-buffer pointer
+    check, test_benchmark.cc:169-181)."""
+    import torch
+
+    ps_mod.use_torch_pool_allocator()  # before any CUDA allocation
+    server = ps_mod.KVServer(0)
+    server.set_gpu_dense_handle(mode="assign")
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker = ps_mod.KVWorker(0, 0)
+    n = 1 << 16
+    src = torch.randn(n, device="cuda:0")          # plain torch tensors,
+    dst = torch.zeros(n, device="cuda:0")          # drawn from the pool
+    torch.cuda.synchronize()
+    before = ps_mod._core.zero_copy_recv_count()
+    keys = np.array([3], dtype=np.uint64)
+    lens = np.array([n], dtype=np.int32)
+    worker.wait(worker.zpush_ptr(keys, src.data_ptr(), n * 4, 0, lens, cmd=1))
+    worker.wait(worker.zpull_ptr(keys, dst.data_ptr(), n * 4, 0, lens))
+    torch.cuda.synchronize()
+    ok_values = bool(torch.allclose(dst, src))
+    zero_copy_delta = ps_mod._core.zero_copy_recv_count() - before
+    return [ok_values, int(zero_copy_delta)], server
+
+
+def test_torch_allocator_zero_copy():
+    results = launch_local(1, 1, _torch_allocator_worker_fn, joint=True, devices={0: 0},
+                           env_extra={"XPS_POOL_GB": 4}, timeout=300)
+    ok_values, zero_copy_delta = results[0]
+    assert ok_values
+    assert zero_copy_delta > 0, "torch tensor did not ride the zero-copy plane"
+
+
 def _gpu_worker_fn(ps_mod, rank):
     server = ps_mod.KVServer(0)
     server.set_gpu_dense_handle(mode="sum")
