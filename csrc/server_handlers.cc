@@ -62,11 +62,17 @@ static EventRef MakeEventRef(Postoffice* po, hipStream_t s) {
 GpuDenseHandler::GpuDenseHandler(Postoffice* po, DenseMode mode) : po_(po), mode_(mode) {
   XPS_CHECK(HbmPool::Get()->initialized()) << "GpuDenseHandler needs the HBM pool";
   num_workers_ = std::max(1, po_->num_workers());
-  // same-key kernels can land on different streams in two cases: >1
-  // worker (one stream set per peer) and >1 lanes per peer (push on
-  // lane 0, pull copies on the pull lane) — both need the last_ev chain
+  // >1 workers: same-key kernels land on different peers' streams
+  chain_ = num_workers_ > 1;
+}
+
+bool GpuDenseHandler::NeedChain(int sender) {
+  if (chain_) return true;
+  // lane split for this peer (cross-device): push kernels on lane 0 and
+  // pull copies on the pull lane touch the same entries — chain them
   auto* plane = ThePlane(po_);
-  chain_ = num_workers_ > 1 || (plane && plane->lanes() > 1);
+  return plane && plane->lanes() > 1 &&
+         plane->PullStreamForPeer(sender) != plane->StreamForPeer(sender);
 }
 
 GpuDenseHandler::~GpuDenseHandler() = default;
@@ -205,6 +211,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   }
   hipStream_t stream = Stream(req.sender);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
+  bool chain = NeedChain(req.sender);
   bool sum_all = mode_ == DenseMode::kAssign ? req.cmd == kCmdSum : req.cmd != kCmdAssign;
   // multi-key device push: one batched kernel launch for all segments
   if (kvs.vals.on_device() && n > 1) {
@@ -239,7 +246,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
       } else {
         kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
       }
-      if (chain_) {
+      if (chain) {
         EventRef ev = MakeEventRef(po_, stream);  // one event covers the batch
         for (Entry* e : ents) e->last_ev = ev;
       }
@@ -275,11 +282,11 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
       } else {
         kern::DenseAssign(e->buf.data(), src, len, stream);
       }
-      if (chain_) e->last_ev = MakeEventRef(po_, stream);
+      if (chain) e->last_ev = MakeEventRef(po_, stream);
     } else {
       // host vals land via synchronous copies below: drain the entry's
       // outstanding cross-stream kernel first
-      if (chain_ && e->last_ev) {
+      if (chain && e->last_ev) {
         XPS_HIP_CHECK(hipEventSynchronize(e->last_ev.get()));
         e->last_ev.reset();
       }
@@ -428,6 +435,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
     return;
   }
   hipStream_t stream = PullStream(req.sender);
+  bool chain = NeedChain(req.sender);
   // multi-key pull with an in-place destination: batched copy of every
   // store entry straight into the requester's mapped pool (no staging
   // buffer, meta-only response)
@@ -464,7 +472,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
         res2.keys = kvs.keys;
         res2.lens = lens2;
         server->Response(r2, res2);  // meta+keys/lens only; plane defers on `stream`
-        if (chain_) {
+        if (chain) {
           EventRef ev = MakeEventRef(po_, stream);  // pushes must wait these reads
           for (Entry* e : ents) e->last_ev = ev;
         }
@@ -529,7 +537,7 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   KVMeta r = req;
   r.option |= kOptPullLane;  // ordering prepared on the pull lane
   server->Response(r, res);  // plane enqueues the in-place read on `stream`
-  if (chain_) {
+  if (chain) {
     EventRef ev = MakeEventRef(po_, stream);  // pushes must wait these reads
     for (Entry* e : touched) e->last_ev = ev;
   }

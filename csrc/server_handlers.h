@@ -93,6 +93,9 @@ class GpuDenseHandler {
   Group* GroupFor(const SArray<Key>& keys);  // find-or-create round group
   hipStream_t Stream(int sender);      // lane 0: push/accumulate kernels
   hipStream_t PullStream(int sender);  // pull lane: response copies
+  // same-key event chaining needed for this sender? (>1 workers, or the
+  // peer's lanes actually split = cross-device)
+  bool NeedChain(int sender);
   void OrderAfter(Entry* e, hipStream_t s);  // wait the entry's last_ev
 
   Postoffice* po_;
