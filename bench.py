@@ -50,6 +50,9 @@ def parse_args():
                         "model does not grow with the cluster -> honest weak "
                         "scaling of per-worker work)")
     p.add_argument("--op", choices=["assign", "sum"], default="assign")
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                   help="payload dtype for the dense sum/reduce kernels "
+                        "(bf16 halves bytes moved; headline stays fp32)")
     p.add_argument("--batch-keys", action="store_true",
                    help="dense: one multi-key message per server per round")
     p.add_argument("--per-key", action="store_true",
@@ -186,7 +189,8 @@ def main():
         server.set_gpu_sparse_handle(spec.rows_local(n), spec.width, accumulate=True,
                                      key_shift=spec.key_shift)
     else:
-        server.set_gpu_dense_handle(mode=("reduce" if args.mode == "rn50" else args.op))
+        server.set_gpu_dense_handle(mode=("reduce" if args.mode == "rn50" else args.op),
+                                    dtype=("bf16" if args.dtype == "bf16" else "f32"))
     trace("server handler installed")
     worker = ps.KVWorker(0, 0)
     ps.barrier("worker", ps.WORKER_GROUP)
@@ -374,7 +378,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": model_name,

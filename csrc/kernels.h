@@ -18,6 +18,11 @@ namespace kern {
 void DenseAssign(void* dst, const void* src, size_t nbytes, hipStream_t s);
 // dst[i] += src[i] (fp32)
 void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s);
+// dst[i] += src[i] (bf16 payloads, fp32 accumulate in-register,
+// round-to-nearest-even back to bf16 — halves bytes moved on the
+// bandwidth-bound dense path)
+void DenseSumBf16(uint16_t* dst, const uint16_t* src, size_t n, hipStream_t s);
+void BatchedSumBf16(const struct CopyDesc* descs_host, int n, hipStream_t s);
 // dst[i] += sum_j srcs[j][i], up to 8 sources in one pass (one read of
 // dst, one write — HBM-optimal multi-worker reduction)
 // Sparse ops: local row index = (rows[r] >> key_shift) - row_base, so a

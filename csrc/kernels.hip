@@ -62,6 +62,57 @@ __global__ void sum_tail_f32(float* __restrict__ dst, const float* __restrict__ 
   if (i < end) dst[i] += src[i];
 }
 
+// bf16 += bf16 with fp32 accumulate in-register, 16 B per lane (8 bf16
+// elements via uint4). bf16 -> fp32 is a 16-bit shift; fp32 -> bf16
+// rounds to nearest-even (matches torch's bf16 semantics).
+__device__ inline float bf16_to_f32(uint16_t h) {
+  union {
+    uint32_t u;
+    float f;
+  } c;
+  c.u = static_cast<uint32_t>(h) << 16;
+  return c.f;
+}
+
+__device__ inline uint16_t f32_to_bf16(float f) {
+  union {
+    uint32_t u;
+    float f;
+  } c;
+  c.f = f;
+  uint32_t lsb = (c.u >> 16) & 1u;
+  return static_cast<uint16_t>((c.u + 0x7FFFu + lsb) >> 16);
+}
+
+__device__ inline uint32_t bf16x2_sum(uint32_t d, uint32_t s) {
+  float lo = bf16_to_f32(static_cast<uint16_t>(d)) + bf16_to_f32(static_cast<uint16_t>(s));
+  float hi = bf16_to_f32(static_cast<uint16_t>(d >> 16)) +
+             bf16_to_f32(static_cast<uint16_t>(s >> 16));
+  return static_cast<uint32_t>(f32_to_bf16(lo)) |
+         (static_cast<uint32_t>(f32_to_bf16(hi)) << 16);
+}
+
+__global__ void sum_kernel_bf16(uint4* __restrict__ dst, const uint4* __restrict__ src,
+                                size_t n8) {
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (; i < n8; i += stride) {
+    uint4 d = dst[i];
+    uint4 s = src[i];
+    d.x = bf16x2_sum(d.x, s.x);
+    d.y = bf16x2_sum(d.y, s.y);
+    d.z = bf16x2_sum(d.z, s.z);
+    d.w = bf16x2_sum(d.w, s.w);
+    dst[i] = d;
+  }
+}
+
+__global__ void sum_tail_bf16(uint16_t* __restrict__ dst, const uint16_t* __restrict__ src,
+                              size_t begin, size_t end) {
+  size_t i = begin + blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  if (i < end) dst[i] = f32_to_bf16(bf16_to_f32(dst[i]) + bf16_to_f32(src[i]));
+}
+
 // flat indexing: thread i handles element i of the CONCATENATED rows
 // (row = i / row_len4), so narrow rows (e.g. 64 floats = 16 float4)
 // still use every lane — a row-per-block mapping left 94 % of the block
@@ -184,6 +235,25 @@ __global__ void batched_assign_kernel(DescArray da) {
   }
 }
 
+__global__ void batched_sum_kernel_bf16(DescArray da) {
+  for (int seg = 0; seg < da.n; ++seg) {
+    uint4* dst = reinterpret_cast<uint4*>(da.d[seg].dst);
+    const uint4* src = reinterpret_cast<const uint4*>(da.d[seg].src);
+    size_t n8 = da.d[seg].nbytes / 16;
+    size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+    size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+    for (; i < n8; i += stride) {
+      uint4 d = dst[i];
+      uint4 s = src[i];
+      d.x = bf16x2_sum(d.x, s.x);
+      d.y = bf16x2_sum(d.y, s.y);
+      d.z = bf16x2_sum(d.z, s.z);
+      d.w = bf16x2_sum(d.w, s.w);
+      dst[i] = d;
+    }
+  }
+}
+
 __global__ void batched_sum_kernel_f32(DescArray da) {
   for (int seg = 0; seg < da.n; ++seg) {
     float4* dst = reinterpret_cast<float4*>(da.d[seg].dst);
@@ -247,6 +317,29 @@ void DenseSumF32(float* dst, const float* src, size_t n, hipStream_t s) {
   }
   if (n % 4) {
     hipLaunchKernelGGL(sum_tail_f32, dim3(1), dim3(kBlock), 0, s, dst, src, n4 * 4, n);
+  }
+}
+
+void DenseSumBf16(uint16_t* dst, const uint16_t* src, size_t n, hipStream_t s) {
+  size_t n8 = n / 8;
+  if (n8) {
+    hipLaunchKernelGGL(sum_kernel_bf16, dim3(GridFor(n8)), dim3(kBlock), 0, s,
+                       reinterpret_cast<uint4*>(dst), reinterpret_cast<const uint4*>(src), n8);
+  }
+  if (n % 8) {
+    hipLaunchKernelGGL(sum_tail_bf16, dim3(1), dim3(kBlock), 0, s, dst, src, n8 * 8, n);
+  }
+}
+
+void BatchedSumBf16(const CopyDesc* descs_host, int n, hipStream_t s) {
+  size_t total = 0;
+  for (int i = 0; i < n; ++i) total += descs_host[i].nbytes;
+  DescArray da{};
+  for (int off = 0; off < n; off += kMaxBatch) {
+    da.n = std::min(n - off, kMaxBatch);
+    for (int i = 0; i < da.n; ++i) da.d[i] = descs_host[off + i];
+    hipLaunchKernelGGL(batched_sum_kernel_bf16, dim3(GridFor(total / 16)), dim3(kBlock), 0, s,
+                       da);
   }
 }
 

@@ -292,11 +292,12 @@ class PyKVServer {
     }
   }
 
-  void SetGpuDenseHandle(const std::string& mode) {
+  void SetGpuDenseHandle(const std::string& mode, const std::string& dtype) {
     DenseMode m = mode == "sum" ? DenseMode::kSum
                   : mode == "reduce" ? DenseMode::kReduce
                                      : DenseMode::kAssign;
-    auto h = std::make_shared<GpuDenseHandler>(s_.postoffice(), m);
+    DenseDtype d = dtype == "bf16" ? DenseDtype::kBf16 : DenseDtype::kF32;
+    auto h = std::make_shared<GpuDenseHandler>(s_.postoffice(), m, d);
     dense_ = h;
     s_.set_request_handle([h](const KVMeta& m2, const KVPairs<float>& kvs, KVServer<float>* srv) {
       (*h)(m2, kvs, srv);
@@ -532,7 +533,8 @@ PYBIND11_MODULE(_core, m) {
       .def(py::init<int, int>(), py::arg("app_id") = 0, py::arg("instance_idx") = 0)
       .def("set_default_handle", &PyKVServer::SetDefaultHandle)
       .def("set_reduce_handle", &PyKVServer::SetReduceHandle, py::arg("num_workers"))
-      .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("mode") = "assign")
+      .def("set_gpu_dense_handle", &PyKVServer::SetGpuDenseHandle, py::arg("mode") = "assign",
+           py::arg("dtype") = "f32")
       .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),
            py::arg("row_len"), py::arg("accumulate") = true, py::arg("key_shift") = 0)
       .def("sparse_table_ptr", &PyKVServer::SparseTablePtr)
@@ -794,6 +796,11 @@ PYBIND11_MODULE(_core, m) {
   });
   m.def("k_dense_sum_f32", [](uintptr_t dst, uintptr_t src, size_t n) {
     kern::DenseSumF32(reinterpret_cast<float*>(dst), reinterpret_cast<float*>(src), n, nullptr);
+    gpu::DeviceSync(-1);
+  });
+  m.def("k_dense_sum_bf16", [](uintptr_t dst, uintptr_t src, size_t n) {
+    kern::DenseSumBf16(reinterpret_cast<uint16_t*>(dst), reinterpret_cast<uint16_t*>(src), n,
+                       nullptr);
     gpu::DeviceSync(-1);
   });
   m.def("k_sparse_gather_f32",

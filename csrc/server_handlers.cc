@@ -59,11 +59,29 @@ static EventRef MakeEventRef(Postoffice* po, hipStream_t s) {
   return EventRef(ev, [po](hipEvent_t e) { EventFree(po, e); });
 }
 
-GpuDenseHandler::GpuDenseHandler(Postoffice* po, DenseMode mode) : po_(po), mode_(mode) {
+GpuDenseHandler::GpuDenseHandler(Postoffice* po, DenseMode mode, DenseDtype dtype)
+    : po_(po), mode_(mode), dtype_(dtype) {
   XPS_CHECK(HbmPool::Get()->initialized()) << "GpuDenseHandler needs the HBM pool";
   num_workers_ = std::max(1, po_->num_workers());
   // >1 workers: same-key kernels land on different peers' streams
   chain_ = num_workers_ > 1;
+}
+
+void GpuDenseHandler::SumKernel(void* dst, const void* src, size_t nbytes, hipStream_t s) {
+  if (dtype_ == DenseDtype::kBf16) {
+    kern::DenseSumBf16(static_cast<uint16_t*>(dst), static_cast<const uint16_t*>(src),
+                       nbytes / 2, s);
+  } else {
+    kern::DenseSumF32(static_cast<float*>(dst), static_cast<const float*>(src), nbytes / 4, s);
+  }
+}
+
+void GpuDenseHandler::BatchedSum(const kern::CopyDesc* descs, int n, hipStream_t s) {
+  if (dtype_ == DenseDtype::kBf16) {
+    kern::BatchedSumBf16(descs, n, s);
+  } else {
+    kern::BatchedSumF32(descs, n, s);
+  }
 }
 
 bool GpuDenseHandler::NeedChain(int sender) {
@@ -175,16 +193,14 @@ void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& 
     if (first) {
       kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
     } else {
-      kern::BatchedSumF32(descs.data(), static_cast<int>(n), stream);
+      BatchedSum(descs.data(), static_cast<int>(n), stream);
     }
   } else {
     for (auto& d : descs) {
       if (first) {
         kern::DenseAssign(d.dst, d.src, d.nbytes, stream);
       } else {
-        kern::DenseSumF32(reinterpret_cast<float*>(d.dst),
-                          reinterpret_cast<const float*>(d.src), d.nbytes / sizeof(float),
-                          stream);
+        SumKernel(d.dst, d.src, d.nbytes, stream);
       }
     }
   }
@@ -242,7 +258,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
     if (aligned) {
       for (Entry* e : ents) OrderAfter(e, stream);
       if (sum_all) {
-        kern::BatchedSumF32(descs.data(), static_cast<int>(n), stream);
+        BatchedSum(descs.data(), static_cast<int>(n), stream);
       } else {
         kern::BatchedAssign(descs.data(), static_cast<int>(n), stream);
       }
@@ -277,8 +293,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
     if (kvs.vals.on_device()) {
       OrderAfter(e, stream);
       if (sum) {
-        kern::DenseSumF32(reinterpret_cast<float*>(e->buf.data()),
-                          reinterpret_cast<const float*>(src), len / sizeof(float), stream);
+        SumKernel(e->buf.data(), src, len, stream);
       } else {
         kern::DenseAssign(e->buf.data(), src, len, stream);
       }
@@ -294,9 +309,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
       if (sum) {
         SArray<char> scratch = HbmPool::Get()->AllocArray(len);
         XPS_HIP_CHECK(hipMemcpy(scratch.data(), src, len, hipMemcpyHostToDevice));
-        kern::DenseSumF32(reinterpret_cast<float*>(e->buf.data()),
-                          reinterpret_cast<const float*>(scratch.data()), len / sizeof(float),
-                          stream);
+        SumKernel(e->buf.data(), scratch.data(), len, stream);
         XPS_HIP_CHECK(hipStreamSynchronize(stream));
         synced = true;
       } else {

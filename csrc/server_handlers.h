@@ -18,6 +18,7 @@
 #include <type_traits>
 #include <unordered_map>
 
+#include "kernels.h"
 #include "kv_app.h"
 
 namespace xps {
@@ -38,9 +39,14 @@ enum class DenseMode {
             // events), and the round resets after num_workers pulls.
 };
 
+// payload dtype of the accumulate kernels (the wire/pools are bytes;
+// bf16 halves bytes moved on the bandwidth-bound dense path). Assign
+// mode is dtype-agnostic byte copies either way.
+enum class DenseDtype { kF32, kBf16 };
+
 class GpuDenseHandler {
  public:
-  GpuDenseHandler(Postoffice* po, DenseMode mode);
+  GpuDenseHandler(Postoffice* po, DenseMode mode, DenseDtype dtype = DenseDtype::kF32);
   ~GpuDenseHandler();
   void operator()(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
 
@@ -98,8 +104,13 @@ class GpuDenseHandler {
   bool NeedChain(int sender);
   void OrderAfter(Entry* e, hipStream_t s);  // wait the entry's last_ev
 
+  // dtype-dispatched accumulate (nbytes, not elements)
+  void SumKernel(void* dst, const void* src, size_t nbytes, hipStream_t s);
+  void BatchedSum(const kern::CopyDesc* descs, int n, hipStream_t s);
+
   Postoffice* po_;
   DenseMode mode_;
+  DenseDtype dtype_;
   int num_workers_ = 1;
   bool chain_ = false;  // >1 workers: cross-stream same-key ordering needed
   std::mutex mu_;

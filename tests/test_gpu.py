@@ -51,6 +51,20 @@ def test_kernel_dense_sum_vs_torch():
     assert torch.allclose(a.cpu(), ref)
 
 
+def test_kernel_dense_sum_bf16_vs_torch():
+    """bf16 accumulate (fp32 in-register, RNE back to bf16) must match
+    torch's bf16 add exactly — same rounding rule."""
+    import torch
+
+    torch.manual_seed(1)
+    n = (1 << 20) + 5  # exercise the tail path
+    a = torch.randn(n, device="cuda:0", dtype=torch.bfloat16)
+    b = torch.randn(n, device="cuda:0", dtype=torch.bfloat16)
+    ref = (a + b).cpu()
+    ps._core.k_dense_sum_bf16(a.data_ptr(), b.data_ptr(), n)
+    assert torch.equal(a.cpu(), ref)
+
+
 def test_kernel_dense_assign_vs_torch():
     import torch
 
@@ -194,6 +208,37 @@ def test_sparse_handler_gpu():
         worker.wait(worker.zpull_ptr(idx, dst.ptr, grads.nbytes, 0, lens))
         out = dst.to_numpy_f32().reshape(nsel, width)
         assert np.allclose(out, grads, atol=1e-6)
+    finally:
+        _down_joint()
+
+
+def test_dense_bf16_accumulate_gpu():
+    """End-to-end bf16 dense sum: handler in dtype=bf16 mode accumulates
+    bf16 payloads exactly like torch's bf16 add (halves bytes moved on
+    the bandwidth-bound path)."""
+    import torch
+
+    _boot_joint_inproc()
+    try:
+        n = 1 << 16
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(mode="sum", dtype="bf16")
+        worker = ps.KVWorker(0, 0)
+        torch.manual_seed(2)
+        vals = torch.randn(n, dtype=torch.bfloat16)
+        src = ps.pool_alloc(n * 2)
+        dst = ps.pool_alloc(n * 2)
+        src.copy_from(vals.view(torch.uint16).numpy())
+        keys = np.array([12], dtype=np.uint64)
+        lens = np.array([n // 2], dtype=np.int32)  # lens are 4-byte units
+        ref = torch.zeros(n, dtype=torch.bfloat16)
+        for _ in range(3):
+            worker.wait(worker.zpush_ptr(keys, src.ptr, n * 2, 0, lens, cmd=2))
+            ref = ref + vals
+        worker.wait(worker.zpull_ptr(keys, dst.ptr, n * 2, 0, lens))
+        got = torch.from_numpy(
+            dst.to_numpy_f32().view(np.uint16)[:n].copy()).view(torch.bfloat16)
+        assert torch.equal(got, ref)
     finally:
         _down_joint()
 
