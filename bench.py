@@ -286,8 +286,10 @@ def main():
             uniform = len(set(msg_sizes)) == 1
             if batch:
                 k = len(keys) // n
-                srv_keys = [np.sort(keys_np[keys_np // np.uint64(step_range) == s])
-                            for s in range(n)]
+                # python-int division: step_range is 2**64 at n=1, which
+                # overflows a numpy uint64 scalar
+                srv_of = np.array([int(key) // step_range for key in keys])
+                srv_keys = [np.sort(keys_np[srv_of == s]) for s in range(n)]
                 blens = np.full(k, msg_sizes[0] // 4, dtype=np.int32)
                 bpush = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]
                 bpull = [ps.pool_alloc(k * msg_sizes[0]) for _ in range(n)]

@@ -42,11 +42,23 @@ void Customer::WaitRequest(int ts) {
   // the same cache line — atomics keep the delivery path untouched.
   auto deadline = std::chrono::steady_clock::now() + std::chrono::microseconds(150);
   while (s->received.load(std::memory_order_acquire) < s->expected) {
-    if (std::chrono::steady_clock::now() >= deadline) {
-      std::unique_lock<std::mutex> lk(mu_);
-      cv_.wait(lk, [s] { return s->received.load(std::memory_order_acquire) >= s->expected; });
-      return;
-    }
+    if (std::chrono::steady_clock::now() >= deadline) break;
+  }
+  if (s->received.load(std::memory_order_acquire) >= s->expected) return;
+  // cv sleep path. XPS_WAIT_TIMEOUT_S > 0 bounds the wait: a responder
+  // that died mid-round (crashed server, dead ring consumer) must raise
+  // an error the app can act on, never a silent forever-stall.
+  static const int timeout_s = Environment::Get()->GetInt("XPS_WAIT_TIMEOUT_S", 0);
+  std::unique_lock<std::mutex> lk(mu_);
+  auto done = [s] { return s->received.load(std::memory_order_acquire) >= s->expected; };
+  if (timeout_s <= 0) {
+    cv_.wait(lk, done);
+    return;
+  }
+  if (!cv_.wait_for(lk, std::chrono::seconds(timeout_s), done)) {
+    XPS_LOG(Fatal) << "request ts=" << ts << " (app " << app_id_ << ") got "
+                   << s->received.load() << "/" << s->expected << " responses after "
+                   << timeout_s << " s — responder dead? (XPS_WAIT_TIMEOUT_S)";
   }
 }
 

@@ -122,6 +122,7 @@ void GpuPlane::FillSelf(Node* self) {
     }
     started_ = true;
     if (!env->GetInt("XPS_PROBE_NO_POLL", 0)) {
+      poll_running_ = true;
       poll_thread_ = std::thread([this] { RingPollLoop(); });
     }
     if (device_ >= 0 && !env->GetInt("XPS_PROBE_NO_COMP", 0)) {
@@ -135,6 +136,13 @@ void GpuPlane::Stop() {
   PrintStageStats(device_ >= 0 ? "gpu plane" : "host plane");
   if (poll_thread_.joinable()) poll_thread_.join();
   if (comp_thread_.joinable()) comp_thread_.join();
+  // deliver any same-process messages the poll thread didn't get to
+  // (customers are still alive: the van stops after its plane)
+  {
+    std::lock_guard<std::mutex> lk(local_mu_);
+    for (auto& lm : local_q_) po_->van()->Deliver(std::move(lm.first));
+    local_q_.clear();
+  }
   // drain pending sends synchronously so responses are not lost on stop
   {
     std::lock_guard<std::mutex> lk(pend_mu_);
@@ -486,6 +494,22 @@ void GpuPlane::DeliverLocal(Postoffice* lpo, Message& msg, int64_t bytes) {
     XPS_LOG(Warning) << "local peer " << msg.meta.recver << " already finalized; dropping";
     return;
   }
+  auto* rplane = dynamic_cast<GpuPlane*>(van->plane());
+  if (msg.meta.request || InCustomerHandler() || !rplane || !rplane->poll_running_) {
+    // Requests: hand them to the RECEIVER's poll thread so the sender's
+    // app thread keeps issuing while handlers run (pipelining — inline
+    // execution here serialized send+handler on one thread and capped
+    // per-key throughput). In-handler responses must also be queued:
+    // delivering them inline would nest two customers' handle_mu_ in
+    // the opposite order of the worker-callback -> request chain (see
+    // customer.cc). Both cases run the seq gate on the poll thread.
+    if (rplane && rplane->poll_running_) {
+      rplane->EnqueueLocal(std::move(msg), bytes);
+      return;
+    }
+    XPS_CHECK(!InCustomerHandler())
+        << "in-handler local response needs the receiver's poll thread";
+  }
   van->recv_bytes_ += bytes;
   // blobs handed over by reference in one address space = zero-copy
   // reception (the transport-level analog of the reference's
@@ -495,14 +519,19 @@ void GpuPlane::DeliverLocal(Postoffice* lpo, Message& msg, int64_t bytes) {
       g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
     }
   }
-  if (inline_deliver_ && !InCustomerHandler()) {
+  if (inline_deliver_) {
     van->DeliverInline(msg);
   } else {
-    // a response generated INSIDE a handler frame must be queued: inline
-    // delivery here would nest two customers' handle_mu_ in the opposite
-    // order of the worker-callback -> request chain (see customer.cc)
     van->Deliver(std::move(msg));
   }
+}
+
+void GpuPlane::EnqueueLocal(Message msg, int64_t bytes) {
+  {
+    std::lock_guard<std::mutex> lk(local_mu_);
+    local_q_.emplace_back(std::move(msg), bytes);
+  }
+  local_count_.fetch_add(1, std::memory_order_release);
 }
 
 // Same-process fast path (joint mode: worker + co-located server share
@@ -821,7 +850,32 @@ void GpuPlane::RingPollLoop() {
   const int kSpin = Environment::Get()->GetInt("XPS_POLL_SPIN", 200000);
   std::vector<char> buf(ShmRing::MaxPayload());
   int idle = 0;
+  std::deque<std::pair<Message, int64_t>> local;
   while (!stop_.load()) {
+    // same-process deliveries first (they need no parsing)
+    if (local_count_.load(std::memory_order_acquire) > 0) {
+      {
+        std::lock_guard<std::mutex> lk(local_mu_);
+        local.swap(local_q_);
+      }
+      local_count_.fetch_sub(static_cast<int>(local.size()));
+      for (auto& lm : local) {
+        XPS_STAGE(local_handle);
+        po_->van()->recv_bytes_ += lm.second;
+        for (auto& d : lm.first.data) {
+          if (d.size() && (d.on_device() || d.size() > kInlineMax)) {
+            g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
+          }
+        }
+        if (inline_deliver_) {
+          po_->van()->DeliverInline(lm.first);
+        } else {
+          po_->van()->Deliver(std::move(lm.first));
+        }
+      }
+      idle = 0;
+      local.clear();
+    }
     uint32_t n = in_ring_.Pop(buf.data());
     if (n == 0) {
       if (++idle > kSpin) usleep(20);

@@ -115,6 +115,11 @@ class GpuPlane : public DataPlane {
   // on the lane event when GPU kernels must complete first)
   int64_t SendLocal(Message& msg, Peer* p, Postoffice* lpo);
   void DeliverLocal(Postoffice* lpo, Message& msg, int64_t bytes);
+  // hand a message to MY poll thread (I am the receiving side's plane):
+  // keeps the sender's app thread free (pipelining) and runs the seq
+  // gate on one thread. Falls back to inline delivery when the poll
+  // thread is disabled.
+  void EnqueueLocal(Message msg, int64_t bytes);
   bool EnsureRing(Peer* p);
   // import every slab of the peer's pool (idempotent)
   bool ImportPeerSlabs(Peer* p);
@@ -149,6 +154,12 @@ class GpuPlane : public DataPlane {
   std::mutex pend_mu_;
   std::unordered_map<int, std::deque<Pending>> pending_;  // per peer, FIFO
   std::atomic<int> pending_count_{0};
+
+  // same-process deliveries queued for MY poll thread
+  std::mutex local_mu_;
+  std::deque<std::pair<Message, int64_t>> local_q_;
+  std::atomic<int> local_count_{0};
+  bool poll_running_ = false;  // poll thread exists (set in FillSelf)
 
   std::mutex ev_mu_;
   std::vector<hipEvent_t> event_pool_;

@@ -82,3 +82,71 @@ def test_worker_recovery():
     assert seen.get("worker2") == "ok", seen
     for p in procs:
         p.join(timeout=30)
+
+
+# ---- plane liveness: a dead responder must RAISE, never stall ----------
+
+
+def _chaos_role(role, port, q, ev_kill):
+    env = dict(ENV_BASE)
+    env["DMLC_PS_ROOT_PORT"] = str(port)
+    env["XPS_WAIT_TIMEOUT_S"] = "5"
+    os.environ.update(env)
+    sys.path.insert(0, str(REPO))
+    import ps_lite_amd as ps
+
+    ps.start(role=role, device=-1)
+    if role == "scheduler":
+        q.put(("scheduler", "up"))
+        ev_kill.wait(90)
+        time.sleep(20)  # outlive the worker's timeout window
+        os._exit(0)
+    elif role == "server":
+        server = ps.KVServer(0)
+        server.set_default_handle()
+        q.put(("server", "up"))
+        ev_kill.wait(90)
+        os._exit(0)  # SIGKILL-style death mid-round: no finalize, no unlink
+    else:  # worker
+        worker = ps.KVWorker(0, 0)
+        keys = np.array([9], dtype=np.uint64)
+        vals = np.ones(256, dtype=np.float32)
+        lens = np.array([256], dtype=np.int32)
+        worker.wait(worker.push(keys, vals, lens))  # round 1: server alive
+        ev_kill.set()
+        time.sleep(2)  # let the server die
+        t0 = time.time()
+        try:
+            worker.wait(worker.push(keys, vals, lens))
+            q.put(("worker", "no-error"))
+        except RuntimeError as e:
+            took = time.time() - t0
+            ok = "responder dead" in str(e) and took < 30
+            q.put(("worker", "raised" if ok else f"bad:{took:.0f}s:{e}"))
+        time.sleep(1)  # let the queue feeder thread flush before dying
+        os._exit(0)  # cluster is broken; no orderly finalize
+
+
+def test_dead_server_raises_not_stalls():
+    """VERDICT round-1 weak #4: a response lost to a dead plane consumer
+    hung the worker forever. With XPS_WAIT_TIMEOUT_S the blocked Wait
+    raises a clear error within the timeout."""
+    port = random.randint(21000, 50000)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ev_kill = ctx.Event()
+    procs = []
+    for role in ("scheduler", "server", "worker"):
+        p = ctx.Process(target=_chaos_role, args=(role, port, q, ev_kill), daemon=True)
+        p.start()
+        procs.append(p)
+    seen = {}
+    deadline = time.time() + 120
+    while "worker" not in seen and time.time() < deadline:
+        role, st = q.get(timeout=120)
+        seen[role] = st
+    assert seen.get("worker") == "raised", seen
+    for p in procs:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
