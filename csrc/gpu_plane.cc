@@ -546,6 +546,36 @@ int64_t GpuPlane::SendLocal(Message& msg, Peer* p, Postoffice* lpo) {
   int64_t bytes = 64;
   for (auto& d : msg.data) bytes += static_cast<int64_t>(d.size());
 
+  // ---- pull response, HOST vals + advertised HOST destination: copy
+  // into the requester's buffer NOW (the synchronous-handler contract —
+  // the store may mutate right after Response returns), meta-only after
+  if (response && msg.meta.pull && msg.data.size() > 1 && !msg.data[1].on_device() &&
+      msg.data[1].size() > 0 && (msg.meta.option & kOptPullAddr) &&
+      (msg.meta.option & kOptHostAddr)) {
+    void* base = HostShmPool::MapPeer(p->node.host_pool_uid, p->node.host_pool_capacity);
+    SArray<char> vals = msg.data[1];
+    if (base && msg.meta.addr <= p->node.host_pool_capacity &&
+        vals.size() <= p->node.host_pool_capacity - msg.meta.addr) {
+      HostPar::CopyBytes(static_cast<char*>(base) + msg.meta.addr, vals.data(), vals.size());
+      g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
+      Message meta_msg;
+      meta_msg.meta = msg.meta;
+      meta_msg.meta.option |= kOptInPlace;
+      meta_msg.meta.val_len = static_cast<int64_t>(vals.size());
+      meta_msg.meta.data_type.clear();
+      for (size_t i = 0; i < msg.data.size(); ++i) {
+        if (i == 1 || msg.data[i].on_device()) continue;
+        SArray<char> copy(msg.data[i].size());  // snapshot small keys/lens too
+        memcpy(copy.data(), msg.data[i].data(), msg.data[i].size());
+        meta_msg.data.push_back(copy);
+        meta_msg.meta.data_type.push_back(msg.meta.data_type[i]);
+      }
+      DeliverLocal(lpo, meta_msg, bytes);
+      p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
+      return bytes;
+    }
+  }
+
   // device-vals pull response with an advertised HBM destination: run
   // the in-place write now, deliver the meta once the copy completes
   if (response && msg.meta.pull && msg.data.size() > 1 && msg.data[1].on_device() &&
@@ -603,6 +633,20 @@ int64_t GpuPlane::SendLocal(Message& msg, Peer* p, Postoffice* lpo) {
     pending_count_.fetch_add(1);
   } else {
     Message copy = msg;
+    if (response) {
+      // synchronous (host) handler responses may be VIEWS of mutable
+      // server state — the old wire paths serialized them before the
+      // handler could mutate again; snapshot host blobs to keep that
+      // contract (requests need no copy: the sender's buffers are
+      // pinned until the response arrives)
+      for (auto& d : copy.data) {
+        if (!d.on_device() && d.size()) {
+          SArray<char> snap(d.size());
+          memcpy(snap.data(), d.data(), d.size());
+          d = snap;
+        }
+      }
+    }
     DeliverLocal(lpo, copy, bytes);
   }
   p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
