@@ -104,7 +104,17 @@ def reduce_max(values, rank, world):
     import torch.distributed as dist
 
     if not dist.is_initialized():
-        dist.init_process_group("gloo", rank=rank, world_size=world)
+        # gloo prints "[Gloo] Rank ... connected ..." to C-level stdout;
+        # the driver parses our stdout for ONE json line — route the
+        # noise to stderr during init
+        sys.stdout.flush()
+        saved = os.dup(1)
+        try:
+            os.dup2(2, 1)
+            dist.init_process_group("gloo", rank=rank, world_size=world)
+        finally:
+            os.dup2(saved, 1)
+            os.close(saved)
     t = torch.tensor(values, dtype=torch.float64)
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     return t.tolist()

@@ -495,20 +495,20 @@ void GpuPlane::DeliverLocal(Postoffice* lpo, Message& msg, int64_t bytes) {
     return;
   }
   auto* rplane = dynamic_cast<GpuPlane*>(van->plane());
-  if (msg.meta.request || InCustomerHandler() || !rplane || !rplane->poll_running_) {
-    // Requests: hand them to the RECEIVER's poll thread so the sender's
-    // app thread keeps issuing while handlers run (pipelining — inline
-    // execution here serialized send+handler on one thread and capped
-    // per-key throughput). In-handler responses must also be queued:
-    // delivering them inline would nest two customers' handle_mu_ in
-    // the opposite order of the worker-callback -> request chain (see
-    // customer.cc). Both cases run the seq gate on the poll thread.
+  // In-handler sends must be queued to the receiver's poll thread:
+  // delivering them inline would nest two customers' handle_mu_ in the
+  // opposite order of the worker-callback -> request chain (see
+  // customer.cc). Requests deliver inline by default (lowest latency —
+  // measured better than queue-pipelining for single-flow traffic);
+  // XPS_LOCAL_QUEUE_REQ=1 opts into the queued/pipelined variant.
+  static const bool queue_req = Environment::Get()->GetInt("XPS_LOCAL_QUEUE_REQ", 0) != 0;
+  if (InCustomerHandler() || (queue_req && msg.meta.request)) {
     if (rplane && rplane->poll_running_) {
       rplane->EnqueueLocal(std::move(msg), bytes);
       return;
     }
     XPS_CHECK(!InCustomerHandler())
-        << "in-handler local response needs the receiver's poll thread";
+        << "in-handler local send needs the receiver's poll thread";
   }
   van->recv_bytes_ += bytes;
   // blobs handed over by reference in one address space = zero-copy
@@ -619,6 +619,7 @@ int64_t GpuPlane::SendLocal(Message& msg, Peer* p, Postoffice* lpo) {
 
   if (response && device_ >= 0) {
     // GPU handler output: deliver once this peer's lane drained
+    XPS_STAGE(local_defer);
     hipStream_t stream = (msg.meta.option & kOptPullLane) ? PullStreamForPeer(p->node.id)
                                                           : StreamForPeer(p->node.id);
     hipEvent_t ev = GetEvent();

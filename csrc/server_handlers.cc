@@ -292,12 +292,18 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
     const char* src = reinterpret_cast<const char*>(kvs.vals.data()) + off;
     if (kvs.vals.on_device()) {
       OrderAfter(e, stream);
-      if (sum) {
-        SumKernel(e->buf.data(), src, len, stream);
-      } else {
-        kern::DenseAssign(e->buf.data(), src, len, stream);
+      {
+        XPS_STAGE(push_kernel_launch);
+        if (sum) {
+          SumKernel(e->buf.data(), src, len, stream);
+        } else {
+          kern::DenseAssign(e->buf.data(), src, len, stream);
+        }
       }
-      if (chain) e->last_ev = MakeEventRef(po_, stream);
+      if (chain) {
+        XPS_STAGE(push_chain_event);
+        e->last_ev = MakeEventRef(po_, stream);
+      }
     } else {
       // host vals land via synchronous copies below: drain the entry's
       // outstanding cross-stream kernel first
@@ -508,7 +514,10 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
     }
     // the plane's in-place write (or the sync below) reads buf on
     // `stream`: order it behind the last writer
-    OrderAfter(touched[0], stream);
+    {
+      XPS_STAGE(pull_order_after);
+      OrderAfter(touched[0], stream);
+    }
     res.vals = SArray<float>::View(entry);  // zero-copy store view
     lens[0] = static_cast<int>(entry.size() / sizeof(float));
   } else {
