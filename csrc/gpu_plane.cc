@@ -662,6 +662,57 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
     std::lock_guard<std::mutex> lk(p->mu);
     if (p->node.shm_uid == 0) p->node = peer_node;
   }
+
+  // ---- one-sided assign push (reference rdma steady state: unsignaled
+  // RDMA_WRITE of vals + write-with-imm of meta, rdma_transport.h:341-
+  // 356): the worker writes the server's advertised store entry with
+  // ITS OWN kernel and sends the meta-only notification once the write
+  // completes. The server's push handling shrinks to an ack.
+  if (msg.meta.request && msg.meta.push && !msg.meta.pull &&
+      (msg.meta.option & kOptEntryPush) && device_ >= 0 && msg.data.size() > 1 &&
+      msg.data[1].on_device()) {
+    Postoffice* elpo = LocalPeer(p);
+    SArray<char> vals = msg.data[1];
+    char* dst = (elpo || EnsureRing(p)) ? ResolvePeer(p, msg.meta.addr, vals.size()) : nullptr;
+    if (dst) {
+      XPS_STAGE(entry_push);
+      hipStream_t stream = StreamForPeer(p->node.id);
+      XPS_HIP_CHECK(hipSetDevice(device_));
+      kern::DenseAssign(dst, vals.data(), vals.size(), stream);
+      g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
+      Message meta_msg;
+      meta_msg.meta = msg.meta;
+      meta_msg.meta.option |= kOptInPlace;
+      meta_msg.meta.val_len = static_cast<int64_t>(vals.size());
+      meta_msg.meta.data_type.clear();
+      for (size_t i = 0; i < msg.data.size(); ++i) {
+        if (i == 1 || msg.data[i].on_device()) continue;
+        meta_msg.data.push_back(msg.data[i]);
+        meta_msg.meta.data_type.push_back(msg.meta.data_type[i]);
+      }
+      int64_t bytes = 64 + static_cast<int64_t>(vals.size());
+      Message keepalive;
+      keepalive.data.push_back(vals);
+      std::string payload;
+      if (!elpo) {
+        std::vector<char> br(meta_msg.data.size(), 0);
+        XPS_CHECK(Serialize(meta_msg, br, &payload));
+      }
+      hipEvent_t ev = GetEvent();
+      XPS_HIP_CHECK(hipEventRecord(ev, stream));
+      {
+        std::lock_guard<std::mutex> lk(pend_mu_);
+        pending_[p->node.id].push_back(Pending{ev, p->node.id, std::move(payload),
+                                               std::move(meta_msg), std::move(keepalive),
+                                               bytes, elpo});
+      }
+      pending_count_.fetch_add(1);
+      p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
+      return bytes;
+    }
+    msg.meta.option &= ~kOptEntryPush;  // entry unmapped: normal path
+  }
+
   if (Postoffice* lpo = LocalPeer(p)) return SendLocal(msg, p, lpo);
   bool response = !msg.meta.request;
   auto sync_peer_lanes = [this, &peer_node] {

@@ -256,6 +256,22 @@ class KVWorker : public SimpleApp {
         msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
         msg.AddData(s.vals);
         if (!s.lens.empty()) msg.AddData(s.lens);
+        // one-sided steady state (reference rdma_van.h:486-508): a
+        // prior ACK advertised this key's store-entry offset — ask the
+        // plane to write vals there itself and send meta only. Assign
+        // semantics only (sum needs the server's accumulate kernel).
+        if (!pull && s.keys.size() == 1 && cmd != kCmdSum && s.vals.on_device()) {
+          std::lock_guard<std::mutex> lk(mu_);
+          auto nit = entry_cache_.find(msg.meta.recver);
+          if (nit != entry_cache_.end()) {
+            auto it = nit->second.find(s.keys[0]);
+            if (it != nit->second.end() &&
+                it->second.second == static_cast<int64_t>(s.vals.nbytes())) {
+              msg.meta.addr = it->second.first;
+              msg.meta.option |= kOptEntryPush;
+            }
+          }
+        }
         if (pull && dst) {
           // fused round: advertise this slice's pull destination — the
           // dst buffer shares the vals geometry, so the slice's element
@@ -318,6 +334,11 @@ class KVWorker : public SimpleApp {
         got.push_back(RecvSlice{kvs, msg.meta.option, msg.meta.val_len});
         last = static_cast<int>(got.size()) >= expected_[ts];
       } else {
+        // a push ACK may advertise the server's entry offset for this
+        // key: cache it so later assign pushes go one-sided
+        if (msg.meta.option & kOptEntryAddr) {
+          entry_cache_[msg.meta.sender][msg.meta.key] = {msg.meta.addr, msg.meta.val_len};
+        }
         int n = ++push_acks_[ts];
         last = n >= expected_[ts];
       }
@@ -418,6 +439,10 @@ class KVWorker : public SimpleApp {
     int option = 0;
     int64_t val_len = 0;
   };
+
+  // server node id -> key -> (entry pool offset, byte len) learned from
+  // push ACKs; guarded by mu_
+  std::unordered_map<int, std::unordered_map<Key, std::pair<uint64_t, int64_t>>> entry_cache_;
 
   Slicer slicer_;
   std::mutex mu_;

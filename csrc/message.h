@@ -82,6 +82,13 @@ static const int kOptHostAddr = 8;  // meta.addr is a HOST-shm-pool offset (else
 // plane must enqueue the in-place write / completion event on that lane
 // (handlers that stay on lane 0, e.g. sparse, leave this unset)
 static const int kOptPullLane = 16;
+// push ACK: meta.addr/val_len advertise the server's store-entry pool
+// offset for this key — the worker may write subsequent assign pushes
+// there ONE-SIDED (the rdma_van push_addr_ steady state, :486-508)
+static const int kOptEntryAddr = 32;
+// push REQUEST: the sender has a cached entry offset in meta.addr and
+// asks the plane to write vals there itself + deliver the meta only
+static const int kOptEntryPush = 64;
 
 enum DataType : int { kChar = 0, kInt32, kInt64, kUint64, kFloat32, kFloat64, kUint8 };
 inline size_t DataTypeSize(int t) {

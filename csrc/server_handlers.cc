@@ -225,6 +225,13 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
     HandleReducePush(req, kvs, server);
     return;
   }
+  if (req.option & kOptInPlace) {
+    // one-sided push: the worker's kernel already wrote our store entry
+    // (the notification is ordered after the write's completion event);
+    // nothing to launch — ack immediately
+    server->Response(req);
+    return;
+  }
   hipStream_t stream = Stream(req.sender);
   XPS_HIP_CHECK(hipSetDevice(HbmPool::Get()->device()));
   bool chain = NeedChain(req.sender);
@@ -274,6 +281,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   }
   size_t off = 0;  // bytes into vals
   bool synced = false;
+  Entry* last_e = nullptr;
   for (size_t i = 0; i < n; ++i) {
     size_t len = kvs.lens.empty() ? kvs.vals.nbytes() / n
                                   : static_cast<size_t>(kvs.lens[i]) * sizeof(float);
@@ -282,12 +290,17 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
       std::lock_guard<std::mutex> lk(mu_);
       e = &store_[kvs.keys[i]];
       if (e->buf.size() < len) {
+        // retire (never free) a smaller buffer: workers may hold its
+        // offset in their one-sided entry caches — a stale write must
+        // land in dead-but-owned memory, not a reused pool region
+        if (!e->buf.empty()) e->retired.push_back(e->buf);
         e->buf = HbmPool::Get()->AllocArray(len);
         // zero before publishing: another peer's stream may accumulate
         // into this entry concurrently with our first push
         XPS_HIP_CHECK(hipMemset(e->buf.data(), 0, len));
       }
     }
+    last_e = e;
     bool sum = sum_all;
     const char* src = reinterpret_cast<const char*>(kvs.vals.data()) + off;
     if (kvs.vals.on_device()) {
@@ -326,6 +339,19 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   }
   auto* plane = po_->van() ? po_->van()->plane() : nullptr;
   if (!plane && !synced) XPS_HIP_CHECK(hipStreamSynchronize(stream));
+  // assign mode, single key: advertise the store entry's pool offset so
+  // this worker's next pushes of the key go one-sided (kOptEntryPush)
+  if (mode_ == DenseMode::kAssign && n == 1 && last_e && plane) {
+    uint64_t eoff = 0;
+    if (HbmPool::Get()->OffsetOf(last_e->buf.data(), &eoff)) {
+      KVMeta r = req;
+      r.addr = eoff;
+      r.val_len = static_cast<int64_t>(last_e->buf.size());
+      r.option |= kOptEntryAddr;
+      server->Response(r);
+      return;
+    }
+  }
   server->Response(req);
 }
 
@@ -553,10 +579,18 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
     res.vals = SArray<float>::View(tmp);  // plane keeps it alive until sent
   }
   res.lens = lens;
+  KVMeta r = req;
+  // the advertised destination can be SMALLER than the store entry
+  // (entry grew since the worker sized its buffer): an in-place write
+  // would overrun the requester's pool region — force the staging path,
+  // whose worker-side merge checks the overflow loudly
+  if ((r.option & kOptPullAddr) &&
+      static_cast<int64_t>(res.vals.nbytes()) > req.val_len && req.val_len > 0) {
+    r.option &= ~kOptPullAddr;
+  }
   // TCP fallback (no in-place destination): the staging D2H copy below in
   // the van is stream-unaware — drain our stream first
-  if (!(req.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
-  KVMeta r = req;
+  if (!(r.option & kOptPullAddr)) XPS_HIP_CHECK(hipStreamSynchronize(stream));
   r.option |= kOptPullLane;  // ordering prepared on the pull lane
   server->Response(r, res);  // plane enqueues the in-place read on `stream`
   if (chain) {

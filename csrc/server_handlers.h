@@ -64,6 +64,11 @@ class GpuDenseHandler {
     // one — otherwise two senders' sum kernels race (lost updates) and
     // a pull can read a half-written store
     EventRef last_ev;
+    // superseded (smaller) buffers, kept alive forever: workers may
+    // still hold their offsets in one-sided entry caches; stale writes
+    // must hit dead-but-owned memory (they self-heal — the mismatched
+    // length makes the worker fall back and re-learn the new offset)
+    std::vector<SArray<char>> retired;
   };
 
   // Reduce-mode round state, per KEY-SET: a worker's buckets for this

@@ -212,6 +212,43 @@ def test_sparse_handler_gpu():
         _down_joint()
 
 
+def test_one_sided_assign_push_gpu():
+    """Steady-state one-sided push (rdma_van push_addr_ analog): after
+    the first ack advertises the store entry's offset, later assign
+    pushes are written by the WORKER's kernel + meta-only notification.
+    Values must round-trip across repeats, and an entry REALLOC (bigger
+    push) must invalidate the cached offset cleanly."""
+    _boot_joint_inproc()
+    try:
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(mode="assign")
+        worker = ps.KVWorker(0, 0)
+        n = 1 << 16
+        src = ps.pool_alloc(n * 4)
+        dst = ps.pool_alloc(n * 4)
+        keys = np.array([44], dtype=np.uint64)
+        lens = np.array([n], dtype=np.int32)
+        for it in range(4):  # push 1 learns the entry; 2..4 go one-sided
+            vals = np.full(n, float(it + 1), dtype=np.float32)
+            src.copy_from(vals)
+            worker.wait(worker.zpush_ptr(keys, src.ptr, n * 4, 0, lens, cmd=1))
+            worker.wait(worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens))
+            assert np.allclose(dst.to_numpy_f32(), vals), it
+        # grow the entry: stale cached offset must fall back + re-learn
+        m = n * 2
+        src2 = ps.pool_alloc(m * 4)
+        dst2 = ps.pool_alloc(m * 4)
+        lens2 = np.array([m], dtype=np.int32)
+        for it in range(3):
+            vals2 = np.full(m, 9.5 + it, dtype=np.float32)
+            src2.copy_from(vals2)
+            worker.wait(worker.zpush_ptr(keys, src2.ptr, m * 4, 0, lens2, cmd=1))
+            worker.wait(worker.zpull_ptr(keys, dst2.ptr, m * 4, 0, lens2))
+            assert np.allclose(dst2.to_numpy_f32(), vals2), it
+    finally:
+        _down_joint()
+
+
 def test_dense_bf16_accumulate_gpu():
     """End-to-end bf16 dense sum: handler in dtype=bf16 mode accumulates
     bf16 payloads exactly like torch's bf16 add (halves bytes moved on
