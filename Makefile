@@ -17,6 +17,13 @@ ifeq ($(ASAN),1)
 CXXFLAGS += -fsanitize=address -fno-omit-frame-pointer -g
 LDFLAGS += -fsanitize=address
 endif
+# `make TSAN=1 cppbench` — thread-sanitized C++ bench (beyond the
+# reference, which has no TSAN config): run build/bench_kv for a
+# race-checked end-to-end cluster
+ifeq ($(TSAN),1)
+CXXFLAGS += -fsanitize=thread -fno-omit-frame-pointer -g
+LDFLAGS += -fsanitize=thread
+endif
 ifneq ($(TORCHLIB),)
 LDFLAGS += -L$(TORCHLIB) -Wl,-rpath,$(TORCHLIB)
 endif
@@ -52,6 +59,9 @@ build/bench_kv.o: examples/cpp/bench_kv.cc csrc/*.h
 CPPBENCH_LD := -O3 --offload-arch=$(ARCH)
 ifeq ($(ASAN),1)
 CPPBENCH_LD += -fsanitize=address
+endif
+ifeq ($(TSAN),1)
+CPPBENCH_LD += -fsanitize=thread
 endif
 cppbench: $(CORE_OBJS) build/bench_kv.o
 	$(HIPCC) $(CPPBENCH_LD) build/bench_kv.o $(CORE_OBJS) \
