@@ -488,6 +488,12 @@ PYBIND11_MODULE(_core, m) {
     return reinterpret_cast<uintptr_t>(ptr);
   });
   m.def("pool_in_use", []() { return HbmPool::Get()->bytes_in_use(); });
+  // is this device pointer inside the zero-copy (bootstrap-exported)
+  // pool window? (torch-allocator tensors usually are)
+  m.def("pool_contains", [](uintptr_t p) {
+    uint64_t off = 0;
+    return HbmPool::Get()->OffsetOf(reinterpret_cast<const void*>(p), &off);
+  });
   py::class_<PoolBuffer>(m, "PoolBuffer")
       .def_property_readonly("ptr", &PoolBuffer::ptr)
       .def_property_readonly("nbytes", &PoolBuffer::nbytes)

@@ -43,6 +43,10 @@ class PSGradSync:
                 off += take
         self.bufs = []
         self.pull_bufs = []
+        # direct mode (decided on first allreduce): when .grad tensors
+        # are pool-resident (ps.use_torch_pool_allocator()), push/pull
+        # straight from grad memory — no staging buffers, no copies
+        self.direct = None
         for _, _, elems in self.buckets:
             if device >= 0:
                 self.bufs.append(ps.pool_alloc(elems * 4))
@@ -61,24 +65,43 @@ class PSGradSync:
     def _ptr(self, buf):
         return buf.ptr if self.device >= 0 else buf.ctypes.data
 
+    def _grad_view(self, b):
+        pi, off, elems = self.buckets[b]
+        return self.params[pi].grad.detach().reshape(-1)[off:off + elems]
+
     def allreduce(self):
         """Average .grad across workers, in place. Call after backward."""
         import torch
 
+        if self.direct is None:
+            # direct (zero-staging) mode needs every grad bucket inside
+            # the zero-copy pool window AND reduce-mode servers (pulls
+            # write grads in place)
+            self.direct = (self.device >= 0 and self.mode == "reduce" and
+                           all(self.ps._core.pool_contains(self._grad_view(b).data_ptr())
+                               for b in range(len(self.buckets))))
         keys = self._keys()
-        # stage grads into the push buffers
-        for b, (pi, off, elems) in enumerate(self.buckets):
-            g = self.params[pi].grad.detach().reshape(-1)[off:off + elems]
-            if self.device >= 0:
-                # pool buffer <- device grad (same GPU)
-                self.ps._core.k_dense_assign(self.bufs[b].ptr, g.data_ptr(), elems * 4)
-            else:
-                self.bufs[b][:] = g.cpu().numpy()
+        # stage grads into the push buffers (skipped in direct mode)
+        if not self.direct:
+            for b, (pi, off, elems) in enumerate(self.buckets):
+                g = self._grad_view(b)
+                if self.device >= 0:
+                    # pool buffer <- device grad (same GPU)
+                    self.ps._core.k_dense_assign(self.bufs[b].ptr, g.data_ptr(), elems * 4)
+                else:
+                    self.bufs[b][:] = g.cpu().numpy()
         tss = []
         for b, k in enumerate(keys):
             ka = np.array([k], dtype=np.uint64)
             elems = self.buckets[b][2]
             lens = np.array([elems], dtype=np.int32)
+            if self.direct:
+                gptr = self._grad_view(b).data_ptr()
+                tss.append(self.worker.zpush_ptr(ka, gptr, elems * 4, self.device, lens,
+                                                 cmd=2))
+                # the reduction is written straight back into .grad
+                tss.append(self.worker.zpull_ptr(ka, gptr, elems * 4, self.device, lens))
+                continue
             tss.append(self.worker.zpush_ptr(ka, self._ptr(self.bufs[b]), elems * 4,
                                              self.device, lens, cmd=2))
             if self.mode == "reduce":
@@ -97,11 +120,14 @@ class PSGradSync:
                                                  self.device, lens))
         for ts in tss:
             self.worker.wait(ts)
-        # write the averaged reduction back into .grad
+        # scale to the average (direct mode: the reduction already sits
+        # in .grad — the pull wrote it in place over xGMI)
         inv = 1.0 / self.num_workers
         for b, (pi, off, elems) in enumerate(self.buckets):
-            g = self.params[pi].grad.detach().reshape(-1)[off:off + elems]
-            if self.device >= 0:
+            g = self._grad_view(b)
+            if self.direct:
+                g.mul_(inv)
+            elif self.device >= 0:
                 self.ps._core.k_dense_assign(g.data_ptr(), self.pull_bufs[b].ptr, elems * 4)
                 g.mul_(inv)
             else:
