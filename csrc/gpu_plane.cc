@@ -155,8 +155,9 @@ void GpuPlane::Stop() {
           continue;
         }
         Peer* peer = GetPeer(p.peer_id);
-        bool sent = peer && EnsureRing(peer) &&
-                    peer->ring.Push(p.payload.data(), static_cast<uint32_t>(p.payload.size()));
+        auto ring = peer ? RingOf(peer) : nullptr;
+        bool sent = ring && ring->Push(p.payload.data(),
+                                       static_cast<uint32_t>(p.payload.size()));
         if (!sent) {
           // van TCP conns are still open (plane stops first): best-effort
           (void)po_->van()->SendOverTcp(p.resend, p.peer_id);
@@ -178,6 +179,24 @@ void GpuPlane::OnPeer(const Node& peer) {
   std::unique_lock<std::shared_timed_mutex> lk(peers_mu_);
   auto& p = peers_[peer.id];
   if (!p) p.reset(new Peer());
+  if (p->node.shm_uid != 0 && peer.shm_uid != 0 && p->node.shm_uid != peer.shm_uid) {
+    // RECOVERY: the id now names a NEW process. Every cached resource
+    // of the old one is stale — the ring segment has no consumer, the
+    // slab mappings point at a dead process's pool (writing through
+    // them would corrupt whatever reused that memory), and local_po may
+    // name the wrong instance. Reset; the next send re-opens/re-imports
+    // lazily. The old ring object stays alive via shared_ptr for any
+    // in-flight push.
+    std::lock_guard<std::mutex> lk2(p->mu);
+    XPS_LOG(Warning) << "peer " << peer.id << " was recovered (new process); resetting "
+                        "plane caches for it";
+    p->ring_ok.store(false, std::memory_order_release);
+    p->ring_tried = false;
+    p->ring.reset();
+    p->pool_tried = false;
+    p->slab_bases.clear();
+    p->local_po.store(nullptr, std::memory_order_release);
+  }
   p->node = peer;
 }
 
@@ -216,13 +235,21 @@ GpuPlane::Peer* GpuPlane::GetPeer(int id) {
 bool GpuPlane::EnsureRing(Peer* p) {
   if (p->ring_ok.load(std::memory_order_acquire)) return true;
   std::lock_guard<std::mutex> lk(p->mu);
-  if (p->ring.ok()) return true;
+  if (p->ring && p->ring->ok()) return true;
   if (p->ring_tried) return false;
   p->ring_tried = true;
   if (p->node.shm_uid == 0) return false;
-  bool ok = p->ring.Open(p->node.shm_uid);
-  if (ok) p->ring_ok.store(true, std::memory_order_release);
-  return ok;
+  auto ring = std::make_shared<ShmRing>();
+  if (!ring->Open(p->node.shm_uid)) return false;
+  p->ring = std::move(ring);
+  p->ring_ok.store(true, std::memory_order_release);
+  return true;
+}
+
+std::shared_ptr<ShmRing> GpuPlane::RingOf(Peer* p) {
+  if (!EnsureRing(p)) return nullptr;
+  std::lock_guard<std::mutex> lk(p->mu);
+  return p->ring;
 }
 
 bool GpuPlane::ImportPeerSlabs(Peer* p) {
@@ -759,7 +786,8 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
         XPS_CHECK(Serialize(meta_msg, {}, &payload));
       }
       int64_t bytes = static_cast<int64_t>(vals.size() + payload.size());
-      if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
+      auto ring = RingOf(p);
+      if (!ring || !ring->Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
       p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
       return bytes;
     }
@@ -849,7 +877,8 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   } else {
     // requests — and every send of a host-only plane — go out now (host
     // responses were produced synchronously; nothing to wait for)
-    if (!p->ring.Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
+    auto ring = RingOf(p);
+    if (!ring || !ring->Push(payload.data(), static_cast<uint32_t>(payload.size()))) return -1;
     XPS_VLOG(3) << "plane send done -> " << peer_node.id;
   }
   p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
@@ -923,9 +952,9 @@ void GpuPlane::CompletionLoop() {
             continue;
           }
           Peer* peer = GetPeer(front.peer_id);
-          bool sent = peer && EnsureRing(peer) &&
-                      peer->ring.Push(front.payload.data(),
-                                      static_cast<uint32_t>(front.payload.size()));
+          auto ring = peer ? RingOf(peer) : nullptr;
+          bool sent = ring && ring->Push(front.payload.data(),
+                                         static_cast<uint32_t>(front.payload.size()));
           if (sent) {
             XPS_VLOG(3) << "deferred send done -> " << front.peer_id;
           } else {

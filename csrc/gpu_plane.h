@@ -79,10 +79,13 @@ class GpuPlane : public DataPlane {
     // set when this peer's Postoffice lives in OUR process (joint
     // mode): sends bypass serialize/ring/poll entirely (direct call)
     std::atomic<Postoffice*> local_po{nullptr};
-    ShmRing ring;  // producer handle on the peer's inbound ring
+    // producer handle on the peer's inbound ring. shared_ptr so a
+    // RECOVERY (same id, new process, new shm_uid) can swap in a fresh
+    // ring while in-flight pushes keep the old mapping alive (swapping
+    // a by-value ring under a concurrent Push would munmap live memory)
+    std::shared_ptr<ShmRing> ring;
     bool ring_tried = false;
-    // fast-path flag: the ring is opened once and never closed while the
-    // plane runs, so senders read this without taking mu
+    // fast-path flag: senders read this without taking mu
     std::atomic<bool> ring_ok{false};
     std::vector<void*> slab_bases;  // peer pool slabs mapped into our space
     bool pool_tried = false;
@@ -121,6 +124,8 @@ class GpuPlane : public DataPlane {
   // thread is disabled.
   void EnqueueLocal(Message msg, int64_t bytes);
   bool EnsureRing(Peer* p);
+  // the peer's current ring (nullptr when none could be opened)
+  std::shared_ptr<ShmRing> RingOf(Peer* p);
   // import every slab of the peer's pool (idempotent)
   bool ImportPeerSlabs(Peer* p);
   // translate a peer-pool GLOBAL offset range to a mapped pointer

@@ -265,9 +265,12 @@ class KVWorker : public SimpleApp {
           auto nit = entry_cache_.find(msg.meta.recver);
           if (nit != entry_cache_.end()) {
             auto it = nit->second.find(s.keys[0]);
+            // uid guard: a RECOVERED server (same id, new process) has a
+            // fresh pool — its old offsets must never be written
             if (it != nit->second.end() &&
-                it->second.second == static_cast<int64_t>(s.vals.nbytes())) {
-              msg.meta.addr = it->second.first;
+                it->second.len == static_cast<int64_t>(s.vals.nbytes()) &&
+                po_->van()->GetNode(msg.meta.recver).shm_uid == it->second.uid) {
+              msg.meta.addr = it->second.off;
               msg.meta.option |= kOptEntryPush;
             }
           }
@@ -337,7 +340,9 @@ class KVWorker : public SimpleApp {
         // a push ACK may advertise the server's entry offset for this
         // key: cache it so later assign pushes go one-sided
         if (msg.meta.option & kOptEntryAddr) {
-          entry_cache_[msg.meta.sender][msg.meta.key] = {msg.meta.addr, msg.meta.val_len};
+          entry_cache_[msg.meta.sender][msg.meta.key] =
+              EntryRec{msg.meta.addr, msg.meta.val_len,
+                       po_->van()->GetNode(msg.meta.sender).shm_uid};
         }
         int n = ++push_acks_[ts];
         last = n >= expected_[ts];
@@ -440,9 +445,15 @@ class KVWorker : public SimpleApp {
     int64_t val_len = 0;
   };
 
-  // server node id -> key -> (entry pool offset, byte len) learned from
-  // push ACKs; guarded by mu_
-  std::unordered_map<int, std::unordered_map<Key, std::pair<uint64_t, int64_t>>> entry_cache_;
+  // server node id -> key -> advertised store-entry location, learned
+  // from push ACKs; uid pins the advert to the server PROCESS that made
+  // it (recovery = new process = new pool). Guarded by mu_.
+  struct EntryRec {
+    uint64_t off;
+    int64_t len;
+    uint64_t uid;
+  };
+  std::unordered_map<int, std::unordered_map<Key, EntryRec>> entry_cache_;
 
   Slicer slicer_;
   std::mutex mu_;
