@@ -192,7 +192,8 @@ void GpuPlane::OnPeer(const Node& peer) {
                         "plane caches for it";
     p->ring_ok.store(false, std::memory_order_release);
     p->ring_tried = false;
-    p->ring.reset();
+    std::atomic_store_explicit(&p->ring, std::shared_ptr<ShmRing>(),
+                               std::memory_order_release);
     p->pool_tried = false;
     p->slab_bases.clear();
     p->local_po.store(nullptr, std::memory_order_release);
@@ -235,21 +236,26 @@ GpuPlane::Peer* GpuPlane::GetPeer(int id) {
 bool GpuPlane::EnsureRing(Peer* p) {
   if (p->ring_ok.load(std::memory_order_acquire)) return true;
   std::lock_guard<std::mutex> lk(p->mu);
-  if (p->ring && p->ring->ok()) return true;
+  {
+    auto cur = std::atomic_load_explicit(&p->ring, std::memory_order_acquire);
+    if (cur && cur->ok()) return true;
+  }
   if (p->ring_tried) return false;
   p->ring_tried = true;
   if (p->node.shm_uid == 0) return false;
   auto ring = std::make_shared<ShmRing>();
   if (!ring->Open(p->node.shm_uid)) return false;
-  p->ring = std::move(ring);
+  std::atomic_store_explicit(&p->ring, std::shared_ptr<ShmRing>(std::move(ring)),
+                             std::memory_order_release);
   p->ring_ok.store(true, std::memory_order_release);
   return true;
 }
 
 std::shared_ptr<ShmRing> GpuPlane::RingOf(Peer* p) {
   if (!EnsureRing(p)) return nullptr;
-  std::lock_guard<std::mutex> lk(p->mu);
-  return p->ring;
+  // lock-free snapshot: the recovery path swaps the pointer with
+  // atomic_store, so the hot send path never takes p->mu
+  return std::atomic_load_explicit(&p->ring, std::memory_order_acquire);
 }
 
 bool GpuPlane::ImportPeerSlabs(Peer* p) {

@@ -622,3 +622,108 @@ def test_multiprocess_hipipc_eight_joint_on_one_gpu():
     assert len(results) == 8
     for rank, (first, last) in results.items():
         assert first == 36.0 and last == 36.0, results
+
+
+# ---- elastic recovery WITH real HBM pools (lazy hipIpc re-import) ------
+
+
+def _gpu_recovery_role(role, port, outdir, behavior):
+    import os
+    import sys
+    import time
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    sys.path.insert(0, str(repo))
+    os.environ.update({
+        "DMLC_NUM_WORKER": "1",
+        "DMLC_NUM_SERVER": "1",
+        "DMLC_PS_ROOT_URI": "127.0.0.1",
+        "DMLC_PS_ROOT_PORT": str(port),
+        "PS_HEARTBEAT_INTERVAL": "1",
+        "PS_HEARTBEAT_TIMEOUT": "2",
+        "XPS_POOL_GB": "2",
+    })
+    import numpy as np
+    import ps_lite_amd as psm
+
+    def report(name, text):
+        with open(os.path.join(outdir, name), "w") as f:
+            f.write(text)
+
+    if role == "scheduler":
+        psm.start(role=role, device=-1)
+        report("scheduler", "up")
+        time.sleep(60)
+        os._exit(0)
+    elif role == "server":
+        psm.start(role=role, device=0)
+        server = psm.KVServer(0)
+        server.set_gpu_dense_handle(mode="sum")
+        report("server", "up")
+        time.sleep(60)
+        os._exit(0)
+    elif behavior == "die":
+        psm.start(role="worker", device=0)
+        report("worker1", "dying")
+        time.sleep(1)
+        os._exit(0)  # dead node; its pool/ring die with it
+    else:  # replacement worker: must re-import the server lazily
+        psm.start(role="worker", device=0)
+        n = 1 << 14
+        src = psm.pool_alloc(n * 4)
+        dst = psm.pool_alloc(n * 4)
+        src.copy_from(np.full(n, 3.5, dtype=np.float32))
+        keys = np.array([7], dtype=np.uint64)
+        lens = np.array([n], dtype=np.int32)
+        w = psm.KVWorker(0, 0)
+        w.wait(w.zpush_ptr(keys, src.ptr, n * 4, 0, lens, cmd=2))
+        w.wait(w.zpull_ptr(keys, dst.ptr, n * 4, 0, lens))
+        out = dst.to_numpy_f32()
+        ok = bool(np.allclose(out, 3.5))
+        zc = psm._core.zero_copy_recv_count()
+        report("worker2", f"{'ok' if ok else 'bad'}:zc={zc}")
+        time.sleep(1)
+        os._exit(0)
+
+
+def test_gpu_worker_recovery_with_pools():
+    """A replacement worker (new process, new pool) joins after the
+    original died: the server's plane must reset its cached mappings for
+    the id and lazily re-import the NEW pool (hipIpc open outside the
+    bootstrap ladder) — pushes and in-place pulls must work zero-copy."""
+    import multiprocessing as mp
+    import random
+    import tempfile
+    import time as _t
+
+    port = random.randint(21000, 50000)
+    outdir = tempfile.mkdtemp(prefix="xps_gpu_recovery_")
+    ctx = mp.get_context("spawn")
+    procs = []
+    for role, behavior in (("scheduler", None), ("server", None), ("worker", "die")):
+        p = ctx.Process(target=_gpu_recovery_role, args=(role, port, outdir, behavior),
+                        daemon=True)
+        p.start()
+        procs.append(p)
+    deadline = _t.time() + 60
+    import os as _os
+    while not _os.path.exists(_os.path.join(outdir, "worker1")) and _t.time() < deadline:
+        _t.sleep(0.5)
+    _t.sleep(4)  # exceed the heartbeat timeout
+    p = ctx.Process(target=_gpu_recovery_role, args=("worker", port, outdir, "recover"),
+                    daemon=True)
+    p.start()
+    procs.append(p)
+    verdict = None
+    wfile = _os.path.join(outdir, "worker2")
+    deadline = _t.time() + 90
+    while _t.time() < deadline:
+        if _os.path.exists(wfile):
+            _t.sleep(0.3)
+            verdict = open(wfile).read()
+            break
+        _t.sleep(0.5)
+    assert verdict is not None and verdict.startswith("ok"), verdict
+    for p in procs:
+        p.terminate()
