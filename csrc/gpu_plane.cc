@@ -559,6 +559,30 @@ void GpuPlane::DeliverLocal(Postoffice* lpo, Message& msg, int64_t bytes) {
   }
 }
 
+bool GpuPlane::LocalPullRead(int peer_id, void* dst, uint64_t entry_off, size_t len,
+                             Message resp) {
+  if (device_ < 0 || !started_ || stop_.load()) return false;
+  Peer* p = GetPeer(peer_id);
+  char* src = ResolvePeer(p, entry_off, len);
+  if (!src) return false;
+  XPS_STAGE(local_pull_read);
+  hipStream_t stream = StreamForPeer(peer_id);  // lane 0: ordered after entry pushes
+  XPS_HIP_CHECK(hipSetDevice(device_));
+  kern::DenseAssign(dst, src, len, stream);
+  g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
+  hipEvent_t ev = GetEvent();
+  XPS_HIP_CHECK(hipEventRecord(ev, stream));
+  int64_t bytes = static_cast<int64_t>(len) + 64;
+  {
+    std::lock_guard<std::mutex> lk(pend_mu_);
+    pending_[peer_id].push_back(Pending{ev, peer_id, std::string(), std::move(resp), Message(),
+                                        bytes, po_});
+  }
+  pending_count_.fetch_add(1);
+  p->rx_bytes.fetch_add(bytes, std::memory_order_relaxed);
+  return true;
+}
+
 void GpuPlane::EnqueueLocal(Message msg, int64_t bytes) {
   {
     std::lock_guard<std::mutex> lk(local_mu_);

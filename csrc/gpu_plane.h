@@ -64,6 +64,14 @@ class GpuPlane : public DataPlane {
   // its own xGMI link)
   std::vector<std::tuple<int, int64_t, int64_t>> PeerBytes();
   int device() const { return device_; }
+
+  // One-sided pull (the RDMA_READ analog): copy `len` bytes from the
+  // peer's advertised store entry straight into `dst` with OUR kernel
+  // on the peer's lane-0 stream (ordered after our one-sided pushes of
+  // the same key), then deliver the synthetic in-place response to
+  // OURSELVES once the copy completes. Returns false when the peer's
+  // pool is not mapped (caller sends a normal pull request).
+  bool LocalPullRead(int peer_id, void* dst, uint64_t entry_off, size_t len, Message resp);
   // pooled events (shared with the server handlers)
   hipEvent_t GetEvent();
   void PutEvent(hipEvent_t ev);

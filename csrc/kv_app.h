@@ -21,6 +21,7 @@
 #include <unordered_map>
 #include <vector>
 
+#include "gpu_plane.h"
 #include "hip_pool.h"
 #include "hip_util.h"
 #include "host_par.h"
@@ -260,18 +261,30 @@ class KVWorker : public SimpleApp {
         // prior ACK advertised this key's store-entry offset — ask the
         // plane to write vals there itself and send meta only. Assign
         // semantics only (sum needs the server's accumulate kernel).
-        if (!pull && s.keys.size() == 1 && cmd != kCmdSum && s.vals.on_device()) {
+        {
           std::lock_guard<std::mutex> lk(mu_);
           auto nit = entry_cache_.find(msg.meta.recver);
-          if (nit != entry_cache_.end()) {
-            auto it = nit->second.find(s.keys[0]);
-            // uid guard: a RECOVERED server (same id, new process) has a
-            // fresh pool — its old offsets must never be written
-            if (it != nit->second.end() &&
-                it->second.len == static_cast<int64_t>(s.vals.nbytes()) &&
-                po_->van()->GetNode(msg.meta.recver).shm_uid == it->second.uid) {
-              msg.meta.addr = it->second.off;
-              msg.meta.option |= kOptEntryPush;
+          if (nit != entry_cache_.end() && !nit->second.empty()) {
+            if (!pull && s.keys.size() == 1 && cmd != kCmdSum && s.vals.on_device()) {
+              auto it = nit->second.find(s.keys[0]);
+              // uid guard: a RECOVERED server (same id, new process) has
+              // a fresh pool — its old offsets must never be written
+              if (it != nit->second.end() &&
+                  it->second.len == static_cast<int64_t>(s.vals.nbytes()) &&
+                  po_->van()->GetNode(msg.meta.recver).shm_uid == it->second.uid) {
+                msg.meta.addr = it->second.off;
+                msg.meta.option |= kOptEntryPush;
+                it->second.one_sided = true;
+              } else if (it != nit->second.end()) {
+                it->second.one_sided = false;  // normal push re-orders via server
+              }
+            } else {
+              // normal / multi-key push: any contained cached key loses
+              // its one-sided ordering chain
+              for (size_t ki = 0; ki < s.keys.size(); ++ki) {
+                auto it = nit->second.find(s.keys[ki]);
+                if (it != nit->second.end()) it->second.one_sided = false;
+              }
             }
           }
         }
@@ -293,6 +306,53 @@ class KVWorker : public SimpleApp {
           }
         }
       } else {
+        // One-sided pull (RDMA_READ analog, assign-mode steady state):
+        // the entry offset is cached, OUR last push of the key was
+        // one-sided (stream-ordered before this read), and the process
+        // uid still matches — copy the server's entry into the dst with
+        // our own kernel and complete the request with a synthetic
+        // deferred response. No server round trip at all. Cross-worker
+        // concurrent writes follow the reference's async-PS semantics
+        // (an RDMA_READ racing another worker's RDMA_WRITE).
+        if (s.keys.size() == 1 && cmd != kCmdSum && s.vals.on_device() &&
+            s.vals.nbytes() > 0) {
+          uint64_t eoff = 0;
+          bool hit = false;
+          {
+            std::lock_guard<std::mutex> lk(mu_);
+            auto nit = entry_cache_.find(msg.meta.recver);
+            if (nit != entry_cache_.end()) {
+              auto it = nit->second.find(s.keys[0]);
+              if (it != nit->second.end() && it->second.one_sided &&
+                  it->second.len == static_cast<int64_t>(s.vals.nbytes()) &&
+                  po_->van()->GetNode(msg.meta.recver).shm_uid == it->second.uid) {
+                eoff = it->second.off;
+                hit = true;
+              }
+            }
+          }
+          if (hit) {
+            auto* plane = dynamic_cast<GpuPlane*>(po_->van()->plane());
+            if (plane) {
+              Message resp;
+              resp.meta.app_id = obj_->app_id();
+              resp.meta.customer_id = obj_->customer_id();
+              resp.meta.request = false;
+              resp.meta.pull = true;
+              resp.meta.head = cmd;
+              resp.meta.timestamp = ts;
+              resp.meta.sender = msg.meta.recver;
+              resp.meta.recver = po_->node_id();
+              resp.meta.key = s.keys[0];
+              resp.meta.option = kOptInPlace;
+              resp.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
+              if (plane->LocalPullRead(msg.meta.recver, s.vals.data(), eoff,
+                                       s.vals.nbytes(), std::move(resp))) {
+                continue;  // request satisfied by the local read
+              }
+            }
+          }
+        }
         // pull request: keys (+lens geometry) only; advertise destination
         msg.meta.val_len = static_cast<int64_t>(s.vals.nbytes());
         if (!s.vals.empty()) {
@@ -452,6 +512,10 @@ class KVWorker : public SimpleApp {
     uint64_t off;
     int64_t len;
     uint64_t uid;
+    // our LAST push of this key was one-sided (same-stream): only then
+    // may a pull be served by a one-sided read — the stream orders the
+    // read after our write. A normal (server-kernel) push clears it.
+    bool one_sided = false;
   };
   std::unordered_map<int, std::unordered_map<Key, EntryRec>> entry_cache_;
 

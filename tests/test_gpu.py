@@ -483,6 +483,49 @@ def _gpu_worker_fn(ps_mod, rank):
     return (float(out[0]), float(out[-1])), server
 
 
+def _one_sided_steady_worker_fn(ps_mod, rank):
+    """Cross-process steady state: after round 1, assign pushes write
+    the server's entry one-sided and pulls are one-sided READS of it
+    (no server round trip). Per-rank distinct keys = single writer, so
+    every pull must return exactly that worker's last push."""
+    server = ps_mod.KVServer(0)
+    server.set_gpu_dense_handle(mode="assign")
+    ps_mod.barrier("worker", ps_mod.WORKER_GROUP)
+    worker = ps_mod.KVWorker(0, 0)
+    n = 1 << 14
+    src = ps_mod.pool_alloc(n * 4)
+    dst = ps_mod.pool_alloc(n * 4)
+    half = (1 << 64) // 2
+    # one key on each server, owned by this rank
+    keys = np.array(sorted([100 + rank, half + 200 + rank]), dtype=np.uint64)
+    lens = np.full(2, n // 2, dtype=np.int32)
+    before = ps_mod._core.zero_copy_recv_count()
+    for step in range(5):
+        vals = np.full(n, float(10 * (rank + 1) + step), dtype=np.float32)
+        src.copy_from(vals)
+        for i in range(2):  # single-key messages (entry-cache path)
+            ka = keys[i:i + 1]
+            la = lens[i:i + 1]
+            worker.wait(worker.zpush_ptr(ka, src.ptr + i * (n // 2) * 4, (n // 2) * 4, 0,
+                                         la, cmd=1))
+        for i in range(2):
+            ka = keys[i:i + 1]
+            la = lens[i:i + 1]
+            worker.wait(worker.zpull_ptr(ka, dst.ptr + i * (n // 2) * 4, (n // 2) * 4, 0,
+                                         la, cmd=1))
+        out = dst.to_numpy_f32()
+        assert np.allclose(out, float(10 * (rank + 1) + step)), (step, out[:3])
+    zc = ps_mod._core.zero_copy_recv_count() - before
+    return int(zc), server
+
+
+def test_one_sided_steady_state_two_procs():
+    results = launch_local(2, 2, _one_sided_steady_worker_fn, joint=True, devices={0: 0, 1: 0},
+                           env_extra={"XPS_POOL_GB": 4}, timeout=300)
+    for rank, zc in results.items():
+        assert zc > 0, results
+
+
 def _reduce_worker_fn(ps_mod, rank):
     server = ps_mod.KVServer(0)
     server.set_gpu_dense_handle(mode="reduce")
