@@ -80,6 +80,22 @@ int main(int argc, char** argv) {
   double gb = 2.0 * msg_bytes * num_keys * iters / 1e9;
   printf("push+pull: %.2f GB/s (%.3f ms/round)\n", gb / sec, sec / iters * 1e3);
 
+  // fused ZPushPull (one trip applies the push and returns the post-push
+  // values): with the default sum handle, round r must return r * vals
+  {
+    SArray<Key> fk({static_cast<Key>(1000)});
+    auto fin = SArray<float>::View(HostShmPool::Get()->AllocArray(msg_bytes));
+    auto fout = SArray<float>::View(HostShmPool::Get()->AllocArray(msg_bytes));
+    for (size_t i = 0; i < n; ++i) fin[i] = 2.0f;
+    for (int r = 1; r <= 3; ++r) {
+      worker.Wait(worker.ZPushPull(fk, fin, &fout, lens, kCmdSum));
+      for (size_t i = 0; i < n; i += n / 5 + 1) {
+        XPS_CHECK_EQ(fout[i], 2.0f * r) << "fused round mismatch at r=" << r;
+      }
+    }
+    printf("fused ZPushPull OK (3 rounds)\n");
+  }
+
   std::thread sfin([] { Finalize(0, "server", true); });
   std::thread schedfin([] { Finalize(0, "scheduler", true); });
   Finalize(0, "worker", true);
