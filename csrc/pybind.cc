@@ -234,6 +234,13 @@ class PyKVServer {
     });
   }
 
+  // RegisterRecvBufferWithRank parity: the app's own device buffer
+  // becomes the dense store entry for `key`
+  void RegisterEntry(uint64_t key, uintptr_t ptr, size_t nbytes, int device) {
+    XPS_CHECK(dense_) << "register_entry needs set_gpu_dense_handle first";
+    dense_->RegisterEntry(key, reinterpret_cast<void*>(ptr), nbytes, device);
+  }
+
   // checkpoint/resume of the installed handler's server state
   void SaveCheckpoint(const std::string& path) {
     py::gil_scoped_release rel;
@@ -543,6 +550,8 @@ PYBIND11_MODULE(_core, m) {
            py::arg("dtype") = "f32")
       .def("set_gpu_sparse_handle", &PyKVServer::SetGpuSparseHandle, py::arg("rows"),
            py::arg("row_len"), py::arg("accumulate") = true, py::arg("key_shift") = 0)
+      .def("register_entry", &PyKVServer::RegisterEntry, py::arg("key"), py::arg("ptr"),
+           py::arg("nbytes"), py::arg("device") = 0)
       .def("sparse_table_ptr", &PyKVServer::SparseTablePtr)
       .def("save_checkpoint", &PyKVServer::SaveCheckpoint)
       .def("load_checkpoint", &PyKVServer::LoadCheckpoint)

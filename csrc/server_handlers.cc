@@ -599,6 +599,15 @@ void GpuDenseHandler::HandlePull(const KVMeta& req, const KVPairs<float>& kvs,
   }
 }
 
+void GpuDenseHandler::RegisterEntry(Key key, void* ptr, size_t nbytes, int device) {
+  XPS_CHECK(ptr && nbytes) << "RegisterEntry needs a real buffer";
+  std::lock_guard<std::mutex> lk(mu_);
+  Entry& e = store_[key];
+  if (!e.buf.empty()) e.retired.push_back(e.buf);  // keep old offsets owned
+  // non-owning view: the app keeps the buffer alive
+  e.buf = SArray<char>(static_cast<char*>(ptr), nbytes, device);
+}
+
 void GpuDenseHandler::Save(const std::string& path) {
   FILE* f = fopen(path.c_str(), "wb");
   XPS_CHECK(f) << "cannot open checkpoint " << path;

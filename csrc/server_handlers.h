@@ -55,6 +55,13 @@ class GpuDenseHandler {
   void Save(const std::string& path);
   void Load(const std::string& path);
 
+  // RegisterRecvBufferWithRank parity (reference kv_app.h:488): make an
+  // APP-owned device buffer this key's store entry — pushes land
+  // straight in it (zero-copy into user memory), pulls serve from it.
+  // The buffer must outlive the handler; if it is pool-resident (e.g.
+  // a torch-allocator tensor) the one-sided steady state applies too.
+  void RegisterEntry(Key key, void* ptr, size_t nbytes, int device);
+
  private:
   struct Entry {
     SArray<char> buf;

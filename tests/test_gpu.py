@@ -249,6 +249,37 @@ def test_one_sided_assign_push_gpu():
         _down_joint()
 
 
+def test_register_entry_recv_buffer_gpu():
+    """RegisterRecvBufferWithRank parity (reference kv_app.h:488 + the
+    EmptyHandler pointer-equality check, test_benchmark.cc:169-181): the
+    app's OWN tensor is the store entry — a push must land in that exact
+    memory with no pull needed."""
+    _boot_joint_inproc()
+    try:
+        n = 1 << 14
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(mode="assign")
+        model = ps.pool_alloc(n * 4)  # the app's own buffer
+        model.copy_from(np.zeros(n, dtype=np.float32))
+        server.register_entry(5050, model.ptr, n * 4, 0)
+        worker = ps.KVWorker(0, 0)
+        src = ps.pool_alloc(n * 4)
+        vals = np.full(n, 4.25, dtype=np.float32)
+        src.copy_from(vals)
+        keys = np.array([5050], dtype=np.uint64)
+        lens = np.array([n], dtype=np.int32)
+        worker.wait(worker.zpush_ptr(keys, src.ptr, n * 4, 0, lens, cmd=1))
+        ps.device_sync(0)
+        # no pull: the push itself must have written the registered buffer
+        assert np.allclose(model.to_numpy_f32(), vals)
+        # pulls serve from the same buffer
+        dst = ps.pool_alloc(n * 4)
+        worker.wait(worker.zpull_ptr(keys, dst.ptr, n * 4, 0, lens))
+        assert np.allclose(dst.to_numpy_f32(), vals)
+    finally:
+        _down_joint()
+
+
 def test_dense_bf16_accumulate_gpu():
     """End-to-end bf16 dense sum: handler in dtype=bf16 mode accumulates
     bf16 payloads exactly like torch's bf16 add (halves bytes moved on
