@@ -692,6 +692,8 @@ struct KVServerReduceHandle {
   explicit KVServerReduceHandle(int num_workers) : num_workers_(std::max(1, num_workers)) {}
 
   void operator()(const KVMeta& req, const KVPairs<V>& kvs, KVServer<V>* server) {
+    XPS_CHECK(!(req.push && req.pull))
+        << "ZPushPull is not supported in reduce mode (pulls are held per round)";
     if (req.push) {
       HandlePush(req, kvs, server);
     } else if (req.pull) {

@@ -109,6 +109,16 @@ hipStream_t GpuDenseHandler::PullStream(int sender) {
 
 void GpuDenseHandler::operator()(const KVMeta& req, const KVPairs<float>& kvs,
                                  KVServer<float>* server) {
+  if (req.push && req.pull) {
+    // fused ZPushPull: apply the push silently, then answer with the
+    // post-push values (same stream -> ordered). Reduce mode holds
+    // pulls for round accounting and cannot fuse them into one message.
+    XPS_CHECK(mode_ != DenseMode::kReduce)
+        << "ZPushPull is not supported in reduce mode (pulls are held per round)";
+    HandlePush(req, kvs, server, /*respond=*/false);
+    HandlePull(req, kvs, server);
+    return;
+  }
   if (req.push) {
     HandlePush(req, kvs, server);
   } else if (req.pull) {
@@ -217,7 +227,7 @@ void GpuDenseHandler::HandleReducePush(const KVMeta& req, const KVPairs<float>& 
 }
 
 void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
-                                 KVServer<float>* server) {
+                                 KVServer<float>* server, bool respond) {
   XPS_STAGE(dense_push);
   size_t n = kvs.keys.size();
   XPS_CHECK_GT(n, 0u);
@@ -229,7 +239,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
     // one-sided push: the worker's kernel already wrote our store entry
     // (the notification is ordered after the write's completion event);
     // nothing to launch — ack immediately
-    server->Response(req);
+    if (respond) server->Response(req);
     return;
   }
   hipStream_t stream = Stream(req.sender);
@@ -275,7 +285,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
       }
       auto* plane0 = po_->van() ? po_->van()->plane() : nullptr;
       if (!plane0) XPS_HIP_CHECK(hipStreamSynchronize(stream));
-      server->Response(req);
+      if (respond) server->Response(req);
       return;
     }
   }
@@ -339,6 +349,7 @@ void GpuDenseHandler::HandlePush(const KVMeta& req, const KVPairs<float>& kvs,
   }
   auto* plane = po_->van() ? po_->van()->plane() : nullptr;
   if (!plane && !synced) XPS_HIP_CHECK(hipStreamSynchronize(stream));
+  if (!respond) return;
   // assign mode, single key: advertise the store entry's pool offset so
   // this worker's next pushes of the key go one-sided (kOptEntryPush)
   if (mode_ == DenseMode::kAssign && n == 1 && last_e && plane) {

@@ -104,7 +104,8 @@ class GpuDenseHandler {
     EventRef last_ev;                    // last kernel touching the group
   };
 
-  void HandlePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
+  void HandlePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server,
+                  bool respond = true);
   void HandleReducePush(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   void HandlePull(const KVMeta& req, const KVPairs<float>& kvs, KVServer<float>* server);
   void RespondPull(const KVMeta& req, Group* g, KVServer<float>* server);

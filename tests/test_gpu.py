@@ -311,6 +311,31 @@ def test_dense_bf16_accumulate_gpu():
         _down_joint()
 
 
+def test_dense_fused_round_gpu():
+    """Fused ZPushPull on the DENSE GPU handler (sum mode): one trip
+    applies the push and returns the accumulated values, multi-key."""
+    _boot_joint_inproc()
+    try:
+        server = ps.KVServer(0)
+        server.set_gpu_dense_handle(mode="sum")
+        worker = ps.KVWorker(0, 0)
+        nk, n = 3, 4096
+        src = ps.pool_alloc(nk * n * 4)
+        dst = ps.pool_alloc(nk * n * 4)
+        vals = np.concatenate([np.full(n, float(i + 1), dtype=np.float32)
+                               for i in range(nk)])
+        src.copy_from(vals)
+        keys = np.array([70, 71, 72], dtype=np.uint64)
+        lens = np.full(nk, n, dtype=np.int32)
+        for r in range(1, 4):  # round r returns r * vals
+            worker.wait(worker.zpushpull_ptr(keys, src.ptr, dst.ptr, nk * n * 4, 0,
+                                             lens, cmd=2))
+            out = dst.to_numpy_f32()
+            assert np.allclose(out, r * vals), (r, out[::n])
+    finally:
+        _down_joint()
+
+
 def test_sparse_fused_round_gpu():
     """Fused ZPushPull on the sparse handler: scatter-add + gather in ONE
     trip; the response must carry the post-update rows."""
