@@ -9,6 +9,7 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <cstdlib>
 
 #include "kernels.h"
 
@@ -221,6 +222,60 @@ struct DescArray {
   int n;
 };
 
+// balanced variant: segments are concatenated into one virtual chunk
+// space (prefix = exclusive 16-B-chunk offsets); each BLOCK owns
+// contiguous tiles of that space, so work splits evenly across mixed
+// segment sizes (the per-segment grid-stride left small buckets on a
+// handful of blocks while every block swept all 64 segment headers).
+// Within a tile lanes stay consecutive (coalesced); the tile's segment
+// is found once and walked forward.
+struct BalancedDescArray {
+  CopyDesc d[kMaxBatch];
+  unsigned long long prefix[kMaxBatch + 1];  // chunk offsets, prefix[n] = total
+  int n;
+};
+
+template <bool kSum>
+__global__ void batched_balanced_kernel(BalancedDescArray da, size_t chunks_per_block) {
+  size_t total = da.prefix[da.n];
+  size_t tile_begin = blockIdx.x * chunks_per_block;
+  size_t tile_end = tile_begin + chunks_per_block;
+  if (tile_end > total) tile_end = total;
+  if (tile_begin >= total) return;
+  // locate the first segment of this tile (binary search once)
+  int seg = 0;
+  {
+    int lo = 0, hi = da.n - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (da.prefix[mid] <= tile_begin) {
+        lo = mid;
+      } else {
+        hi = mid - 1;
+      }
+    }
+    seg = lo;
+  }
+  for (size_t g = tile_begin + threadIdx.x; g < tile_end; g += blockDim.x) {
+    while (g >= da.prefix[seg + 1]) ++seg;  // walk forward (tiles are contiguous)
+    size_t local = g - da.prefix[seg];
+    if (kSum) {
+      float4* dst = reinterpret_cast<float4*>(da.d[seg].dst) + local;
+      const float4* src = reinterpret_cast<const float4*>(da.d[seg].src) + local;
+      float4 a = *dst;
+      float4 b = *src;
+      a.x += b.x;
+      a.y += b.y;
+      a.z += b.z;
+      a.w += b.w;
+      *dst = a;
+    } else {
+      reinterpret_cast<uint4*>(da.d[seg].dst)[local] =
+          reinterpret_cast<const uint4*>(da.d[seg].src)[local];
+    }
+  }
+}
+
 // every block strides over every segment's 16B chunks (grid sized for
 // the concatenated total, so all segments together fill the 8 XCDs)
 // (segment loop per block; fine for <= kMaxBatch segments)
@@ -275,7 +330,46 @@ __global__ void batched_sum_kernel_f32(DescArray da) {
 
 }  // namespace
 
+namespace {
+
+bool BalancedBatchEnabled() {
+  static const bool on = [] {
+    const char* v = getenv("XPS_BALANCED_BATCH");
+    return !v || atoi(v) != 0;  // default on
+  }();
+  return on;
+}
+
+template <bool kSum>
+void LaunchBalanced(const CopyDesc* descs_host, int n, hipStream_t s) {
+  for (int off = 0; off < n; off += kMaxBatch) {
+    BalancedDescArray da{};
+    da.n = std::min(n - off, kMaxBatch);
+    unsigned long long acc = 0;
+    for (int i = 0; i < da.n; ++i) {
+      da.d[i] = descs_host[off + i];
+      da.prefix[i] = acc;
+      acc += da.d[i].nbytes / 16;
+    }
+    da.prefix[da.n] = acc;
+    if (acc == 0) continue;
+    // tile size: fill ~8192 blocks (8 XCDs want >> 256 workgroups),
+    // whole multiples of the block so lanes sweep full strides
+    size_t cpb = (acc + 8191) / 8192;
+    cpb = ((cpb + kBlock - 1) / kBlock) * kBlock;
+    int grid = static_cast<int>((acc + cpb - 1) / cpb);
+    hipLaunchKernelGGL((batched_balanced_kernel<kSum>), dim3(grid), dim3(kBlock), 0, s, da,
+                       cpb);
+  }
+}
+
+}  // namespace
+
 void BatchedAssign(const CopyDesc* descs_host, int n, hipStream_t s) {
+  if (BalancedBatchEnabled()) {
+    LaunchBalanced<false>(descs_host, n, s);
+    return;
+  }
   size_t total = 0;
   for (int i = 0; i < n; ++i) total += descs_host[i].nbytes;
   DescArray da{};
@@ -287,6 +381,10 @@ void BatchedAssign(const CopyDesc* descs_host, int n, hipStream_t s) {
 }
 
 void BatchedSumF32(const CopyDesc* descs_host, int n, hipStream_t s) {
+  if (BalancedBatchEnabled()) {
+    LaunchBalanced<true>(descs_host, n, s);
+    return;
+  }
   size_t total = 0;
   for (int i = 0; i < n; ++i) total += descs_host[i].nbytes;
   DescArray da{};
