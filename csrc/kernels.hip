@@ -235,7 +235,8 @@ struct BalancedDescArray {
   int n;
 };
 
-template <bool kSum>
+// kOp: 0 = byte assign, 1 = fp32 sum, 2 = bf16 sum (fp32 accumulate)
+template <int kOp>
 __global__ void batched_balanced_kernel(BalancedDescArray da, size_t chunks_per_block) {
   size_t total = da.prefix[da.n];
   size_t tile_begin = blockIdx.x * chunks_per_block;
@@ -259,7 +260,7 @@ __global__ void batched_balanced_kernel(BalancedDescArray da, size_t chunks_per_
   for (size_t g = tile_begin + threadIdx.x; g < tile_end; g += blockDim.x) {
     while (g >= da.prefix[seg + 1]) ++seg;  // walk forward (tiles are contiguous)
     size_t local = g - da.prefix[seg];
-    if (kSum) {
+    if (kOp == 1) {
       float4* dst = reinterpret_cast<float4*>(da.d[seg].dst) + local;
       const float4* src = reinterpret_cast<const float4*>(da.d[seg].src) + local;
       float4 a = *dst;
@@ -268,6 +269,16 @@ __global__ void batched_balanced_kernel(BalancedDescArray da, size_t chunks_per_
       a.y += b.y;
       a.z += b.z;
       a.w += b.w;
+      *dst = a;
+    } else if (kOp == 2) {
+      uint4* dst = reinterpret_cast<uint4*>(da.d[seg].dst) + local;
+      const uint4* src = reinterpret_cast<const uint4*>(da.d[seg].src) + local;
+      uint4 a = *dst;
+      uint4 b = *src;
+      a.x = bf16x2_sum(a.x, b.x);
+      a.y = bf16x2_sum(a.y, b.y);
+      a.z = bf16x2_sum(a.z, b.z);
+      a.w = bf16x2_sum(a.w, b.w);
       *dst = a;
     } else {
       reinterpret_cast<uint4*>(da.d[seg].dst)[local] =
@@ -340,7 +351,7 @@ bool BalancedBatchEnabled() {
   return on;
 }
 
-template <bool kSum>
+template <int kOp>
 void LaunchBalanced(const CopyDesc* descs_host, int n, hipStream_t s) {
   for (int off = 0; off < n; off += kMaxBatch) {
     BalancedDescArray da{};
@@ -358,7 +369,7 @@ void LaunchBalanced(const CopyDesc* descs_host, int n, hipStream_t s) {
     size_t cpb = (acc + 8191) / 8192;
     cpb = ((cpb + kBlock - 1) / kBlock) * kBlock;
     int grid = static_cast<int>((acc + cpb - 1) / cpb);
-    hipLaunchKernelGGL((batched_balanced_kernel<kSum>), dim3(grid), dim3(kBlock), 0, s, da,
+    hipLaunchKernelGGL((batched_balanced_kernel<kOp>), dim3(grid), dim3(kBlock), 0, s, da,
                        cpb);
   }
 }
@@ -367,7 +378,7 @@ void LaunchBalanced(const CopyDesc* descs_host, int n, hipStream_t s) {
 
 void BatchedAssign(const CopyDesc* descs_host, int n, hipStream_t s) {
   if (BalancedBatchEnabled()) {
-    LaunchBalanced<false>(descs_host, n, s);
+    LaunchBalanced<0>(descs_host, n, s);
     return;
   }
   size_t total = 0;
@@ -382,7 +393,7 @@ void BatchedAssign(const CopyDesc* descs_host, int n, hipStream_t s) {
 
 void BatchedSumF32(const CopyDesc* descs_host, int n, hipStream_t s) {
   if (BalancedBatchEnabled()) {
-    LaunchBalanced<true>(descs_host, n, s);
+    LaunchBalanced<1>(descs_host, n, s);
     return;
   }
   size_t total = 0;
@@ -430,6 +441,10 @@ void DenseSumBf16(uint16_t* dst, const uint16_t* src, size_t n, hipStream_t s) {
 }
 
 void BatchedSumBf16(const CopyDesc* descs_host, int n, hipStream_t s) {
+  if (BalancedBatchEnabled()) {
+    LaunchBalanced<2>(descs_host, n, s);
+    return;
+  }
   size_t total = 0;
   for (int i = 0; i < n; ++i) total += descs_host[i].nbytes;
   DescArray da{};
