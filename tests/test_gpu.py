@@ -307,6 +307,23 @@ def test_dense_bf16_accumulate_gpu():
         got = torch.from_numpy(
             dst.to_numpy_f32().view(np.uint16)[:n].copy()).view(torch.bfloat16)
         assert torch.equal(got, ref)
+
+        # multi-key message: the BALANCED batched bf16 sum kernel
+        nk = 3
+        msrc = ps.pool_alloc(nk * n * 2)
+        mdst = ps.pool_alloc(nk * n * 2)
+        mvals = torch.randn(nk * n, dtype=torch.bfloat16)
+        msrc.copy_from(mvals.view(torch.uint16).numpy())
+        mkeys = np.array([201, 202, 203], dtype=np.uint64)
+        mlens = np.full(nk, n // 2, dtype=np.int32)
+        mref = torch.zeros(nk * n, dtype=torch.bfloat16)
+        for _ in range(2):
+            worker.wait(worker.zpush_ptr(mkeys, msrc.ptr, nk * n * 2, 0, mlens, cmd=2))
+            mref = mref + mvals
+        worker.wait(worker.zpull_ptr(mkeys, mdst.ptr, nk * n * 2, 0, mlens))
+        mgot = torch.from_numpy(
+            mdst.to_numpy_f32().view(np.uint16)[:nk * n].copy()).view(torch.bfloat16)
+        assert torch.equal(mgot, mref)
     finally:
         _down_joint()
 
