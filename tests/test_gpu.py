@@ -843,3 +843,28 @@ def test_gpu_worker_recovery_with_pools():
     assert verdict is not None and verdict.startswith("ok"), verdict
     for p in procs:
         p.terminate()
+
+
+def test_pool_growth_local_slabs():
+    """Exhausting the bootstrap pool must GROW it with local-only slabs
+    (allocation succeeds; 288 GB HBM has room) — grown memory is outside
+    the advertised zero-copy window (pool_contains False), so traffic
+    from it stages instead of corrupting peer mappings."""
+    ps.pool_init(0)  # idempotent: reuses the suite's pool
+    normal = ps.pool_alloc(1 << 20)
+    assert ps._core.pool_contains(normal.ptr)
+    # drain the remaining exported capacity, then one more forces growth
+    held = []
+    grown = None
+    for _ in range(64):  # pool is at most 4 GB in this suite
+        b = ps.pool_alloc(512 << 20)
+        held.append(b)
+        if not ps._core.pool_contains(b.ptr):
+            grown = b
+            break
+    assert grown is not None, "growth slab never engaged"
+    # grown memory is usable (device round-trip) but not wire-referenceable
+    data = np.random.default_rng(9).standard_normal(1024).astype(np.float32)
+    grown.copy_from(data)
+    assert np.allclose(grown.to_numpy_f32()[:1024], data)
+    del held, grown, normal  # release back to the pool
