@@ -21,6 +21,9 @@ def test_bench_cpu_smoke():
     assert j["n_gpus"] == 1
     assert j["higher_is_better"] is True
     assert j["data"] == "synthetic"
+    # the metric string is the driver's join key against BASELINE.json
+    baseline = json.loads((REPO / "BASELINE.json").read_text())
+    assert j["metric"] == baseline["metric"], (j["metric"], baseline["metric"])
 
 
 def test_rn50_buckets_shape():
