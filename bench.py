@@ -54,7 +54,10 @@ def parse_args():
                    help="payload dtype for the dense sum/reduce kernels "
                         "(bf16 halves bytes moved; headline stays fp32)")
     p.add_argument("--batch-keys", action="store_true",
-                   help="dense: one multi-key message per server per round")
+                   help="dense: one multi-key message per server per round. "
+                        "Meant for SMALL keys (e.g. --size-mb 1): the batched "
+                        "buffer must stay under the <2 GiB hipIpc slab limit "
+                        "or it falls off the zero-copy path")
     p.add_argument("--per-key", action="store_true",
                    help="rn50: legacy one-message-per-bucket reduce rounds "
                         "(default is one multi-key message per server per round)")

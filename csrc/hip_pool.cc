@@ -92,6 +92,11 @@ void* HbmPool::Alloc(size_t nbytes) {
     // oversized (> the <2 GiB hipIpc slab ceiling): dedicated local-only
     // slab — cannot be zero-copy-shared, but torch-allocator users may
     // hold multi-GiB tensors that never cross the wire
+    XPS_LOG(Warning) << "pool allocation of " << (nbytes >> 20)
+                     << " MiB exceeds the exported slab size (" << (slab_bytes_ >> 20)
+                     << " MiB, a <2 GiB hipIpc limit): it gets a LOCAL-ONLY slab and "
+                        "any wire traffic from it stages over TCP — split the buffer "
+                        "into <= slab-sized pieces to stay zero-copy";
     std::lock_guard<std::mutex> lk(mu_);
     Slab s;
     s.capacity = nbytes;
