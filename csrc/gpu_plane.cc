@@ -566,7 +566,17 @@ bool GpuPlane::LocalPullRead(int peer_id, void* dst, uint64_t entry_off, size_t 
   char* src = ResolvePeer(p, entry_off, len);
   if (!src) return false;
   XPS_STAGE(local_pull_read);
-  hipStream_t stream = StreamForPeer(peer_id);  // lane 0: ordered after entry pushes
+  // lane 0 of the stream our entry pushes used: for a same-process peer
+  // that is the RECEIVER plane's stream for us (also the stream its
+  // handler kernels run on), else our own per-peer stream
+  hipStream_t stream;
+  if (Postoffice* elpo = LocalPeer(p)) {
+    auto* rplane = dynamic_cast<GpuPlane*>(elpo->van() ? elpo->van()->plane() : nullptr);
+    if (!rplane) return false;
+    stream = rplane->StreamForPeer(po_->node_id());
+  } else {
+    stream = StreamForPeer(peer_id);
+  }
   XPS_HIP_CHECK(hipSetDevice(device_));
   kern::DenseAssign(dst, src, len, stream);
   g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
@@ -729,19 +739,27 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   // 356): the worker writes the server's advertised store entry with
   // ITS OWN kernel and sends the meta-only notification once the write
   // completes. The server's push handling shrinks to an ack.
+  static const bool local_one_sided =
+      Environment::Get()->GetInt("XPS_LOCAL_ONE_SIDED", 1) != 0;
   if (msg.meta.request && msg.meta.push && !msg.meta.pull &&
       (msg.meta.option & kOptEntryPush) && device_ >= 0 && msg.data.size() > 1 &&
-      msg.data[1].on_device() && !LocalPeer(p)) {
-    // (same-PROCESS peers skip this: there is no transport to save, and
-    // writing here just moves the kernel launch from the server's
-    // delivery thread onto this one — measured slower. Cross-process it
-    // removes the server-side launch entirely: 2012 -> 2603 GB/s on the
-    // 2-joint-procs-1-GPU config.)
+      msg.data[1].on_device()) {
+    // Cross-process this removes the server-side launch entirely
+    // (2012 -> 2603 GB/s on the 2-joint-procs-1-GPU config). For a
+    // SAME-process peer the write runs on the RECEIVER plane's stream
+    // for us — the same stream its handler kernels and our one-sided
+    // reads use, so ordering is by stream, and the (seq-consistent)
+    // notification just acks.
+    Postoffice* elpo = LocalPeer(p);
+    GpuPlane* rplane =
+        elpo ? dynamic_cast<GpuPlane*>(elpo->van() ? elpo->van()->plane() : nullptr) : nullptr;
+    bool can = elpo ? (local_one_sided && rplane != nullptr) : EnsureRing(p);
     SArray<char> vals = msg.data[1];
-    char* dst = EnsureRing(p) ? ResolvePeer(p, msg.meta.addr, vals.size()) : nullptr;
+    char* dst = can ? ResolvePeer(p, msg.meta.addr, vals.size()) : nullptr;
     if (dst) {
       XPS_STAGE(entry_push);
-      hipStream_t stream = StreamForPeer(p->node.id);
+      hipStream_t stream = elpo ? rplane->StreamForPeer(po_->node_id())
+                                : StreamForPeer(p->node.id);
       XPS_HIP_CHECK(hipSetDevice(device_));
       kern::DenseAssign(dst, vals.data(), vals.size(), stream);
       g_zero_copy_recv.fetch_add(1, std::memory_order_relaxed);
@@ -759,7 +777,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
       Message keepalive;
       keepalive.data.push_back(vals);
       std::string payload;
-      {
+      if (!elpo) {
         std::vector<char> br(meta_msg.data.size(), 0);
         XPS_CHECK(Serialize(meta_msg, br, &payload));
       }
@@ -769,7 +787,7 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
         std::lock_guard<std::mutex> lk(pend_mu_);
         pending_[p->node.id].push_back(Pending{ev, p->node.id, std::move(payload),
                                                std::move(meta_msg), std::move(keepalive),
-                                               bytes, nullptr});
+                                               bytes, elpo});
       }
       pending_count_.fetch_add(1);
       p->tx_bytes.fetch_add(bytes, std::memory_order_relaxed);
