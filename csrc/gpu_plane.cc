@@ -740,16 +740,17 @@ int64_t GpuPlane::Send(Message& msg, const Node& peer_node) {
   // ITS OWN kernel and sends the meta-only notification once the write
   // completes. The server's push handling shrinks to an ack.
   static const bool local_one_sided =
-      Environment::Get()->GetInt("XPS_LOCAL_ONE_SIDED", 1) != 0;
+      Environment::Get()->GetInt("XPS_LOCAL_ONE_SIDED", 0) != 0;
   if (msg.meta.request && msg.meta.push && !msg.meta.pull &&
       (msg.meta.option & kOptEntryPush) && device_ >= 0 && msg.data.size() > 1 &&
       msg.data[1].on_device()) {
     // Cross-process this removes the server-side launch entirely
     // (2012 -> 2603 GB/s on the 2-joint-procs-1-GPU config). For a
-    // SAME-process peer the write runs on the RECEIVER plane's stream
-    // for us — the same stream its handler kernels and our one-sided
-    // reads use, so ordering is by stream, and the (seq-consistent)
-    // notification just acks.
+    // SAME-process peer it is measured SLOWER (61 vs 84 GB/s per-key
+    // 1 MB, RTT 50 vs 42 us, same-box A/B: the app thread becomes the
+    // single launch lane) — default off locally; XPS_LOCAL_ONE_SIDED=1
+    // re-enables it (the write then runs on the RECEIVER plane's stream
+    // for us, so ordering stays stream-based).
     Postoffice* elpo = LocalPeer(p);
     GpuPlane* rplane =
         elpo ? dynamic_cast<GpuPlane*>(elpo->van() ? elpo->van()->plane() : nullptr) : nullptr;
