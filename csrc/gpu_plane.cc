@@ -573,6 +573,13 @@ bool GpuPlane::LocalPullRead(int peer_id, void* dst, uint64_t entry_off, size_t 
   if (Postoffice* elpo = LocalPeer(p)) {
     auto* rplane = dynamic_cast<GpuPlane*>(elpo->van() ? elpo->van()->plane() : nullptr);
     if (!rplane) return false;
+    // ordering relies on the preceding LOCAL push's kernel being
+    // enqueued on this same stream BEFORE we enqueue the read — true
+    // only with synchronous inline request delivery (the default). With
+    // queued delivery the push launch races us: fall back to a normal
+    // pull request, which the seq gate orders.
+    static const bool queue_req = Environment::Get()->GetInt("XPS_LOCAL_QUEUE_REQ", 0) != 0;
+    if (!inline_deliver_ || queue_req) return false;
     stream = rplane->StreamForPeer(po_->node_id());
   } else {
     stream = StreamForPeer(peer_id);
